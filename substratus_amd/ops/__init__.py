@@ -1,0 +1,174 @@
+"""Op dispatch: hand-written CDNA4 HIP kernels on GPU, reference PyTorch on CPU.
+
+Policy (task contract): on a ROCm GPU the HIP extension is MANDATORY — if a
+tensor lives on a CUDA/HIP device and `substratus_amd._C` is not importable the
+op raises instead of silently falling back to eager PyTorch. CPU tensors run
+the fp32 reference implementations (substratus_amd/ops/ref.py), which are also
+the parity oracle for the GPU tests.
+"""
+
+from __future__ import annotations
+
+import os
+from typing import Optional
+
+import torch
+
+from . import ref
+
+_EXT = None
+_EXT_ERR: Optional[str] = None
+
+
+def _load_ext():
+    global _EXT, _EXT_ERR
+    if _EXT is not None or _EXT_ERR is not None:
+        return _EXT
+    try:
+        from substratus_amd import _C  # built in-tree by setup.py build_ext --inplace
+        _EXT = _C
+    except ImportError as e:  # pragma: no cover - exercised only without the .so
+        _EXT_ERR = str(e)
+    return _EXT
+
+
+def have_ext() -> bool:
+    return _load_ext() is not None
+
+
+def _dispatch(t: torch.Tensor):
+    """Return the extension for GPU tensors (raising loudly if absent), None for CPU."""
+    if t.is_cuda:
+        ext = _load_ext()
+        if ext is None:
+            raise RuntimeError(
+                "substratus_amd HIP extension (substratus_amd/_C*.so) is not built "
+                f"but a GPU tensor was passed — no eager fallback on GPU. Build it "
+                f"with `python setup.py build_ext --inplace` (PYTORCH_ROCM_ARCH=gfx950). "
+                f"Import error: {_EXT_ERR}")
+        return ext
+    return None
+
+
+def rms_norm(x: torch.Tensor, weight: torch.Tensor, eps: float) -> torch.Tensor:
+    ext = _dispatch(x)
+    if ext is None:
+        return ref.rms_norm(x, weight, eps)
+    out = torch.empty_like(x)
+    ext.rms_norm(out, x.contiguous(), weight, eps)
+    return out
+
+
+def rms_norm_residual(x: torch.Tensor, residual: torch.Tensor,
+                      weight: torch.Tensor, eps: float) -> tuple[torch.Tensor, torch.Tensor]:
+    """Fused: residual += x; y = rms_norm(residual). Returns (y, residual).
+
+    The producing kernel fuses the residual add so the hidden state is read
+    once from HBM (guide: fuse elementwise work into the producing kernel).
+    """
+    ext = _dispatch(x)
+    if ext is None:
+        residual = residual + x
+        return ref.rms_norm(residual, weight, eps), residual
+    y = torch.empty_like(x)
+    ext.rms_norm_residual(y, residual, x.contiguous(), weight, eps)
+    return y, residual
+
+
+def layer_norm(x: torch.Tensor, weight: torch.Tensor, bias: torch.Tensor,
+               eps: float) -> torch.Tensor:
+    ext = _dispatch(x)
+    if ext is None:
+        return ref.layer_norm(x, weight, bias, eps)
+    out = torch.empty_like(x)
+    ext.layer_norm(out, x.contiguous(), weight, bias, eps)
+    return out
+
+
+def rope_kv_append(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
+                   positions: torch.Tensor, k_cache: torch.Tensor,
+                   v_cache: torch.Tensor, slots: torch.Tensor,
+                   theta: float) -> torch.Tensor:
+    ext = _dispatch(q)
+    if ext is None:
+        return ref.rope_kv_append(q, k, v, positions, k_cache, v_cache, slots, theta)
+    q = q.contiguous()
+    ext.rope_kv_append(q, k.contiguous(), v.contiguous(),
+                       positions, k_cache, v_cache, slots, theta)
+    return q
+
+
+def paged_attn_decode(q: torch.Tensor, k_cache: torch.Tensor,
+                      v_cache: torch.Tensor, block_table: torch.Tensor,
+                      seq_lens: torch.Tensor, scale: float) -> torch.Tensor:
+    ext = _dispatch(q)
+    if ext is None:
+        return ref.paged_attn_decode(q, k_cache, v_cache, block_table, seq_lens, scale)
+    out = torch.empty_like(q)
+    ext.paged_attn_decode(out, q.contiguous(), k_cache, v_cache,
+                          block_table, seq_lens, scale)
+    return out
+
+
+def attn_prefill(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
+                 seq_start: torch.Tensor, scale: float) -> torch.Tensor:
+    ext = _dispatch(q)
+    if ext is None:
+        return ref.attn_prefill(q, k, v, seq_start, scale)
+    out = torch.empty_like(q)
+    max_len = int((seq_start[1:] - seq_start[:-1]).max())
+    ext.attn_prefill(out, q.contiguous(), k.contiguous(), v.contiguous(),
+                     seq_start, max_len, scale)
+    return out
+
+
+def silu_mul(x: torch.Tensor) -> torch.Tensor:
+    ext = _dispatch(x)
+    if ext is None:
+        return ref.silu_mul(x)
+    out = x.new_empty(*x.shape[:-1], x.shape[-1] // 2)
+    ext.silu_mul(out, x.contiguous())
+    return out
+
+
+def gelu(x: torch.Tensor) -> torch.Tensor:
+    ext = _dispatch(x)
+    if ext is None:
+        return ref.gelu(x)
+    out = torch.empty_like(x)
+    ext.gelu_tanh(out, x.contiguous())
+    return out
+
+
+def softmax_sample(logits: torch.Tensor, temperature: torch.Tensor,
+                   top_k: torch.Tensor, top_p: torch.Tensor,
+                   uniform: torch.Tensor) -> torch.Tensor:
+    ext = _dispatch(logits)
+    if ext is None:
+        return ref.softmax_sample(logits, temperature, top_k, top_p, uniform)
+    out = torch.empty(logits.shape[0], dtype=torch.int64, device=logits.device)
+    ext.softmax_sample(out, logits.contiguous(), temperature, top_k, top_p, uniform)
+    return out
+
+
+def cross_entropy_fwd_bwd(logits: torch.Tensor, targets: torch.Tensor,
+                          ignore_index: int = -100) -> tuple[torch.Tensor, torch.Tensor]:
+    ext = _dispatch(logits)
+    if ext is None:
+        return ref.cross_entropy_fwd_bwd(logits, targets, ignore_index)
+    loss = torch.zeros((), dtype=torch.float32, device=logits.device)
+    dlogits = torch.empty_like(logits)
+    ext.cross_entropy_fwd_bwd(loss, dlogits, logits.contiguous(), targets, ignore_index)
+    return loss, dlogits
+
+
+def adamw_step(param: torch.Tensor, grad: torch.Tensor, exp_avg: torch.Tensor,
+               exp_avg_sq: torch.Tensor, lr: float, beta1: float, beta2: float,
+               eps: float, weight_decay: float, step: int) -> None:
+    ext = _dispatch(param)
+    if ext is None:
+        ref.adamw_step(param, grad, exp_avg, exp_avg_sq, lr, beta1, beta2,
+                       eps, weight_decay, step)
+        return
+    ext.adamw_step(param, grad, exp_avg, exp_avg_sq, lr, beta1, beta2,
+                   eps, weight_decay, step)

@@ -1,0 +1,128 @@
+"""CPU tests of the serve engine: continuous batching over the paged KV cache.
+
+Mirrors the reference's system-test success criterion (a completion request
+returns tokens — test/system.sh:73-77) at the engine level, plus correctness
+properties the reference never had: paged decode must equal a dense forward.
+"""
+
+import pytest
+import torch
+
+from substratus_amd.models import CausalLM, get_config
+from substratus_amd.serve.engine import Engine, EngineConfig
+
+
+@pytest.fixture(scope="module")
+def tiny_engine():
+    cfg = EngineConfig(model="llama-tiny", dtype=torch.float32, max_batch=4,
+                       max_seq_len=128, kv_blocks=64)
+    return Engine(cfg, device="cpu")
+
+
+def test_greedy_generation_finishes(tiny_engine):
+    eng = tiny_engine
+    sid = eng.add_request([1, 2, 3, 4, 5], max_new_tokens=8)
+    toks = []
+    for _ in range(20):
+        for out in eng.step():
+            if out.seq_id == sid:
+                toks.append(out.token)
+        if not eng.has_work():
+            break
+    assert len(toks) == 8
+    assert not eng.running and not eng.waiting
+    assert eng.kv.allocator.free_count == 64  # all blocks freed
+
+
+def test_continuous_batching_interleaves(tiny_engine):
+    eng = tiny_engine
+    a = eng.add_request([1, 2, 3], max_new_tokens=4)
+    b = eng.add_request([4, 5, 6, 7, 8, 9, 10], max_new_tokens=6)
+    got = {a: [], b: []}
+    for _ in range(30):
+        for out in eng.step():
+            got[out.seq_id].append(out.token)
+        if not eng.has_work():
+            break
+    assert len(got[a]) == 4
+    assert len(got[b]) == 6
+
+
+def test_paged_decode_matches_dense_forward():
+    """Greedy tokens from the paged engine must equal a dense forward_train
+    argmax continuation — the decode path's end-to-end numerics oracle."""
+    torch.manual_seed(0)
+    mcfg = get_config("llama-tiny")
+    model = CausalLM(mcfg, device="cpu", dtype=torch.float32)
+    model.init_random(seed=3)
+    eng_cfg = EngineConfig(model="llama-tiny", dtype=torch.float32,
+                           max_batch=2, max_seq_len=64, kv_blocks=32)
+    eng = Engine(eng_cfg, device="cpu", model=model)
+    prompt = [5, 9, 2, 7, 1, 3]
+    sid = eng.add_request(list(prompt), max_new_tokens=5)
+    paged_tokens = []
+    while eng.has_work():
+        for out in eng.step():
+            paged_tokens.append(out.token)
+    # dense reference: feed the whole growing sequence each step
+    seq = list(prompt)
+    dense_tokens = []
+    with torch.no_grad():
+        for _ in range(5):
+            logits = model.forward_train(torch.tensor([seq]))
+            nxt = int(logits[0, -1].float().argmax())
+            dense_tokens.append(nxt)
+            seq.append(nxt)
+    assert paged_tokens == dense_tokens
+
+
+def test_sampling_params_accepted(tiny_engine):
+    eng = tiny_engine
+    sid = eng.add_request([1, 2, 3], max_new_tokens=3, temperature=0.8,
+                          top_k=50, top_p=0.9)
+    n = 0
+    while eng.has_work():
+        n += len([o for o in eng.step() if o.seq_id == sid])
+    assert n == 3
+
+
+def test_preemption_under_block_pressure():
+    cfg = EngineConfig(model="llama-tiny", dtype=torch.float32, max_batch=4,
+                       max_seq_len=64, kv_blocks=10)
+    eng = Engine(cfg, device="cpu")
+    for i in range(3):
+        eng.add_request(list(range(1, 17)), max_new_tokens=12)
+    counts = {}
+    for _ in range(200):
+        for out in eng.step():
+            counts[out.seq_id] = counts.get(out.seq_id, 0) + 1
+        if not eng.has_work():
+            break
+    assert not eng.has_work(), "requests starved under block pressure"
+    # every request eventually produced its tokens (>= because preempted
+    # sequences re-emit their regenerated continuation)
+    for sid, n in counts.items():
+        assert n >= 12
+    assert eng.kv.allocator.free_count == 10
+
+
+def test_falcon_and_opt_families_run():
+    for name in ("falcon-tiny", "opt-125m"):
+        mcfg = get_config(name)
+        if name == "opt-125m":
+            # shrink for CPU test speed
+            from substratus_amd.models.config import ModelConfig, register
+            mcfg = register(ModelConfig(
+                name="opt-tiny", family="opt", hidden_size=128,
+                intermediate_size=512, num_layers=2, num_heads=4,
+                num_kv_heads=4, vocab_size=256, max_position_embeddings=256,
+                learned_pos_emb=True, activation="gelu",
+                tie_word_embeddings=True))
+        cfg = EngineConfig(model=mcfg.name, dtype=torch.float32, max_batch=2,
+                           max_seq_len=64, kv_blocks=16)
+        eng = Engine(cfg, device="cpu")
+        sid = eng.add_request([1, 2, 3, 4], max_new_tokens=3)
+        toks = []
+        while eng.has_work():
+            toks.extend(o.token for o in eng.step() if o.seq_id == sid)
+        assert len(toks) == 3, name
