@@ -1,0 +1,157 @@
+#!/usr/bin/env python3
+"""Flagship benchmark: llama2-7b bf16 serving throughput (served tokens/sec).
+
+This measures BASELINE.json's headline metric — served tokens/sec for the
+llama2-7b Server config — on synthetic prompts and random-init weights (no
+network for datasets/checkpoints). One step = one continuous-batching decode
+iteration over the whole running batch, every op on the hand-written CDNA4
+HIP path.
+
+Contract (driver): `python bench.py --gpus N --steps K --warmup W`; for N>1
+launched via torch.distributed.run with one rank per GPU over RCCL. Each rank
+runs an independent engine replica (weak scaling, data-parallel serving —
+replicated Servers are the reference's scale-out shape; TP=8 for llama2-70b is
+exercised separately). Prefill + W decode steps are untimed warmup; exactly K
+decode steps are timed between barrier+synchronize pairs; the slowest rank's
+time is used; rank 0 prints ONE JSON line.
+
+Also supports `--mode train` (fine-tune tokens/s, falcon-7b-instruct config —
+BASELINE config 3) once the trainer lands.
+"""
+
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import time
+
+import torch
+
+
+def parse_args():
+    p = argparse.ArgumentParser()
+    p.add_argument("--gpus", type=int, default=1)
+    p.add_argument("--steps", type=int, default=64)
+    p.add_argument("--warmup", type=int, default=16)
+    p.add_argument("--mode", choices=["serve", "train"], default="serve")
+    p.add_argument("--model", default=None, help="override model config")
+    p.add_argument("--batch", type=int, default=64)
+    p.add_argument("--prompt-len", type=int, default=512)
+    p.add_argument("--seq-len", type=int, default=None)
+    p.add_argument("--tp", type=int, default=1,
+                   help="tensor-parallel degree (serve); default DP replicas")
+    return p.parse_args()
+
+
+def log(msg: str) -> None:
+    if int(os.environ.get("RANK", "0")) == 0:
+        print(msg, flush=True)
+
+
+def bench_serve(args) -> None:
+    from substratus_amd.parallel.comm import init_distributed, state, barrier
+    import torch.distributed as dist
+    from substratus_amd.serve.engine import Engine, EngineConfig
+
+    st = init_distributed(tp_size=args.tp)
+    n_gpus = max(st.world_size, 1)
+    model_name = args.model or "llama2-7b"
+    on_gpu = torch.cuda.is_available()
+    device = f"cuda:{st.local_rank}" if on_gpu else "cpu"
+    dtype = torch.bfloat16 if on_gpu else torch.float32
+
+    max_seq = args.prompt_len + args.warmup + args.steps + 64
+    cfg = EngineConfig(model=model_name, dtype=dtype, max_batch=args.batch,
+                       max_seq_len=max_seq,
+                       prefill_chunk=args.batch * args.prompt_len,
+                       seed=1234 + st.dp_rank)
+    log(f"[bench] building {model_name} engine on {device} ...")
+    t_build = time.time()
+    eng = Engine(cfg, device=device)
+    log(f"[bench] engine ready in {time.time()-t_build:.1f}s; "
+        f"kv_blocks={eng.kv.num_blocks}")
+
+    g = torch.Generator().manual_seed(7 + st.rank)
+    vocab = eng.mcfg.vocab_size
+    for _ in range(args.batch):
+        prompt = torch.randint(0, vocab, (args.prompt_len,), generator=g).tolist()
+        # never finishes inside the timed region (+63 headroom < max_seq)
+        eng.add_request(prompt, max_new_tokens=args.warmup + args.steps + 32,
+                        temperature=0.0)
+
+    # prefill all requests (untimed), then warmup decode steps
+    t_pref = time.time()
+    while eng.waiting:
+        eng.step()
+    log(f"[bench] prefill of {args.batch}x{args.prompt_len} tokens: "
+        f"{time.time()-t_pref:.2f}s")
+    assert len(eng.running) == args.batch
+    for _ in range(args.warmup):
+        eng.decode_only_step()
+
+    if on_gpu:
+        torch.cuda.synchronize()
+    barrier()
+    t0 = time.time()
+    for _ in range(args.steps):
+        eng.decode_only_step()
+    if on_gpu:
+        torch.cuda.synchronize()
+    barrier()
+    elapsed = time.time() - t0
+    # max over ranks
+    if st.initialized:
+        t = torch.tensor([elapsed], dtype=torch.float64)
+        if on_gpu:
+            t = t.to(device)
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        elapsed = float(t.item())
+
+    assert len(eng.running) == args.batch, "sequences finished mid-bench"
+    per_rank_batch = args.batch
+    tokens = per_rank_batch * args.steps * n_gpus
+    value = tokens / elapsed
+    ms_per_step = elapsed / args.steps * 1000.0
+    parallelism = (f"tp{args.tp}" if args.tp > 1 else f"dp{n_gpus}")
+    result = {
+        "metric": "served_tokens_per_s",
+        "value": round(value, 2),
+        "unit": "tokens/s",
+        "n_gpus": n_gpus,
+        "steps": args.steps,
+        "warmup": args.warmup,
+        "ms_per_step": round(ms_per_step, 3),
+        "higher_is_better": True,
+        "scaling": "weak",
+        "vs_baseline": None,   # reference publishes no numbers
+        "dtype": "bf16" if on_gpu else "fp32",
+        "data": "synthetic",
+        "config": {
+            "model": model_name,
+            "global_batch": per_rank_batch * n_gpus,
+            "seq_len": args.prompt_len + args.warmup + args.steps,
+            "prompt_len": args.prompt_len,
+            "parallelism": parallelism,
+            "p50_decode_latency_ms": round(ms_per_step, 3),
+        },
+    }
+    if st.rank == 0:
+        print(json.dumps(result), flush=True)
+
+
+def bench_train(args) -> None:
+    from substratus_amd.train.bench import run_train_bench
+    run_train_bench(args)
+
+
+def main() -> None:
+    args = parse_args()
+    if args.mode == "serve":
+        bench_serve(args)
+    else:
+        bench_train(args)
+
+
+if __name__ == "__main__":
+    main()

@@ -1,0 +1,38 @@
+"""Build the CDNA4 HIP extension in-tree:
+
+    PYTORCH_ROCM_ARCH=gfx950 python setup.py build_ext --inplace
+
+The .so lands at substratus_amd/_C*.so (git-ignored; it travels to the GPU box
+with the gpurun snapshot).
+"""
+
+import os
+
+from setuptools import setup
+
+os.environ.setdefault("PYTORCH_ROCM_ARCH", "gfx950")
+
+from torch.utils.cpp_extension import BuildExtension, CUDAExtension  # noqa: E402
+
+ROOT = os.path.dirname(os.path.abspath(__file__))
+HIP_DIR = os.path.join(ROOT, "substratus_amd", "ops", "hip")
+
+sources = [os.path.join(HIP_DIR, f) for f in sorted(os.listdir(HIP_DIR))
+           if f.endswith(".hip")]
+
+setup(
+    name="substratus_amd",
+    version="0.1.0",
+    packages=["substratus_amd"],
+    ext_modules=[
+        CUDAExtension(
+            name="substratus_amd._C",
+            sources=sources,
+            extra_compile_args={
+                "cxx": ["-O3", "-std=c++17"],
+                "nvcc": ["-O3", "-std=c++17"],
+            },
+        )
+    ],
+    cmdclass={"build_ext": BuildExtension},
+)

@@ -84,9 +84,10 @@ register(ModelConfig(
     num_layers=80, num_heads=64, num_kv_heads=8, vocab_size=32000,
     max_position_embeddings=4096, rms_norm_eps=1e-5))
 # A tiny llama for CPU tests and CI.
+# Tiny llama for CPU tests/CI and GPU numerics tests (head_dim=64, GQA=2).
 register(ModelConfig(
     name="llama-tiny", family="llama", hidden_size=256, intermediate_size=688,
-    num_layers=4, num_heads=8, num_kv_heads=4, vocab_size=512,
+    num_layers=4, num_heads=4, num_kv_heads=2, vocab_size=512,
     max_position_embeddings=2048, rms_norm_eps=1e-5))
 
 # Falcon-7b-instruct (reference: examples/falcon-7b-instruct/finetuned-model.yaml).
@@ -98,7 +99,7 @@ register(ModelConfig(
     parallel_attn=True, activation="gelu", tie_word_embeddings=True))
 register(ModelConfig(
     name="falcon-tiny", family="falcon", hidden_size=256,
-    intermediate_size=1024, num_layers=2, num_heads=8, num_kv_heads=1,
+    intermediate_size=1024, num_layers=2, num_heads=4, num_kv_heads=1,
     vocab_size=512, max_position_embeddings=2048, parallel_attn=True,
     activation="gelu", tie_word_embeddings=True))
 

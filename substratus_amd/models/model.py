@@ -231,12 +231,11 @@ class CausalLM(nn.Module):
 
     @torch.no_grad()
     def init_random(self, seed: int = 0) -> None:
-        g = torch.Generator(device="cpu").manual_seed(seed)
+        dev = next(self.parameters()).device
+        g = torch.Generator(device=dev).manual_seed(seed)
         for p in self.parameters():
             if p.dim() >= 2:
-                init = torch.empty(p.shape, dtype=torch.float32)
-                init.normal_(0.0, 0.02, generator=g)
-                p.copy_(init.to(p.dtype))
+                p.normal_(0.0, 0.02, generator=g)
 
     # -- decode --------------------------------------------------------------
     def forward_decode(self, tokens: torch.Tensor, positions: torch.Tensor,

@@ -1,0 +1,70 @@
+#include "hip/hip_runtime.h"
+// Fused activation kernels: SwiGLU (silu(gate)*up) and tanh-GELU.
+// Pure HBM-streaming ops; bf16×8 vector loads/stores, fp32 math.
+#include "common.h"
+#include <algorithm>
+
+// x: [rows, 2F] laid out (gate ‖ up); out: [rows, F].
+template <int BLOCK>
+__global__ void silu_mul_kernel(bf16_t* __restrict__ out,
+                                const bf16_t* __restrict__ x, long rows,
+                                int f) {
+  const long row = blockIdx.x;
+  const uint4* gv = reinterpret_cast<const uint4*>(x + row * 2L * f);
+  const uint4* uv = reinterpret_cast<const uint4*>(x + row * 2L * f + f);
+  uint4* ov = reinterpret_cast<uint4*>(out + row * (long)f);
+  const int nvec = f / 8;
+  for (int i = threadIdx.x; i < nvec; i += BLOCK) {
+    uint4 g = gv[i], u = uv[i];
+    float2 gf[4] = {unpack_bf16x2(g.x), unpack_bf16x2(g.y),
+                    unpack_bf16x2(g.z), unpack_bf16x2(g.w)};
+    float2 uf[4] = {unpack_bf16x2(u.x), unpack_bf16x2(u.y),
+                    unpack_bf16x2(u.z), unpack_bf16x2(u.w)};
+    unsigned int o[4];
+    #pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      float a = gf[j].x, b = gf[j].y;
+      float sa = a / (1.f + __expf(-a));
+      float sb = b / (1.f + __expf(-b));
+      o[j] = pack_bf16x2(sa * uf[j].x, sb * uf[j].y);
+    }
+    ov[i] = make_uint4(o[0], o[1], o[2], o[3]);
+  }
+}
+
+DEV_INLINE float gelu_tanh_f(float v) {
+  const float c = 0.7978845608028654f;  // sqrt(2/pi)
+  float inner = c * (v + 0.044715f * v * v * v);
+  return 0.5f * v * (1.f + tanhf(inner));
+}
+
+template <int BLOCK>
+__global__ void gelu_kernel(bf16_t* __restrict__ out,
+                            const bf16_t* __restrict__ x, long n8) {
+  const uint4* xv = reinterpret_cast<const uint4*>(x);
+  uint4* ov = reinterpret_cast<uint4*>(out);
+  for (long i = blockIdx.x * BLOCK + threadIdx.x; i < n8;
+       i += (long)gridDim.x * BLOCK) {
+    uint4 u = xv[i];
+    float2 f[4] = {unpack_bf16x2(u.x), unpack_bf16x2(u.y),
+                   unpack_bf16x2(u.z), unpack_bf16x2(u.w)};
+    unsigned int o[4];
+    #pragma unroll
+    for (int j = 0; j < 4; ++j)
+      o[j] = pack_bf16x2(gelu_tanh_f(f[j].x), gelu_tanh_f(f[j].y));
+    ov[i] = make_uint4(o[0], o[1], o[2], o[3]);
+  }
+}
+
+void launch_silu_mul(bf16_t* out, const bf16_t* x, long rows, int f,
+                     hipStream_t stream) {
+  constexpr int BLOCK = 256;
+ hipLaunchKernelGGL(( silu_mul_kernel<BLOCK>), dim3(dim3(rows)), dim3(dim3(BLOCK)), 0, stream, out, x, rows, f);
+}
+
+void launch_gelu(bf16_t* out, const bf16_t* x, long n, hipStream_t stream) {
+  constexpr int BLOCK = 256;
+  long n8 = n / 8;
+  int grid = (int)std::min((n8 + BLOCK - 1) / BLOCK, (long)4096);
+ hipLaunchKernelGGL(( gelu_kernel<BLOCK>), dim3(dim3(grid)), dim3(dim3(BLOCK)), 0, stream, out, x, n8);
+}

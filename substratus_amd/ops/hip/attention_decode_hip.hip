@@ -1,0 +1,217 @@
+#include "hip/hip_runtime.h"
+// Paged-KV decode attention (SURVEY.md §2.5 ledger item 1) — the serving hot
+// kernel: per sequence and KV head, softmax(q·K^T/√d)·V over a block-paged
+// cache, one new token per sequence.
+//
+// Decode at batch B is HBM-bound on the K/V stream (e.g. llama2-7b @ctx 512,
+// B=64: ~16 GB of KV per step vs 13.5 GB of weights), so the kernel is
+// organized around wide coalesced loads of 4 KiB cache panels:
+//
+//   * grid = (num_seqs, n_kv_heads, ceil(group/GC)); 4 waves per workgroup.
+//   * Each wave owns KV blocks stride-4 (flash-decoding style split over the
+//     sequence) keeping ≥ (16 tokens × D/8-lane uint4) loads in flight.
+//   * K panel row [16, D] bf16: D/8 lanes × uint4 = one full 256 B row per
+//     token; scores reduced via shfl within the lane group (no LDS on the
+//     score path; guide §5 common-mistake #6: wave-parallel, never serial).
+//   * Online softmax per wave; P·V with dims distributed over lanes (uint
+//     bf16×2 loads, conflict-free); cross-wave merge via LDS at the end.
+//
+// Cache layout: [num_blocks, n_kv_heads, BS=16, D]; block_table [B, maxb]
+// int32; seq_lens [B] int32 counts INCLUDING the current token.
+#include "common.h"
+
+#define BS 16   // tokens per KV block
+#define NW 4    // waves per workgroup
+
+template <int D, int GC>
+__global__ __launch_bounds__(NW * WAVE)
+void paged_attn_decode_kernel(
+    bf16_t* __restrict__ out,            // [B, HQ, D]
+    const bf16_t* __restrict__ q,        // [B, HQ, D]
+    const bf16_t* __restrict__ k_cache,  // [NB, HKV, BS, D]
+    const bf16_t* __restrict__ v_cache,
+    const int* __restrict__ block_table, // [B, max_blocks]
+    const int* __restrict__ seq_lens,    // [B]
+    int hq, int hkv, int max_blocks, float scale) {
+  constexpr int LPT = D / 8;        // lanes per token for K rows (uint4 each)
+  constexpr int TPP = WAVE / LPT;   // tokens per pass
+  constexpr int PASSES = BS / TPP;
+  constexpr int DPL = D / WAVE;     // accum dims per lane (1 or 2)
+  static_assert(DPL >= 1 && DPL <= 2, "head_dim must be 64..128");
+
+  const int b = blockIdx.x;
+  const int h = blockIdx.y;           // kv head
+  const int gchunk = blockIdx.z;      // q-head chunk within the GQA group
+  const int group = hq / hkv;
+  const int g0 = gchunk * GC;
+  const int ng = min(GC, group - g0); // live q heads this WG
+  const int wave = threadIdx.x / WAVE;
+  const int lane = threadIdx.x % WAVE;
+  const int sub = lane / LPT;         // token-within-pass
+  const int ll = lane % LPT;          // lane-within-token
+
+  const int n = seq_lens[b];
+  const int nblk = (n + BS - 1) / BS;
+
+  // ---- load this WG's q fragments: q[b][h*group+g0+g][ll*8..+8] ----------
+  float qf[GC][8];
+  #pragma unroll
+  for (int g = 0; g < GC; ++g) {
+    #pragma unroll
+    for (int j = 0; j < 8; ++j) qf[g][j] = 0.f;
+    if (g < ng) {
+      const uint4 u = reinterpret_cast<const uint4*>(
+          q + ((long)b * hq + h * group + g0 + g) * D)[ll];
+      float2 f0 = unpack_bf16x2(u.x), f1 = unpack_bf16x2(u.y);
+      float2 f2 = unpack_bf16x2(u.z), f3 = unpack_bf16x2(u.w);
+      qf[g][0] = f0.x; qf[g][1] = f0.y; qf[g][2] = f1.x; qf[g][3] = f1.y;
+      qf[g][4] = f2.x; qf[g][5] = f2.y; qf[g][6] = f3.x; qf[g][7] = f3.y;
+    }
+  }
+
+  // online-softmax state per wave
+  float m[GC], l[GC], acc[GC][DPL];
+  #pragma unroll
+  for (int g = 0; g < GC; ++g) {
+    m[g] = -INFINITY; l[g] = 0.f;
+    #pragma unroll
+    for (int j = 0; j < DPL; ++j) acc[g][j] = 0.f;
+  }
+
+  __shared__ float s_scores[NW][GC][BS];
+  const int* bt = block_table + (long)b * max_blocks;
+  const long head_off = (long)h * BS * D;
+  const long blk_stride = (long)hkv * BS * D;
+
+  for (int bi = wave; bi < nblk; bi += NW) {
+    const int blk = bt[bi];
+    const bf16_t* kbase = k_cache + (long)blk * blk_stride + head_off;
+    // ---- scores for the 16 tokens of this block ------------------------
+    #pragma unroll
+    for (int p = 0; p < PASSES; ++p) {
+      const int t = p * TPP + sub;
+      const int tok_idx = bi * BS + t;
+      const uint4 u = reinterpret_cast<const uint4*>(kbase + t * D)[ll];
+      float2 f0 = unpack_bf16x2(u.x), f1 = unpack_bf16x2(u.y);
+      float2 f2 = unpack_bf16x2(u.z), f3 = unpack_bf16x2(u.w);
+      #pragma unroll
+      for (int g = 0; g < GC; ++g) {
+        float s = qf[g][0] * f0.x + qf[g][1] * f0.y + qf[g][2] * f1.x +
+                  qf[g][3] * f1.y + qf[g][4] * f2.x + qf[g][5] * f2.y +
+                  qf[g][6] * f3.x + qf[g][7] * f3.y;
+        // reduce over the LPT lanes of this token
+        #pragma unroll
+        for (int off = LPT / 2; off > 0; off >>= 1)
+          s += __shfl_xor(s, off, WAVE);
+        if (ll == 0)
+          s_scores[wave][g][t] = (tok_idx < n) ? s * scale : -INFINITY;
+      }
+    }
+    // wave-synchronous: LDS writes above are visible to the same wave after
+    // lgkmcnt drain (compiler-inserted); no __syncthreads needed (same wave).
+    __builtin_amdgcn_wave_barrier();
+
+    // ---- online softmax + P·V -----------------------------------------
+    #pragma unroll
+    for (int g = 0; g < GC; ++g) {
+      float bm = -INFINITY;
+      #pragma unroll
+      for (int t = 0; t < BS; ++t) bm = fmaxf(bm, s_scores[wave][g][t]);
+      if (bm == -INFINITY) continue;  // fully masked block (can't happen mid-seq)
+      const float mn = fmaxf(m[g], bm);
+      const float alpha = __expf(m[g] - mn);
+      m[g] = mn;
+      l[g] *= alpha;
+      #pragma unroll
+      for (int j = 0; j < DPL; ++j) acc[g][j] *= alpha;
+      const bf16_t* vbase = v_cache + (long)blk * blk_stride + head_off;
+      const int valid = min(n - bi * BS, BS);  // mask tail of last block
+      for (int t = 0; t < valid; ++t) {
+        const float p = __expf(s_scores[wave][g][t] - mn);
+        l[g] += p;  // same value in every lane; lane 0's copy is merged
+        // lane-owned dims: D=128 → uint load (2 dims/lane);
+        // D=64 → one bf16 scalar per lane.
+        if (DPL == 2) {
+          const unsigned int uv = reinterpret_cast<const unsigned int*>(
+              vbase + t * D)[lane];
+          float2 vf = unpack_bf16x2(uv);
+          acc[g][0] += p * vf.x;
+          acc[g][1] += p * vf.y;
+        } else {
+          acc[g][0] += p * bf2f(vbase[t * D + lane]);
+        }
+      }
+    }
+  }
+
+  // ---- cross-wave merge via LDS ----------------------------------------
+  __shared__ float red_o[NW][GC][D];
+  __shared__ float red_m[NW][GC], red_l[NW][GC];
+  #pragma unroll
+  for (int g = 0; g < GC; ++g) {
+    if (lane == 0) { red_m[wave][g] = m[g]; red_l[wave][g] = l[g]; }
+    #pragma unroll
+    for (int j = 0; j < DPL; ++j)
+      red_o[wave][g][lane * DPL + j] = acc[g][j];
+  }
+  __syncthreads();
+  if (wave == 0) {
+    #pragma unroll
+    for (int g = 0; g < GC; ++g) {
+      if (g >= ng) continue;
+      float M = -INFINITY;
+      #pragma unroll
+      for (int w = 0; w < NW; ++w) M = fmaxf(M, red_m[w][g]);
+      float L = 0.f;
+      float o[DPL];
+      #pragma unroll
+      for (int j = 0; j < DPL; ++j) o[j] = 0.f;
+      #pragma unroll
+      for (int w = 0; w < NW; ++w) {
+        const float f = (red_m[w][g] == -INFINITY) ? 0.f
+                                                   : __expf(red_m[w][g] - M);
+        L += f * red_l[w][g];
+        #pragma unroll
+        for (int j = 0; j < DPL; ++j)
+          o[j] += f * red_o[w][g][lane * DPL + j];
+      }
+      const float invl = (L > 0.f) ? 1.f / L : 0.f;
+      bf16_t* orow = out + ((long)b * hq + h * group + g0 + g) * D;
+      if (DPL == 2) {
+        reinterpret_cast<unsigned int*>(orow)[lane] =
+            pack_bf16x2(o[0] * invl, o[1] * invl);
+      } else {
+        orow[lane] = f2bf(o[0] * invl);
+      }
+    }
+  }
+}
+
+void launch_paged_attn_decode(bf16_t* out, const bf16_t* q,
+                              const bf16_t* k_cache, const bf16_t* v_cache,
+                              const int* block_table, const int* seq_lens,
+                              int bsz, int hq, int hkv, int d, int max_blocks,
+                              int block_size, float scale,
+                              hipStream_t stream) {
+  const int group = hq / hkv;
+  // GC=8 when the GQA group tiles evenly by 8 (llama2-70b, llama3-style);
+  // GC=1 otherwise (MHA group=1; falcon's prime group=71 — its MQA KV is
+  // tiny and L3-resident, so per-head K re-reads are cheap).
+  const int gc = (group % 8 == 0) ? 8 : 1;
+  const int gchunks = (group + gc - 1) / gc;
+  dim3 grid(bsz, hkv, gchunks);
+  dim3 block(NW * WAVE);
+  #define LAUNCH(DD, GG) \
+   hipLaunchKernelGGL(( paged_attn_decode_kernel<DD, GG>), dim3(grid), dim3(block), 0, stream,  \
+        out, q, k_cache, v_cache, block_table, seq_lens, hq, hkv, \
+        max_blocks, scale)
+  if (d == 128 && gc == 8) LAUNCH(128, 8);
+  else if (d == 128) LAUNCH(128, 1);
+  else if (d == 64 && gc == 8) LAUNCH(64, 8);
+  else if (d == 64) LAUNCH(64, 1);
+  #undef LAUNCH
+}
+
+bool paged_attn_decode_supported(int d, int block_size) {
+  return (d == 128 || d == 64) && block_size == BS;
+}

@@ -1,0 +1,237 @@
+// Python bindings for the CDNA4 HIP kernel library (substratus_amd._C).
+// Validation lives here; kernels are thin and trusting.
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+#include <hip/hip_runtime.h>
+#include <hip/hip_bf16.h>
+
+typedef __hip_bfloat16 bf16_t;
+
+// launchers (defined in the .hip translation units)
+void launch_rms_norm(bf16_t*, const bf16_t*, const bf16_t*, long, int, float, hipStream_t);
+void launch_rms_norm_residual(bf16_t*, bf16_t*, const bf16_t*, const bf16_t*, long, int, float, hipStream_t);
+void launch_layer_norm(bf16_t*, const bf16_t*, const bf16_t*, const bf16_t*, long, int, float, hipStream_t);
+void launch_silu_mul(bf16_t*, const bf16_t*, long, int, hipStream_t);
+void launch_gelu(bf16_t*, const bf16_t*, long, hipStream_t);
+void launch_rope_kv_append(bf16_t*, const bf16_t*, const bf16_t*, const int*, bf16_t*, bf16_t*, const int*, int, int, int, int, int, float, hipStream_t);
+void launch_paged_attn_decode(bf16_t*, const bf16_t*, const bf16_t*, const bf16_t*, const int*, const int*, int, int, int, int, int, int, float, hipStream_t);
+bool paged_attn_decode_supported(int, int);
+void launch_attn_prefill(bf16_t*, const bf16_t*, const bf16_t*, const bf16_t*, const int*, int, int, int, int, int, float, hipStream_t);
+bool attn_prefill_supported(int);
+void launch_mfma_probe(float*, const bf16_t*, const bf16_t*, hipStream_t);
+void launch_softmax_sample(long*, const bf16_t*, const float*, const int*, const float*, const float*, int, int, hipStream_t);
+void launch_cross_entropy(float*, bf16_t*, const bf16_t*, const long*, long, int, long, float, hipStream_t);
+void launch_adamw(bf16_t*, const bf16_t*, float*, float*, long, float, float, float, float, float, int, hipStream_t);
+void launch_adamw_f32(float*, const float*, float*, float*, long, float, float, float, float, float, int, hipStream_t);
+
+namespace {
+
+inline hipStream_t stream() {
+  return at::hip::getCurrentHIPStreamMasqueradingAsCUDA().stream();
+}
+
+inline bf16_t* bf(torch::Tensor& t) {
+  return reinterpret_cast<bf16_t*>(t.data_ptr<at::BFloat16>());
+}
+inline const bf16_t* bfc(const torch::Tensor& t) {
+  return reinterpret_cast<const bf16_t*>(t.data_ptr<at::BFloat16>());
+}
+
+void check_bf16(const torch::Tensor& t, const char* name) {
+  TORCH_CHECK(t.is_cuda(), name, " must be on GPU");
+  TORCH_CHECK(t.scalar_type() == torch::kBFloat16, name, " must be bf16");
+  TORCH_CHECK(t.is_contiguous(), name, " must be contiguous");
+}
+
+void post_launch_check(const char* op) {
+  hipError_t e = hipGetLastError();
+  TORCH_CHECK(e == hipSuccess, op, ": HIP launch failed: ", hipGetErrorString(e));
+}
+
+void rms_norm(torch::Tensor out, torch::Tensor x, torch::Tensor w, double eps) {
+  check_bf16(out, "out"); check_bf16(x, "x"); check_bf16(w, "w");
+  int hidden = x.size(-1);
+  TORCH_CHECK(hidden % 8 == 0, "hidden must be a multiple of 8");
+  long rows = x.numel() / hidden;
+  launch_rms_norm(bf(out), bfc(x), bfc(w), rows, hidden, (float)eps, stream());
+  post_launch_check("rms_norm");
+}
+
+void rms_norm_residual(torch::Tensor out, torch::Tensor residual,
+                       torch::Tensor x, torch::Tensor w, double eps) {
+  check_bf16(out, "out"); check_bf16(residual, "residual");
+  check_bf16(x, "x"); check_bf16(w, "w");
+  int hidden = x.size(-1);
+  TORCH_CHECK(hidden % 8 == 0, "hidden must be a multiple of 8");
+  long rows = x.numel() / hidden;
+  launch_rms_norm_residual(bf(out), bf(residual), bfc(x), bfc(w), rows,
+                           hidden, (float)eps, stream());
+  post_launch_check("rms_norm_residual");
+}
+
+void layer_norm(torch::Tensor out, torch::Tensor x, torch::Tensor w,
+                torch::Tensor b, double eps) {
+  check_bf16(out, "out"); check_bf16(x, "x"); check_bf16(w, "w");
+  check_bf16(b, "b");
+  int hidden = x.size(-1);
+  TORCH_CHECK(hidden % 8 == 0, "hidden must be a multiple of 8");
+  long rows = x.numel() / hidden;
+  launch_layer_norm(bf(out), bfc(x), bfc(w), bfc(b), rows, hidden, (float)eps,
+                    stream());
+  post_launch_check("layer_norm");
+}
+
+void silu_mul(torch::Tensor out, torch::Tensor x) {
+  check_bf16(out, "out"); check_bf16(x, "x");
+  int f = out.size(-1);
+  TORCH_CHECK(x.size(-1) == 2 * f, "x last dim must be 2*out last dim");
+  TORCH_CHECK(f % 8 == 0, "intermediate size must be a multiple of 8");
+  long rows = out.numel() / f;
+  launch_silu_mul(bf(out), bfc(x), rows, f, stream());
+  post_launch_check("silu_mul");
+}
+
+void gelu_tanh(torch::Tensor out, torch::Tensor x) {
+  check_bf16(out, "out"); check_bf16(x, "x");
+  TORCH_CHECK(x.numel() % 8 == 0, "numel must be a multiple of 8");
+  launch_gelu(bf(out), bfc(x), x.numel(), stream());
+  post_launch_check("gelu_tanh");
+}
+
+void rope_kv_append(torch::Tensor q, torch::Tensor k, torch::Tensor v,
+                    torch::Tensor positions, torch::Tensor k_cache,
+                    torch::Tensor v_cache, torch::Tensor slots, double theta) {
+  check_bf16(q, "q"); check_bf16(k, "k"); check_bf16(v, "v");
+  check_bf16(k_cache, "k_cache"); check_bf16(v_cache, "v_cache");
+  TORCH_CHECK(positions.scalar_type() == torch::kInt, "positions must be int32");
+  TORCH_CHECK(slots.scalar_type() == torch::kInt, "slots must be int32");
+  int ntok = q.size(0), hq = q.size(1), d = q.size(2);
+  int hkv = k.size(1);
+  int block_size = k_cache.size(2);
+  TORCH_CHECK(k_cache.size(1) == hkv && k_cache.size(3) == d, "cache shape");
+  TORCH_CHECK(d % 8 == 0 && (d / 2) % 8 == 0, "head_dim constraints");
+  launch_rope_kv_append(bf(q), bfc(k), bfc(v), positions.data_ptr<int>(),
+                        bf(k_cache), bf(v_cache), slots.data_ptr<int>(), ntok,
+                        hq, hkv, d, block_size, (float)theta, stream());
+  post_launch_check("rope_kv_append");
+}
+
+void paged_attn_decode(torch::Tensor out, torch::Tensor q,
+                       torch::Tensor k_cache, torch::Tensor v_cache,
+                       torch::Tensor block_table, torch::Tensor seq_lens,
+                       double scale) {
+  check_bf16(out, "out"); check_bf16(q, "q");
+  check_bf16(k_cache, "k_cache"); check_bf16(v_cache, "v_cache");
+  TORCH_CHECK(block_table.scalar_type() == torch::kInt, "block_table int32");
+  TORCH_CHECK(seq_lens.scalar_type() == torch::kInt, "seq_lens int32");
+  TORCH_CHECK(block_table.is_contiguous() && seq_lens.is_contiguous());
+  int bsz = q.size(0), hq = q.size(1), d = q.size(2);
+  int hkv = k_cache.size(1), block_size = k_cache.size(2);
+  int max_blocks = block_table.size(1);
+  TORCH_CHECK(paged_attn_decode_supported(d, block_size),
+              "paged_attn_decode: unsupported head_dim ", d, " / block_size ",
+              block_size);
+  TORCH_CHECK(hq % hkv == 0, "hq must be divisible by hkv");
+  launch_paged_attn_decode(bf(out), bfc(q), bfc(k_cache), bfc(v_cache),
+                           block_table.data_ptr<int>(),
+                           seq_lens.data_ptr<int>(), bsz, hq, hkv, d,
+                           max_blocks, block_size, (float)scale, stream());
+  post_launch_check("paged_attn_decode");
+}
+
+void attn_prefill(torch::Tensor out, torch::Tensor q, torch::Tensor k,
+                  torch::Tensor v, torch::Tensor seq_start, long max_len,
+                  double scale) {
+  check_bf16(out, "out"); check_bf16(q, "q"); check_bf16(k, "k");
+  check_bf16(v, "v");
+  TORCH_CHECK(seq_start.scalar_type() == torch::kInt, "seq_start int32");
+  int hq = q.size(1), d = q.size(2), hkv = k.size(1);
+  int num_seqs = seq_start.size(0) - 1;
+  TORCH_CHECK(attn_prefill_supported(d), "unsupported head_dim ", d);
+  TORCH_CHECK(hq % hkv == 0, "hq must be divisible by hkv");
+  launch_attn_prefill(bf(out), bfc(q), bfc(k), bfc(v),
+                      seq_start.data_ptr<int>(), num_seqs, (int)max_len, hq,
+                      hkv, d, (float)scale, stream());
+  post_launch_check("attn_prefill");
+}
+
+torch::Tensor mfma_probe(torch::Tensor a, torch::Tensor b) {
+  check_bf16(a, "a"); check_bf16(b, "b");
+  TORCH_CHECK(a.size(0) == 16 && a.size(1) == 32 && b.size(0) == 32 &&
+              b.size(1) == 16, "probe shapes: a[16,32] b[32,16]");
+  auto out = torch::empty({16, 16}, a.options().dtype(torch::kFloat));
+  launch_mfma_probe(out.data_ptr<float>(), bfc(a), bfc(b), stream());
+  post_launch_check("mfma_probe");
+  return out;
+}
+
+void softmax_sample(torch::Tensor out, torch::Tensor logits,
+                    torch::Tensor temperature, torch::Tensor top_k,
+                    torch::Tensor top_p, torch::Tensor uniform) {
+  check_bf16(logits, "logits");
+  TORCH_CHECK(out.scalar_type() == torch::kLong, "out must be int64");
+  int bsz = logits.size(0), vocab = logits.size(1);
+  TORCH_CHECK((size_t)vocab * 2 <= 160 * 1024 - 4096,
+              "vocab too large for LDS staging: ", vocab);
+  launch_softmax_sample(out.data_ptr<long>(), bfc(logits),
+                        temperature.data_ptr<float>(), top_k.data_ptr<int>(),
+                        top_p.data_ptr<float>(), uniform.data_ptr<float>(),
+                        bsz, vocab, stream());
+  post_launch_check("softmax_sample");
+}
+
+void cross_entropy_fwd_bwd(torch::Tensor loss, torch::Tensor dlogits,
+                           torch::Tensor logits, torch::Tensor targets,
+                           long ignore_index) {
+  check_bf16(dlogits, "dlogits"); check_bf16(logits, "logits");
+  TORCH_CHECK(loss.scalar_type() == torch::kFloat, "loss must be fp32");
+  TORCH_CHECK(targets.scalar_type() == torch::kLong, "targets must be int64");
+  long rows = logits.size(0);
+  int vocab = logits.size(1);
+  TORCH_CHECK(vocab % 8 == 0, "vocab must be a multiple of 8");
+  long nvalid = (targets != ignore_index).sum().item<long>();
+  float inv = nvalid > 0 ? 1.f / (float)nvalid : 0.f;
+  launch_cross_entropy(loss.data_ptr<float>(), bf(dlogits), bfc(logits),
+                       targets.data_ptr<long>(), rows, vocab, ignore_index,
+                       inv, stream());
+  post_launch_check("cross_entropy_fwd_bwd");
+}
+
+void adamw_step(torch::Tensor p, torch::Tensor g, torch::Tensor m,
+                torch::Tensor v, double lr, double beta1, double beta2,
+                double eps, double wd, long step) {
+  TORCH_CHECK(p.is_cuda() && p.is_contiguous());
+  TORCH_CHECK(m.scalar_type() == torch::kFloat && v.scalar_type() == torch::kFloat);
+  long n = p.numel();
+  TORCH_CHECK(g.numel() == n && m.numel() == n && v.numel() == n);
+  if (p.scalar_type() == torch::kBFloat16) {
+    TORCH_CHECK(g.scalar_type() == torch::kBFloat16);
+    launch_adamw(bf(p), bfc(g), m.data_ptr<float>(), v.data_ptr<float>(), n,
+                 (float)lr, (float)beta1, (float)beta2, (float)eps, (float)wd,
+                 (int)step, stream());
+  } else {
+    TORCH_CHECK(p.scalar_type() == torch::kFloat && g.scalar_type() == torch::kFloat);
+    launch_adamw_f32(p.data_ptr<float>(), g.data_ptr<float>(),
+                     m.data_ptr<float>(), v.data_ptr<float>(), n, (float)lr,
+                     (float)beta1, (float)beta2, (float)eps, (float)wd,
+                     (int)step, stream());
+  }
+  post_launch_check("adamw_step");
+}
+
+}  // namespace
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("rms_norm", &rms_norm);
+  m.def("rms_norm_residual", &rms_norm_residual);
+  m.def("layer_norm", &layer_norm);
+  m.def("silu_mul", &silu_mul);
+  m.def("gelu_tanh", &gelu_tanh);
+  m.def("rope_kv_append", &rope_kv_append);
+  m.def("paged_attn_decode", &paged_attn_decode);
+  m.def("attn_prefill", &attn_prefill);
+  m.def("mfma_probe", &mfma_probe);
+  m.def("softmax_sample", &softmax_sample);
+  m.def("cross_entropy_fwd_bwd", &cross_entropy_fwd_bwd);
+  m.def("adamw_step", &adamw_step);
+}
