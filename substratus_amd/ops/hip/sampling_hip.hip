@@ -60,14 +60,15 @@ void softmax_sample_kernel(long* __restrict__ out,          // [B]
     return;
   }
 
-  // ---- stage scaled logits into LDS (bf16) + row max ------------------
+  // ---- stage the ORIGINAL bf16 bits into LDS + row max ----------------
+  // (temperature is applied in fp32 at every read: staging pre-scaled bf16
+  // would re-quantize x/T and shift CDF boundaries vs the fp32 reference)
   const float inv_t = 1.f / temp;
   float mx = -INFINITY;
   for (int i = tid; i < vocab; i += SBLOCK) {
-    float v = bf2f(row[i]) * inv_t;
-    bf16_t h = f2bf(v);
+    bf16_t h = row[i];
     s_vals[i] = *reinterpret_cast<unsigned short*>(&h);
-    mx = fmaxf(mx, __bfloat162float(h));
+    mx = fmaxf(mx, bf2f(h) * inv_t);
   }
   __syncthreads();
   float m = block_max<SNW>(mx, scratch);
@@ -83,7 +84,7 @@ void softmax_sample_kernel(long* __restrict__ out,          // [B]
       int cnt = 0;
       for (int i = tid; i < vocab; i += SBLOCK) {
         bf16_t h; *reinterpret_cast<unsigned short*>(&h) = s_vals[i];
-        if (__bfloat162float(h) >= mid) ++cnt;
+        if (bf2f(h) * inv_t >= mid) ++cnt;
       }
       __syncthreads();
       int total = (int)block_sum<SNW>((float)cnt, scratch);
@@ -99,7 +100,7 @@ void softmax_sample_kernel(long* __restrict__ out,          // [B]
   float zs = 0.f;
   for (int i = tid; i < vocab; i += SBLOCK) {
     bf16_t h; *reinterpret_cast<unsigned short*>(&h) = s_vals[i];
-    float v = __bfloat162float(h);
+    float v = bf2f(h) * inv_t;
     if (v >= thresh) zs += __expf(v - m);
   }
   __syncthreads();
@@ -115,7 +116,7 @@ void softmax_sample_kernel(long* __restrict__ out,          // [B]
       float ms = 0.f;
       for (int i = tid; i < vocab; i += SBLOCK) {
         bf16_t h; *reinterpret_cast<unsigned short*>(&h) = s_vals[i];
-        float v = __bfloat162float(h);
+        float v = bf2f(h) * inv_t;
         if (v >= mid && v >= thresh) ms += __expf(v - m);
       }
       __syncthreads();
@@ -128,7 +129,7 @@ void softmax_sample_kernel(long* __restrict__ out,          // [B]
     float zs2 = 0.f;
     for (int i = tid; i < vocab; i += SBLOCK) {
       bf16_t h; *reinterpret_cast<unsigned short*>(&h) = s_vals[i];
-      float v = __bfloat162float(h);
+      float v = bf2f(h) * inv_t;
       if (v >= thresh) zs2 += __expf(v - m);
     }
     __syncthreads();
@@ -144,7 +145,7 @@ void softmax_sample_kernel(long* __restrict__ out,          // [B]
   float csum = 0.f;
   for (int i = beg; i < end; ++i) {
     bf16_t h; *reinterpret_cast<unsigned short*>(&h) = s_vals[i];
-    float v = __bfloat162float(h);
+    float v = bf2f(h) * inv_t;
     if (v >= thresh) csum += __expf(v - m);
   }
   // exclusive prefix over the 256 per-thread sums (via LDS, simple scan)
@@ -163,7 +164,7 @@ void softmax_sample_kernel(long* __restrict__ out,          // [B]
     int sel = -1;
     for (int i = beg; i < end; ++i) {
       bf16_t h; *reinterpret_cast<unsigned short*>(&h) = s_vals[i];
-      float v = __bfloat162float(h);
+      float v = bf2f(h) * inv_t;
       if (v >= thresh) {
         acc += __expf(v - m);
         if (acc > r) { sel = i; break; }
@@ -178,7 +179,7 @@ void softmax_sample_kernel(long* __restrict__ out,          // [B]
       // numerical edge (r ~ z): fall back to the last kept token
       for (int i = vocab - 1; i >= 0; --i) {
         bf16_t h; *reinterpret_cast<unsigned short*>(&h) = s_vals[i];
-        if (__bfloat162float(h) >= thresh) { sel = i; break; }
+        if (bf2f(h) * inv_t >= thresh) { sel = i; break; }
       }
     }
     out[b] = sel < 0 ? 0 : sel;
