@@ -31,6 +31,13 @@ def _key(kind: str, namespace: str, name: str) -> tuple:
     return (kind, namespace or "default", name)
 
 
+def _content_equal(a: dict, b: dict) -> bool:
+    sa, sb = copy.deepcopy(a), copy.deepcopy(b)
+    for o in (sa, sb):
+        o.get("metadata", {}).pop("resourceVersion", None)
+    return sa == sb
+
+
 def deep_merge(base: dict, patch: dict) -> dict:
     """Strategic-merge-lite: dicts merge recursively, everything else replaces;
     None deletes a key."""
@@ -48,12 +55,14 @@ def deep_merge(base: dict, patch: dict) -> dict:
 class MemKube:
     def __init__(self):
         self._objs: dict[tuple, dict] = {}
+        self.changes = 0   # bumps on every event; managers use it to detect fixpoint
         self._rv = itertools.count(1)
         self._lock = threading.RLock()
         self._watchers: list[tuple[Optional[str], Callable[[str, dict], None]]] = []
 
     # ------------------------------------------------------------- helpers
     def _notify(self, event: str, obj: dict) -> None:
+        self.changes += 1
         for kind, cb in list(self._watchers):
             if kind is None or kind == obj.get("kind"):
                 cb(event, copy.deepcopy(obj))
@@ -93,13 +102,17 @@ class MemKube:
             return None
 
     def update(self, obj: dict) -> dict:
-        """Full replace; bumps generation when spec changes."""
+        """Full replace; bumps generation when spec changes. A content-
+        identical update (ignoring resourceVersion) is a no-op — no RV bump,
+        no event — so reconcile→apply cycles settle."""
         with self._lock:
             md = obj["metadata"]
             k = _key(obj["kind"], md.get("namespace", "default"), md["name"])
             if k not in self._objs:
                 raise NotFound(str(k))
             old = self._objs[k]
+            if _content_equal(old, obj):
+                return copy.deepcopy(old)
             new = copy.deepcopy(obj)
             new["metadata"]["resourceVersion"] = str(next(self._rv))
             new["metadata"]["uid"] = old["metadata"]["uid"]
