@@ -231,11 +231,18 @@ class CausalLM(nn.Module):
 
     @torch.no_grad()
     def init_random(self, seed: int = 0) -> None:
+        """Initialize EVERY parameter (models are built on meta + to_empty,
+        so nothing may be left untouched): matrices ~ N(0, 0.02), norm
+        weights = 1, biases = 0."""
         dev = next(self.parameters()).device
         g = torch.Generator(device=dev).manual_seed(seed)
-        for p in self.parameters():
+        for name, p in self.named_parameters():
             if p.dim() >= 2:
                 p.normal_(0.0, 0.02, generator=g)
+            elif name.endswith("bias"):
+                p.zero_()
+            else:  # norm weights
+                p.fill_(1.0)
 
     # -- decode --------------------------------------------------------------
     def forward_decode(self, tokens: torch.Tensor, positions: torch.Tensor,

@@ -1,0 +1,189 @@
+"""Trainer — replaces the reference's external
+`substratusai/model-trainer-huggingface` image.
+
+Container contract (reference docs/container-contract.md + examples/
+llama2-7b/finetuned-model.yaml:10-21): base model RO at /content/model,
+dataset RO at /content/data, artifacts RW at /content/artifacts; params
+(num_train_epochs, save_steps, prompt_template, lr, ...) arrive via
+/content/params.json / PARAM_* env. Writes checkpoint-{step}/ directories
+into artifacts and resumes from the newest on restart; writes completed.json
+when done (docs/design.md "Reconcile Logic").
+
+Compute path: dense differentiable forward (torch autograd + SDPA on ROCm),
+loss + dlogits from the fused cross-entropy HIP kernel (backward starts from
+dlogits — no autograd through the loss), first-party DDP gradient buckets
+over RCCL, fused AdamW HIP kernel for the update.
+"""
+
+from __future__ import annotations
+
+import json
+import os
+import time
+from dataclasses import dataclass, field
+from typing import Optional
+
+import torch
+
+from .. import ops
+from ..models.config import get_config
+from ..models.model import CausalLM
+from ..parallel.comm import barrier, state
+from .ddp import GradBuckets
+
+
+@dataclass
+class TrainConfig:
+    model: str = "falcon-7b-instruct"
+    lr: float = 2e-5
+    beta1: float = 0.9
+    beta2: float = 0.95
+    eps: float = 1e-8
+    weight_decay: float = 0.0
+    micro_batch: int = 4
+    seq_len: int = 2048
+    grad_accum: int = 1
+    num_train_epochs: int = 1
+    max_steps: int = 0          # 0 = derive from epochs × data
+    save_steps: int = 0         # 0 = only final checkpoint
+    artifacts_dir: str = "/content/artifacts"
+    resume: bool = True
+    seed: int = 0
+    dtype: torch.dtype = torch.bfloat16
+    grad_clip: float = 1.0
+
+
+class AdamWState:
+    """fp32 m/v per parameter, driven through the fused HIP kernel."""
+
+    def __init__(self, params: list[torch.Tensor]):
+        self.params = params
+        self.m = [torch.zeros(p.shape, dtype=torch.float32, device=p.device)
+                  for p in params]
+        self.v = [torch.zeros_like(m) for m in self.m]
+        self.step_num = 0
+
+    def step(self, cfg: TrainConfig) -> None:
+        self.step_num += 1
+        for p, m, v in zip(self.params, self.m, self.v):
+            if p.grad is None:
+                continue
+            ops.adamw_step(p.data, p.grad, m, v, cfg.lr, cfg.beta1, cfg.beta2,
+                           cfg.eps, cfg.weight_decay, self.step_num)
+
+    def zero_grad(self) -> None:
+        for p in self.params:
+            p.grad = None
+
+    def state_dict(self) -> dict:
+        return {"step": self.step_num,
+                "m": [t.cpu() for t in self.m],
+                "v": [t.cpu() for t in self.v]}
+
+    def load_state_dict(self, sd: dict, device) -> None:
+        self.step_num = sd["step"]
+        for dst, src in zip(self.m, sd["m"]):
+            dst.copy_(src.to(device))
+        for dst, src in zip(self.v, sd["v"]):
+            dst.copy_(src.to(device))
+
+
+class Trainer:
+    def __init__(self, cfg: TrainConfig, device: Optional[str] = None,
+                 model: Optional[CausalLM] = None):
+        self.cfg = cfg
+        self.st = state()
+        if device is None:
+            device = (f"cuda:{self.st.local_rank}"
+                      if torch.cuda.is_available() else "cpu")
+        self.device = torch.device(device)
+        self.mcfg = get_config(cfg.model)
+        if model is None:
+            model = CausalLM(self.mcfg, device="meta", dtype=cfg.dtype)
+            model = model.to_empty(device=self.device)
+            model.init_random(cfg.seed)  # all DP ranks: same seed = same init
+        self.model = model.train()
+        params = [p for p in self.model.parameters()]
+        for p in params:
+            p.requires_grad_(True)
+        self.opt = AdamWState(params)
+        self.buckets = GradBuckets(params)
+        self.global_step = 0
+        self.tokens_seen = 0
+
+    # ------------------------------------------------------------- one step
+    def train_step(self, tokens: torch.Tensor) -> float:
+        """tokens: [micro_batch, seq_len+1] int64 (inputs ‖ shifted targets).
+        Returns the scalar loss."""
+        inputs = tokens[:, :-1].to(self.device)
+        targets = tokens[:, 1:].to(self.device).reshape(-1)
+        logits = self.model.forward_train(inputs)
+        flat = logits.reshape(-1, logits.shape[-1])
+        loss, dlogits = ops.cross_entropy_fwd_bwd(flat.detach(), targets)
+        flat.backward(gradient=dlogits)
+        self.buckets.finalize()
+        if self.cfg.grad_clip > 0:
+            torch.nn.utils.clip_grad_norm_(self.opt.params, self.cfg.grad_clip)
+        self.opt.step(self.cfg)
+        self.opt.zero_grad()
+        self.global_step += 1
+        self.tokens_seen += inputs.numel() * max(self.st.dp_size, 1)
+        return float(loss)
+
+    # ---------------------------------------------------------- checkpoints
+    def save_checkpoint(self, step: Optional[int] = None) -> str:
+        step = step if step is not None else self.global_step
+        out = os.path.join(self.cfg.artifacts_dir, f"checkpoint-{step}")
+        if self.st.rank == 0:
+            os.makedirs(out, exist_ok=True)
+            from safetensors.torch import save_file
+            tensors = {k: v.detach().cpu().contiguous()
+                       for k, v in self.model.state_dict().items()}
+            save_file(tensors, os.path.join(out, "model.safetensors"))
+            torch.save(self.opt.state_dict(),
+                       os.path.join(out, "optimizer.pt"))
+            with open(os.path.join(out, "trainer_state.json"), "w") as f:
+                json.dump({"global_step": step,
+                           "tokens_seen": self.tokens_seen,
+                           "model": self.cfg.model}, f)
+        barrier()
+        return out
+
+    def maybe_resume(self) -> int:
+        """Resume from the newest checkpoint-{N} under artifacts_dir
+        (reference behavior: training-level resume is the trainer's job —
+        SURVEY.md §5 'Checkpoint / resume')."""
+        if not self.cfg.resume or not os.path.isdir(self.cfg.artifacts_dir):
+            return 0
+        ckpts = []
+        for d in os.listdir(self.cfg.artifacts_dir):
+            if d.startswith("checkpoint-"):
+                try:
+                    ckpts.append((int(d.split("-")[1]), d))
+                except ValueError:
+                    pass
+        if not ckpts:
+            return 0
+        _, newest = max(ckpts)
+        path = os.path.join(self.cfg.artifacts_dir, newest)
+        from safetensors.torch import load_file
+        sd = load_file(os.path.join(path, "model.safetensors"))
+        self.model.load_state_dict(
+            {k: v.to(self.device) for k, v in sd.items()})
+        opt_path = os.path.join(path, "optimizer.pt")
+        if os.path.exists(opt_path):
+            self.opt.load_state_dict(torch.load(opt_path), self.device)
+        with open(os.path.join(path, "trainer_state.json")) as f:
+            st = json.load(f)
+        self.global_step = st["global_step"]
+        self.tokens_seen = st.get("tokens_seen", 0)
+        return self.global_step
+
+    def finish(self) -> None:
+        self.save_checkpoint()
+        if self.st.rank == 0:
+            with open(os.path.join(self.cfg.artifacts_dir,
+                                   "completed.json"), "w") as f:
+                json.dump({"completed": True, "global_step": self.global_step,
+                           "ts": time.time()}, f)
+        barrier()

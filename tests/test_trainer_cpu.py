@@ -1,0 +1,156 @@
+"""Trainer CPU tests: fused-CE gradient path vs torch autograd, loss descent,
+checkpoint/resume, and 2-process DDP gradient parity over gloo
+(multi-GPU paths are covered by multi-process CPU tests)."""
+
+import json
+import os
+
+import pytest
+import torch
+
+from substratus_amd.models import CausalLM, get_config
+from substratus_amd.train.data import JsonlDataset, SyntheticDataset
+from substratus_amd.train.trainer import TrainConfig, Trainer
+
+
+def make_trainer(tmp_path, model="llama-tiny", mb=2, seq=32, **kw):
+    cfg = TrainConfig(model=model, micro_batch=mb, seq_len=seq,
+                      dtype=torch.float32,
+                      artifacts_dir=str(tmp_path / "artifacts"),
+                      lr=1e-3, grad_clip=1.0, **kw)
+    return Trainer(cfg, device="cpu")
+
+
+def test_fused_ce_path_matches_autograd(tmp_path):
+    """Gradients via (fused CE dlogits → logits.backward) must equal plain
+    torch cross_entropy autograd."""
+    torch.manual_seed(0)
+    t1 = make_trainer(tmp_path)
+    t2 = make_trainer(tmp_path)
+    t2.model.load_state_dict(t1.model.state_dict())
+    tokens = torch.randint(0, 512, (2, 33))
+    # path 1: trainer's fused path (but don't step the optimizer)
+    logits = t1.model.forward_train(tokens[:, :-1])
+    flat = logits.reshape(-1, 512)
+    from substratus_amd import ops
+    loss1, dlogits = ops.cross_entropy_fwd_bwd(flat.detach(),
+                                               tokens[:, 1:].reshape(-1))
+    flat.backward(gradient=dlogits)
+    # path 2: torch autograd
+    logits2 = t2.model.forward_train(tokens[:, :-1])
+    loss2 = torch.nn.functional.cross_entropy(
+        logits2.reshape(-1, 512).float(), tokens[:, 1:].reshape(-1))
+    loss2.backward()
+    assert abs(float(loss1) - float(loss2)) < 1e-4
+    for p1, p2 in zip(t1.model.parameters(), t2.model.parameters()):
+        torch.testing.assert_close(p1.grad, p2.grad, atol=1e-5, rtol=1e-4)
+
+
+def test_loss_decreases(tmp_path):
+    torch.manual_seed(1)
+    tr = make_trainer(tmp_path)
+    data = SyntheticDataset(512, 32, seed=3)
+    it = data.batches(2)
+    fixed = next(it)  # overfit one batch
+    losses = [tr.train_step(fixed) for _ in range(15)]
+    assert losses[-1] < losses[0] * 0.8, losses
+
+
+def test_checkpoint_resume(tmp_path):
+    torch.manual_seed(2)
+    tr = make_trainer(tmp_path)
+    data = SyntheticDataset(512, 32, seed=4)
+    it = data.batches(2)
+    for _ in range(3):
+        tr.train_step(next(it))
+    tr.save_checkpoint()
+    tr.finish()
+    assert os.path.exists(tmp_path / "artifacts" / "completed.json")
+    tr2 = make_trainer(tmp_path)
+    resumed = tr2.maybe_resume()
+    assert resumed == 3
+    for p1, p2 in zip(tr.model.parameters(), tr2.model.parameters()):
+        torch.testing.assert_close(p1.data, p2.data)
+    assert tr2.opt.step_num == tr.opt.step_num
+
+
+def test_jsonl_dataset(tmp_path):
+    path = tmp_path / "data"
+    path.mkdir()
+    with open(path / "train.jsonl", "w") as f:
+        for i in range(50):
+            f.write(json.dumps({"prompt": f"q{i} " * 10,
+                                "completion": f"a{i} " * 10}) + "\n")
+    ds = JsonlDataset(str(path), vocab_size=512, seq_len=32)
+    assert len(ds) == 50
+    batches = list(ds.batches(2))
+    assert batches and batches[0].shape == (2, 33)
+    assert batches[0].max() < 512
+    # DP sharding partitions the data
+    b0 = list(ds.batches(2, dp_rank=0, dp_size=2))
+    b1 = list(ds.batches(2, dp_rank=1, dp_size=2))
+    assert len(b0) + len(b1) <= len(batches) + 1
+
+
+# ------------------------------------------------------------------ DDP gloo
+def _ddp_worker(rank, world, tmpdir, q):
+    import torch.distributed as dist
+    os.environ.update({"RANK": str(rank), "WORLD_SIZE": str(world),
+                       "LOCAL_RANK": str(rank),
+                       "MASTER_ADDR": "127.0.0.1", "MASTER_PORT": "29761"})
+    from substratus_amd.parallel import comm
+    comm._STATE = comm.ParallelState()  # reset in forked proc
+    comm.init_distributed(tp_size=1, backend="gloo")
+    torch.manual_seed(0)
+    cfg = TrainConfig(model="llama-tiny", micro_batch=1, seq_len=16,
+                      dtype=torch.float32, artifacts_dir=tmpdir, lr=1e-3)
+    tr = Trainer(cfg, device="cpu")
+    # global batch of 2 rows, rank r takes row r
+    g = torch.Generator().manual_seed(42)
+    batch = torch.randint(0, 512, (2, 17), generator=g)
+    tr.train_step(batch[rank:rank + 1])
+    # serialize to bytes: fd-passing tensors through a Queue races with
+    # worker exit (resource_sharer gone before the parent detaches)
+    import io
+    buf = io.BytesIO()
+    torch.save({k: v.detach().clone()
+                for k, v in tr.model.state_dict().items()}, buf)
+    q.put((rank, buf.getvalue()))
+    dist.destroy_process_group()
+
+
+def test_ddp_two_ranks_match_single(tmp_path):
+    """2-rank DDP step (gloo) must produce the same params as a single
+    process stepping on the combined batch."""
+    import torch.multiprocessing as mp
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    procs = [ctx.Process(target=_ddp_worker, args=(r, 2, str(tmp_path), q))
+             for r in range(2)]
+    for p in procs:
+        p.start()
+    import io
+    results = {}
+    for _ in range(2):
+        rank, blob = q.get(timeout=120)
+        assert not isinstance(blob, str), blob
+        results[rank] = torch.load(io.BytesIO(blob))
+    for p in procs:
+        p.join(timeout=60)
+    # single-process reference on the combined batch
+    torch.manual_seed(0)
+    cfg = TrainConfig(model="llama-tiny", micro_batch=2, seq_len=16,
+                      dtype=torch.float32, artifacts_dir=str(tmp_path),
+                      lr=1e-3)
+    tr = Trainer(cfg, device="cpu")
+    g = torch.Generator().manual_seed(42)
+    batch = torch.randint(0, 512, (2, 17), generator=g)
+    tr.train_step(batch)
+    ref = tr.model.state_dict()
+    # ranks agree with each other and with the single-process reference
+    for k in ref:
+        torch.testing.assert_close(results[0][k], results[1][k])
+        # fp accumulation order differs (per-rank mean + allreduce-average
+        # vs single-process batch mean): tolerance covers that, not a bug
+        torch.testing.assert_close(results[0][k], ref[k], atol=1e-4,
+                                   rtol=5e-3)
