@@ -1,0 +1,95 @@
+"""Model loader — replaces the reference's external
+`substratusai/model-loader-huggingface` image.
+
+Contract (reference examples/llama2-7b/base-model.yaml:8-13,
+examples/llama2-13b-chat-gguf/base-model.yaml:7-9): reads PARAM_NAME (model
+id), optional PARAM_FILES filter; HUGGING_FACE_HUB_TOKEN via secret-ref env;
+writes weights into /content/artifacts/ and a completed.json sentinel; skips
+when completed.json already exists (docs/design.md "Reconcile Logic").
+
+Offline path (this environment has no egress): PARAM_NAME naming a known
+architecture produces deterministic random-init safetensors of that
+architecture. Online path: huggingface_hub snapshot_download when the hub is
+reachable. GGUF files are passed through for the quantized serve path.
+"""
+
+from __future__ import annotations
+
+import json
+import os
+import sys
+
+import torch
+
+
+def load_params(path: str = "/content/params.json") -> dict:
+    params = {}
+    if os.path.exists(path):
+        with open(path) as f:
+            params.update(json.load(f))
+    for k, v in os.environ.items():
+        if k.startswith("PARAM_"):
+            params[k[len("PARAM_"):].lower()] = v
+    return params
+
+
+def run(artifacts_dir: str = "/content/artifacts",
+        params: dict | None = None) -> str:
+    params = params if params is not None else load_params()
+    os.makedirs(artifacts_dir, exist_ok=True)
+    sentinel = os.path.join(artifacts_dir, "completed.json")
+    if os.path.exists(sentinel):
+        print("completed.json present; skipping (resume semantics)")
+        return artifacts_dir
+    name = params.get("name")
+    if not name:
+        raise SystemExit("PARAM_NAME is required (model id)")
+    files_filter = params.get("files", "")
+
+    if _try_hub_download(name, files_filter, artifacts_dir):
+        pass
+    else:
+        _write_random_init(name, artifacts_dir, params)
+
+    with open(sentinel, "w") as f:
+        json.dump({"completed": True, "name": name}, f)
+    print(f"loaded {name} into {artifacts_dir}")
+    return artifacts_dir
+
+
+def _try_hub_download(name: str, files_filter: str, out: str) -> bool:
+    if "/" not in name:
+        return False  # config names like llama2-7b are always offline
+    try:
+        from huggingface_hub import snapshot_download
+        kwargs = {"repo_id": name, "local_dir": out}
+        if files_filter:
+            kwargs["allow_patterns"] = files_filter.split(",")
+        snapshot_download(**kwargs)
+        return True
+    except Exception as e:  # offline / no token
+        print(f"hub download unavailable ({type(e).__name__}); "
+              f"falling back to random-init weights")
+        return False
+
+
+def _write_random_init(name: str, out: str, params: dict) -> None:
+    from safetensors.torch import save_file
+    from ..models.config import get_config
+    from ..models.model import CausalLM
+    cfg = get_config(name)
+    dtype = torch.bfloat16 if params.get("dtype", "bf16") == "bf16" \
+        else torch.float32
+    model = CausalLM(cfg, device="meta", dtype=dtype)
+    model = model.to_empty(device="cpu")
+    model.init_random(seed=int(params.get("seed", 0)))
+    tensors = {k: v.detach().contiguous()
+               for k, v in model.state_dict().items()}
+    save_file(tensors, os.path.join(out, "model.safetensors"))
+    with open(os.path.join(out, "config.json"), "w") as f:
+        json.dump({"model": cfg.name, "dtype": str(dtype).split(".")[-1],
+                   "source": "random-init"}, f)
+
+
+if __name__ == "__main__":
+    run(sys.argv[1] if len(sys.argv) > 1 else "/content/artifacts")

@@ -1,0 +1,276 @@
+"""HTTP inference server — replaces the reference's external
+`substratusai/model-server-basaran` image.
+
+Container contract (reference docs/container-contract.md:50-55 +
+test/system.sh:73-77): listens on 8080; readiness = GET / → 200; OpenAI-style
+POST /v1/completions with {prompt, max_tokens, temperature, top_p, top_k,
+stream}. Loads the model from /content/model (loader artifacts: config.json +
+model.safetensors) or from PARAM_NAME with random-init weights.
+
+Observability: /metrics (Prometheus) exposes served tokens,
+request latency p50/p95, and per-GPU HBM usage — the metrics the reference's
+control plane never had from its external images.
+
+Engine loop runs in a background thread; requests enqueue into the continuous
+batcher and stream tokens back through per-request queues.
+"""
+
+from __future__ import annotations
+
+import json
+import os
+import queue
+import threading
+import time
+from typing import Optional
+
+import torch
+from fastapi import FastAPI, Request
+from fastapi.responses import JSONResponse, StreamingResponse
+
+from ..models.config import get_config
+from . import tokenizer as tok
+from .engine import Engine, EngineConfig, StepOutput
+
+
+def load_params() -> dict:
+    params = {}
+    path = os.environ.get("PARAMS_PATH", "/content/params.json")
+    if os.path.exists(path):
+        with open(path) as f:
+            params.update(json.load(f))
+    for k, v in os.environ.items():
+        if k.startswith("PARAM_"):
+            params[k[len("PARAM_"):].lower()] = v
+    return params
+
+
+def resolve_model_name(params: dict, model_dir: str = "/content/model") -> str:
+    cfg_path = os.path.join(model_dir, "config.json")
+    if os.path.exists(cfg_path):
+        with open(cfg_path) as f:
+            return json.load(f)["model"]
+    return params.get("name", params.get("model", "llama2-7b"))
+
+
+class InferenceWorker:
+    """Owns the engine; steps continuously while there is work."""
+
+    def __init__(self, engine: Engine):
+        self.engine = engine
+        self.inbox: queue.Queue = queue.Queue()
+        self.outputs: dict[int, queue.Queue] = {}
+        self.lock = threading.Lock()
+        self.total_tokens = 0
+        self.latencies: list[float] = []
+        self._stop = False
+        self.thread = threading.Thread(target=self._run, daemon=True)
+        self.thread.start()
+
+    def submit(self, prompt_tokens, max_new_tokens, temperature, top_k,
+               top_p, eos_token=-1) -> tuple[int, queue.Queue]:
+        out_q: queue.Queue = queue.Queue()
+        done = threading.Event()
+        holder = {}
+        self.inbox.put((prompt_tokens, max_new_tokens, temperature, top_k,
+                        top_p, eos_token, out_q, holder, done))
+        done.wait()
+        return holder["sid"], out_q
+
+    def _run(self) -> None:
+        while not self._stop:
+            moved = False
+            while True:
+                try:
+                    (pt, mnt, temp, tk, tp, eos, out_q, holder,
+                     done) = self.inbox.get_nowait()
+                except queue.Empty:
+                    break
+                sid = self.engine.add_request(pt, max_new_tokens=mnt,
+                                              temperature=temp, top_k=tk,
+                                              top_p=tp, eos_token=eos)
+                with self.lock:
+                    self.outputs[sid] = out_q
+                holder["sid"] = sid
+                done.set()
+                moved = True
+            if self.engine.has_work():
+                t0 = time.time()
+                outs = self.engine.step()
+                dt = time.time() - t0
+                with self.lock:
+                    self.latencies.append(dt)
+                    if len(self.latencies) > 1000:
+                        self.latencies = self.latencies[-1000:]
+                    for o in outs:
+                        self.total_tokens += 1
+                        q_ = self.outputs.get(o.seq_id)
+                        if q_ is not None:
+                            q_.put(o)
+                            if o.finished:
+                                self.outputs.pop(o.seq_id, None)
+                moved = True
+            if not moved:
+                time.sleep(0.002)
+
+    def stop(self) -> None:
+        self._stop = True
+
+
+def create_app(engine: Optional[Engine] = None,
+               model_name: Optional[str] = None) -> FastAPI:
+    params = load_params()
+    if engine is None:
+        name = model_name or resolve_model_name(params)
+        on_gpu = torch.cuda.is_available()
+        ecfg = EngineConfig(
+            model=name,
+            dtype=torch.bfloat16 if on_gpu else torch.float32,
+            max_batch=int(params.get("max_batch", 64)),
+            max_seq_len=int(params.get("max_length", 2048)),
+        )
+        engine = Engine(ecfg)
+        _maybe_load_weights(engine)
+    worker = InferenceWorker(engine)
+    app = FastAPI(title="substratus-amd model server")
+    app.state.worker = worker
+    app.state.engine = engine
+    vocab = engine.mcfg.vocab_size
+    model_id = engine.cfg.model
+
+    try:
+        from prometheus_client import (CONTENT_TYPE_LATEST, Gauge, Counter,
+                                       generate_latest)
+        tokens_counter = Counter("served_tokens_total",
+                                 "tokens generated")
+        hbm_gauge = Gauge("hbm_allocated_bytes", "GPU memory allocated")
+        have_prom = True
+    except ImportError:  # pragma: no cover
+        have_prom = False
+
+    @app.get("/")
+    def root():
+        # readiness contract: GET / -> 200
+        return {"status": "ok", "model": model_id}
+
+    @app.get("/v1/models")
+    def models():
+        return {"object": "list",
+                "data": [{"id": model_id, "object": "model"}]}
+
+    @app.get("/stats")
+    def stats():
+        lat = sorted(worker.latencies) or [0.0]
+        return {
+            "served_tokens_total": worker.total_tokens,
+            "p50_step_latency_ms": lat[len(lat) // 2] * 1000,
+            "p95_step_latency_ms": lat[int(len(lat) * 0.95)] * 1000,
+            "running": len(engine.running),
+            "waiting": len(engine.waiting),
+            "kv_blocks_free": engine.kv.allocator.free_count,
+            "hbm_allocated_bytes": (torch.cuda.memory_allocated()
+                                    if torch.cuda.is_available() else 0),
+        }
+
+    if have_prom:
+        @app.get("/metrics")
+        def metrics():
+            from fastapi import Response
+            tokens_counter._value.set(worker.total_tokens)  # type: ignore
+            if torch.cuda.is_available():
+                hbm_gauge.set(torch.cuda.memory_allocated())
+            return Response(generate_latest(), media_type=CONTENT_TYPE_LATEST)
+
+    @app.post("/v1/completions")
+    async def completions(req: Request):
+        body = await req.json()
+        prompt = body.get("prompt", "")
+        if isinstance(prompt, list):
+            prompt_tokens = [int(t) for t in prompt]
+        else:
+            prompt_tokens = tok.encode(str(prompt), vocab)
+        max_tokens = int(body.get("max_tokens", 16))
+        temperature = float(body.get("temperature", 1.0))
+        top_p = float(body.get("top_p", 1.0))
+        top_k = int(body.get("top_k", 0))
+        stream = bool(body.get("stream", False))
+        t_start = time.time()
+        sid, out_q = worker.submit(prompt_tokens, max_tokens, temperature,
+                                   top_k, top_p, eos_token=tok.EOS)
+
+        def collect() -> list[int]:
+            toks = []
+            while True:
+                o: StepOutput = out_q.get(timeout=600)
+                toks.append(o.token)
+                if o.finished:
+                    return toks
+
+        if stream:
+            def gen():
+                idx = 0
+                while True:
+                    o: StepOutput = out_q.get(timeout=600)
+                    chunk = {
+                        "id": f"cmpl-{sid}", "object": "text_completion",
+                        "model": model_id,
+                        "choices": [{"index": 0,
+                                     "text": tok.decode([o.token]),
+                                     "finish_reason": "stop" if o.finished
+                                     else None}],
+                    }
+                    yield f"data: {json.dumps(chunk)}\n\n"
+                    idx += 1
+                    if o.finished:
+                        yield "data: [DONE]\n\n"
+                        return
+            return StreamingResponse(gen(), media_type="text/event-stream")
+
+        toks = collect()
+        return JSONResponse({
+            "id": f"cmpl-{sid}",
+            "object": "text_completion",
+            "created": int(t_start),
+            "model": model_id,
+            "choices": [{
+                "index": 0,
+                "text": tok.decode(toks),
+                "finish_reason": "length" if len(toks) >= max_tokens
+                                 else "stop",
+            }],
+            "usage": {
+                "prompt_tokens": len(prompt_tokens),
+                "completion_tokens": len(toks),
+                "total_tokens": len(prompt_tokens) + len(toks),
+            },
+        })
+
+    return app
+
+
+def _maybe_load_weights(engine: Engine,
+                        model_dir: str = "/content/model") -> None:
+    path = os.path.join(model_dir, "model.safetensors")
+    if not os.path.exists(path):
+        # search newest checkpoint (trainer artifacts layout)
+        if os.path.isdir(model_dir):
+            ckpts = sorted((d for d in os.listdir(model_dir)
+                            if d.startswith("checkpoint-")),
+                           key=lambda d: int(d.split("-")[1]))
+            if ckpts:
+                path = os.path.join(model_dir, ckpts[-1], "model.safetensors")
+    if os.path.exists(path):
+        from safetensors.torch import load_file
+        sd = load_file(path)
+        engine.model.load_state_dict(
+            {k: v.to(engine.device, engine.cfg.dtype) for k, v in sd.items()})
+
+
+def main() -> None:
+    import uvicorn
+    app = create_app()
+    uvicorn.run(app, host="0.0.0.0", port=int(os.environ.get("PORT", 8080)))
+
+
+if __name__ == "__main__":
+    main()
