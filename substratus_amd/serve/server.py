@@ -45,7 +45,8 @@ def load_params() -> dict:
     return params
 
 
-def resolve_model_name(params: dict, model_dir: str = "/content/model") -> str:
+def resolve_model_name(params: dict, model_dir: str | None = None) -> str:
+    model_dir = model_dir or os.environ.get("MODEL_DIR", "/content/model")
     cfg_path = os.path.join(model_dir, "config.json")
     if os.path.exists(cfg_path):
         with open(cfg_path) as f:
@@ -249,7 +250,8 @@ def create_app(engine: Optional[Engine] = None,
 
 
 def _maybe_load_weights(engine: Engine,
-                        model_dir: str = "/content/model") -> None:
+                        model_dir: str | None = None) -> None:
+    model_dir = model_dir or os.environ.get("MODEL_DIR", "/content/model")
     path = os.path.join(model_dir, "model.safetensors")
     if not os.path.exists(path):
         # search newest checkpoint (trainer artifacts layout)

@@ -1,0 +1,77 @@
+"""System test — the reference's test/system.sh end-to-end on CPU with no
+cluster: apply a base Model + Server, drive the controllers + local runtime
+until .status.ready, then POST /v1/completions and get text back
+(reference test/system.sh:28-77; facebook-opt-125m stands in as llama-tiny
+for CPU speed — the reference used opt-125m for the same reason)."""
+
+import json
+import os
+import time
+
+import httpx
+import pytest
+
+from substratus_amd.api import types as apiv1
+from substratus_amd.cloud import new_cloud
+from substratus_amd.controller.manager import Manager
+from substratus_amd.kube.memapi import MemKube
+from substratus_amd.runtime.local import LocalRuntime
+from substratus_amd.sci.base import FakeSCIClient
+
+
+@pytest.mark.timeout(300)
+def test_system_model_server_completion(tmp_path):
+    bucket = str(tmp_path / "bucket")
+    os.makedirs(bucket)
+    kube = MemKube()
+    cloud = new_cloud({"CLOUD": "kind", "CLUSTER_NAME": "sys",
+                       "BUCKET_DIR": bucket})
+    mgr = Manager(kube, cloud, FakeSCIClient())
+    rt = LocalRuntime(kube, bucket)
+
+    # kubectl apply -f base-model.yaml (reference examples/*/base-model.yaml)
+    kube.create(apiv1.new_object("Model", "model-tiny", {
+        "image": "substratus-amd/model-loader",
+        "params": {"name": "llama-tiny"},
+    }))
+    # kubectl apply -f server.yaml
+    kube.create(apiv1.new_object("Server", "server-tiny", {
+        "image": "substratus-amd/model-server",
+        "model": {"name": "model-tiny"},
+        "params": {"max_length": 128, "max_batch": 2},
+    }))
+
+    # kubectl wait .status.ready=true (720s budget in reference)
+    deadline = time.time() + 240
+    while time.time() < deadline:
+        rt.run_until_quiet(mgr, timeout=10)
+        model = kube.get("Model", "model-tiny")
+        server = kube.get("Server", "server-tiny")
+        if apiv1.get_status_ready(model) and apiv1.get_status_ready(server):
+            break
+        time.sleep(1)
+    try:
+        assert apiv1.get_status_ready(kube.get("Model", "model-tiny")), \
+            kube.get("Model", "model-tiny").get("status")
+        assert apiv1.get_status_ready(kube.get("Server", "server-tiny")), \
+            kube.get("Server", "server-tiny").get("status")
+
+        # loader wrote artifacts into the deterministic bucket path
+        url = cloud.object_artifact_url(kube.get("Model", "model-tiny"))
+        digest = url.rsplit("/", 1)[-1]
+        art = os.path.join(bucket, digest, "artifacts")
+        assert os.path.exists(os.path.join(art, "model.safetensors"))
+        assert os.path.exists(os.path.join(art, "completed.json"))
+
+        # the system test's completion request (test/system.sh:73-77)
+        svc = kube.get("Service", "server-tiny-server")
+        endpoint = svc["metadata"]["annotations"][
+            "substratus.ai/local-endpoint"]
+        r = httpx.post(f"{endpoint}/v1/completions", json={
+            "prompt": "Who was the first president of the United States? ",
+            "max_tokens": 8}, timeout=120)
+        assert r.status_code == 200
+        body = r.json()
+        assert body["usage"]["completion_tokens"] >= 1
+    finally:
+        rt.stop()
