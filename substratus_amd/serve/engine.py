@@ -40,6 +40,7 @@ class EngineConfig:
     kv_blocks: Optional[int] = None        # None = size from gpu_mem_fraction
     gpu_mem_fraction: float = 0.85         # of free HBM after weights
     prefill_chunk: int = 8192              # max packed prefill tokens per step
+    use_graphs: bool = True                # hipGraph-captured decode on GPU
     seed: int = 0
 
 
@@ -87,6 +88,7 @@ class Engine:
         self.waiting: list[SequenceState] = []
         self.running: list[SequenceState] = []
         self._gen = torch.Generator(device="cpu").manual_seed(cfg.seed)
+        self.graph_runner = None
         self.steps = 0
         self.tokens_generated = 0
 
@@ -153,6 +155,43 @@ class Engine:
 
     # ---------------------------------------------------------------- decode
     def _decode(self) -> list[StepOutput]:
+        if self.cfg.use_graphs and self.device.type == "cuda":
+            return self._decode_graph()
+        return self._decode_eager()
+
+    def _decode_graph(self) -> list[StepOutput]:
+        """hipGraph-captured decode (see serve/graph_runner.py): the token/
+        position/slot state lives on device; the host only grows block
+        tables and reads sampled tokens back. Sampling uses the device RNG
+        (not the engine's CPU generator)."""
+        from .kv_cache import OutOfBlocks
+        from .graph_runner import GraphDecodeRunner
+        if self.graph_runner is None:
+            self.graph_runner = GraphDecodeRunner(self)
+        runner = self.graph_runner
+        ids = [s.seq_id for s in self.running]
+        try:
+            if [s.seq_id for s in runner.active] != ids:
+                runner.prime(self.running)
+            toks = runner.step()
+        except OutOfBlocks:
+            runner.invalidate()
+            self._preempt_youngest()
+            return self._decode() if self.running else []
+        out = []
+        for seq, tok in zip(self.running, toks):
+            self.tokens_generated += 1
+            done = (seq.num_generated >= seq.max_new_tokens or
+                    (seq.eos_token >= 0 and tok == seq.eos_token) or
+                    seq.length >= self.cfg.max_seq_len)
+            if done:
+                seq.finished = True
+            out.append(StepOutput(seq.seq_id, int(tok), done))
+        if any(o.finished for o in out):
+            runner.invalidate()
+        return out
+
+    def _decode_eager(self) -> list[StepOutput]:
         seqs = self.running
         bs = self.cfg.block_size
         dev = self.device
