@@ -1,0 +1,89 @@
+"""Kernel microbenchmarks at llama2-7b decode shapes (run on MI355X).
+
+Reports achieved bytes/s against the HBM ceiling (guide: 8 TB/s peak,
+~6.3 TB/s achievable) for the decode hot kernels.
+"""
+
+import argparse
+import time
+
+import torch
+
+from substratus_amd import ops
+
+
+def timeit(fn, iters=200, warmup=20):
+    for _ in range(warmup):
+        fn()
+    torch.cuda.synchronize()
+    t0 = time.time()
+    for _ in range(iters):
+        fn()
+    torch.cuda.synchronize()
+    return (time.time() - t0) / iters
+
+
+def bench_attn(B=64, hq=32, hkv=32, d=128, ctx=512, bs=16):
+    torch.manual_seed(0)
+    nblk_per = (ctx + bs - 1) // bs
+    nb = B * nblk_per + 1
+    kc = torch.randn(nb, hkv, bs, d, device="cuda").bfloat16()
+    vc = torch.randn_like(kc)
+    bt = torch.arange(B * nblk_per, dtype=torch.int32,
+                      device="cuda").reshape(B, nblk_per)
+    q = torch.randn(B, hq, d, device="cuda").bfloat16()
+    lens = torch.full((B,), ctx, dtype=torch.int32, device="cuda")
+    scale = d ** -0.5
+    dt = timeit(lambda: ops.paged_attn_decode(q, kc, vc, bt, lens, scale))
+    kv_bytes = 2 * B * ctx * hkv * d * 2
+    print(f"paged_attn_decode B={B} ctx={ctx} hq={hq} hkv={hkv} d={d}: "
+          f"{dt*1e6:.1f} us, {kv_bytes/dt/1e12:.2f} TB/s")
+
+
+def bench_rope(B=64, hq=32, hkv=32, d=128, bs=16):
+    torch.manual_seed(0)
+    nb = B * 4
+    q = torch.randn(B, hq, d, device="cuda").bfloat16()
+    k = torch.randn(B, hkv, d, device="cuda").bfloat16()
+    v = torch.randn(B, hkv, d, device="cuda").bfloat16()
+    pos = torch.full((B,), 500, dtype=torch.int32, device="cuda")
+    slots = torch.arange(B, dtype=torch.int32, device="cuda")
+    kc = torch.zeros(nb, hkv, bs, d, device="cuda").bfloat16()
+    vc = torch.zeros_like(kc)
+    dt = timeit(lambda: ops.rope_kv_append(q, k, v, pos, kc, vc, slots,
+                                           10000.0))
+    print(f"rope_kv_append B={B}: {dt*1e6:.1f} us")
+
+
+def bench_norm(B=64, h=4096):
+    x = torch.randn(B, h, device="cuda").bfloat16()
+    r = torch.randn(B, h, device="cuda").bfloat16()
+    w = torch.randn(h, device="cuda").bfloat16()
+    dt = timeit(lambda: ops.rms_norm_residual(x, r, w, 1e-5))
+    print(f"rms_norm_residual B={B} h={h}: {dt*1e6:.1f} us")
+
+
+def bench_sample(B=64, V=32000):
+    logits = torch.randn(B, V, device="cuda").bfloat16()
+    temp = torch.zeros(B, device="cuda")
+    tk = torch.zeros(B, dtype=torch.int32, device="cuda")
+    tp = torch.ones(B, device="cuda")
+    u = torch.rand(B, device="cuda")
+    dt = timeit(lambda: ops.softmax_sample(logits, temp, tk, tp, u))
+    print(f"softmax_sample greedy B={B} V={V}: {dt*1e6:.1f} us")
+
+
+if __name__ == "__main__":
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--which", default="all")
+    a = ap.parse_args()
+    if a.which in ("all", "attn"):
+        bench_attn()
+        bench_attn(ctx=2048)
+        bench_attn(hq=64, hkv=8, ctx=512)  # llama2-70b-style GQA
+    if a.which in ("all", "rope"):
+        bench_rope()
+    if a.which in ("all", "norm"):
+        bench_norm()
+    if a.which in ("all", "sample"):
+        bench_sample()
