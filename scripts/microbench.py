@@ -3,6 +3,9 @@
 Reports achieved bytes/s against the HBM ceiling (guide: 8 TB/s peak,
 ~6.3 TB/s achievable) for the decode hot kernels.
 """
+import os
+import sys
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 
 import argparse
 import time

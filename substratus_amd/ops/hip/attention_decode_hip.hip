@@ -4,17 +4,21 @@
 // cache, one new token per sequence.
 //
 // Decode at batch B is HBM-bound on the K/V stream (e.g. llama2-7b @ctx 512,
-// B=64: ~16 GB of KV per step vs 13.5 GB of weights), so the kernel is
-// organized around wide coalesced loads of 4 KiB cache panels:
+// B=64: ~16 GB of KV per step vs 13.5 GB of weights). Structure:
 //
-//   * grid = (num_seqs, n_kv_heads, ceil(group/GC)); 4 waves per workgroup.
-//   * Each wave owns KV blocks stride-4 (flash-decoding style split over the
-//     sequence) keeping ≥ (16 tokens × D/8-lane uint4) loads in flight.
+//   * grid = (num_seqs, n_kv_heads, ceil(group/GC)); 4 waves per workgroup;
+//     each wave owns KV blocks stride-4 (flash-decoding split over the
+//     sequence).
 //   * K panel row [16, D] bf16: D/8 lanes × uint4 = one full 256 B row per
-//     token; scores reduced via shfl within the lane group (no LDS on the
-//     score path; guide §5 common-mistake #6: wave-parallel, never serial).
-//   * Online softmax per wave; P·V with dims distributed over lanes (uint
-//     bf16×2 loads, conflict-free); cross-wave merge via LDS at the end.
+//     token. K tiles are PREFETCHED one block ahead into registers (guide
+//     T14 issue-early/consume-late: the next block's loads are issued before
+//     this block's score+PV compute, hiding the ~900-cycle HBM latency), so
+//     the score phase runs entirely from registers.
+//   * Scores reduced via shfl within the lane group (guide §5
+//     common-mistake #6: wave-parallel softmax, never serial).
+//   * Online softmax per wave; P·V with dims distributed over lanes; for
+//     full blocks all 16 V row-loads are issued before the FMA chain.
+//   * Cross-wave merge through LDS at the end.
 //
 // Cache layout: [num_blocks, n_kv_heads, BS=16, D]; block_table [B, maxb]
 // int32; seq_lens [B] int32 counts INCLUDING the current token.
@@ -44,7 +48,7 @@ void paged_attn_decode_kernel(
   const int gchunk = blockIdx.z;      // q-head chunk within the GQA group
   const int group = hq / hkv;
   const int g0 = gchunk * GC;
-  const int ng = min(GC, group - g0); // live q heads this WG
+  const int ng = min(GC, group - g0);
   const int wave = threadIdx.x / WAVE;
   const int lane = threadIdx.x % WAVE;
   const int sub = lane / LPT;         // token-within-pass
@@ -53,7 +57,7 @@ void paged_attn_decode_kernel(
   const int n = seq_lens[b];
   const int nblk = (n + BS - 1) / BS;
 
-  // ---- load this WG's q fragments: q[b][h*group+g0+g][ll*8..+8] ----------
+  // ---- q fragments ------------------------------------------------------
   float qf[GC][8];
   #pragma unroll
   for (int g = 0; g < GC; ++g) {
@@ -69,7 +73,6 @@ void paged_attn_decode_kernel(
     }
   }
 
-  // online-softmax state per wave
   float m[GC], l[GC], acc[GC][DPL];
   #pragma unroll
   for (int g = 0; g < GC; ++g) {
@@ -83,23 +86,43 @@ void paged_attn_decode_kernel(
   const long head_off = (long)h * BS * D;
   const long blk_stride = (long)hkv * BS * D;
 
+  // ---- K prefetch pipeline: tile bi held in regs while bi+NW streams ----
+  uint4 kt[PASSES];
+  int blk_cur = (wave < nblk) ? bt[wave] : 0;
+  if (wave < nblk) {
+    const bf16_t* kbase = k_cache + (long)blk_cur * blk_stride + head_off;
+    #pragma unroll
+    for (int p = 0; p < PASSES; ++p)
+      kt[p] = reinterpret_cast<const uint4*>(kbase + (p * TPP + sub) * D)[ll];
+  }
+
   for (int bi = wave; bi < nblk; bi += NW) {
-    const int blk = bt[bi];
-    const bf16_t* kbase = k_cache + (long)blk * blk_stride + head_off;
-    // ---- scores for the 16 tokens of this block ------------------------
+    // issue next tile's loads before any compute on the current tile
+    uint4 kt_next[PASSES];
+    const int bnext = bi + NW;
+    int blk_next = 0;
+    if (bnext < nblk) {
+      blk_next = bt[bnext];
+      const bf16_t* kb = k_cache + (long)blk_next * blk_stride + head_off;
+      #pragma unroll
+      for (int p = 0; p < PASSES; ++p)
+        kt_next[p] =
+            reinterpret_cast<const uint4*>(kb + (p * TPP + sub) * D)[ll];
+    }
+    const int blk = blk_cur;
+
+    // ---- scores from registers ----------------------------------------
     #pragma unroll
     for (int p = 0; p < PASSES; ++p) {
       const int t = p * TPP + sub;
       const int tok_idx = bi * BS + t;
-      const uint4 u = reinterpret_cast<const uint4*>(kbase + t * D)[ll];
-      float2 f0 = unpack_bf16x2(u.x), f1 = unpack_bf16x2(u.y);
-      float2 f2 = unpack_bf16x2(u.z), f3 = unpack_bf16x2(u.w);
+      float2 f0 = unpack_bf16x2(kt[p].x), f1 = unpack_bf16x2(kt[p].y);
+      float2 f2 = unpack_bf16x2(kt[p].z), f3 = unpack_bf16x2(kt[p].w);
       #pragma unroll
       for (int g = 0; g < GC; ++g) {
         float s = qf[g][0] * f0.x + qf[g][1] * f0.y + qf[g][2] * f1.x +
                   qf[g][3] * f1.y + qf[g][4] * f2.x + qf[g][5] * f2.y +
                   qf[g][6] * f3.x + qf[g][7] * f3.y;
-        // reduce over the LPT lanes of this token
         #pragma unroll
         for (int off = LPT / 2; off > 0; off >>= 1)
           s += __shfl_xor(s, off, WAVE);
@@ -107,41 +130,78 @@ void paged_attn_decode_kernel(
           s_scores[wave][g][t] = (tok_idx < n) ? s * scale : -INFINITY;
       }
     }
-    // wave-synchronous: LDS writes above are visible to the same wave after
-    // lgkmcnt drain (compiler-inserted); no __syncthreads needed (same wave).
     __builtin_amdgcn_wave_barrier();
 
     // ---- online softmax + P·V -----------------------------------------
-    #pragma unroll
-    for (int g = 0; g < GC; ++g) {
-      float bm = -INFINITY;
+    const bf16_t* vbase = v_cache + (long)blk * blk_stride + head_off;
+    const int valid = min(n - bi * BS, BS);
+    if (valid == BS) {
+      // full block: hoist all 16 V loads ahead of the FMA chains
+      unsigned int vv[BS];
+      bf16_t vs[BS];
       #pragma unroll
-      for (int t = 0; t < BS; ++t) bm = fmaxf(bm, s_scores[wave][g][t]);
-      if (bm == -INFINITY) continue;  // fully masked block (can't happen mid-seq)
-      const float mn = fmaxf(m[g], bm);
-      const float alpha = __expf(m[g] - mn);
-      m[g] = mn;
-      l[g] *= alpha;
+      for (int t = 0; t < BS; ++t) {
+        if (DPL == 2)
+          vv[t] = reinterpret_cast<const unsigned int*>(vbase + t * D)[lane];
+        else
+          vs[t] = vbase[t * D + lane];
+      }
       #pragma unroll
-      for (int j = 0; j < DPL; ++j) acc[g][j] *= alpha;
-      const bf16_t* vbase = v_cache + (long)blk * blk_stride + head_off;
-      const int valid = min(n - bi * BS, BS);  // mask tail of last block
-      for (int t = 0; t < valid; ++t) {
-        const float p = __expf(s_scores[wave][g][t] - mn);
-        l[g] += p;  // same value in every lane; lane 0's copy is merged
-        // lane-owned dims: D=128 → uint load (2 dims/lane);
-        // D=64 → one bf16 scalar per lane.
-        if (DPL == 2) {
-          const unsigned int uv = reinterpret_cast<const unsigned int*>(
-              vbase + t * D)[lane];
-          float2 vf = unpack_bf16x2(uv);
-          acc[g][0] += p * vf.x;
-          acc[g][1] += p * vf.y;
-        } else {
-          acc[g][0] += p * bf2f(vbase[t * D + lane]);
+      for (int g = 0; g < GC; ++g) {
+        float bm = -INFINITY;
+        #pragma unroll
+        for (int t = 0; t < BS; ++t) bm = fmaxf(bm, s_scores[wave][g][t]);
+        const float mn = fmaxf(m[g], bm);
+        const float alpha = (m[g] == -INFINITY) ? 0.f : __expf(m[g] - mn);
+        m[g] = mn;
+        l[g] *= alpha;
+        #pragma unroll
+        for (int j = 0; j < DPL; ++j) acc[g][j] *= alpha;
+        #pragma unroll
+        for (int t = 0; t < BS; ++t) {
+          const float p = __expf(s_scores[wave][g][t] - mn);
+          l[g] += p;
+          if (DPL == 2) {
+            float2 vf = unpack_bf16x2(vv[t]);
+            acc[g][0] += p * vf.x;
+            acc[g][1] += p * vf.y;
+          } else {
+            acc[g][0] += p * bf2f(vs[t]);
+          }
+        }
+      }
+    } else {
+      #pragma unroll
+      for (int g = 0; g < GC; ++g) {
+        float bm = -INFINITY;
+        #pragma unroll
+        for (int t = 0; t < BS; ++t) bm = fmaxf(bm, s_scores[wave][g][t]);
+        if (bm == -INFINITY) continue;
+        const float mn = fmaxf(m[g], bm);
+        const float alpha = (m[g] == -INFINITY) ? 0.f : __expf(m[g] - mn);
+        m[g] = mn;
+        l[g] *= alpha;
+        #pragma unroll
+        for (int j = 0; j < DPL; ++j) acc[g][j] *= alpha;
+        for (int t = 0; t < valid; ++t) {
+          const float p = __expf(s_scores[wave][g][t] - mn);
+          l[g] += p;
+          if (DPL == 2) {
+            const unsigned int uv = reinterpret_cast<const unsigned int*>(
+                vbase + t * D)[lane];
+            float2 vf = unpack_bf16x2(uv);
+            acc[g][0] += p * vf.x;
+            acc[g][1] += p * vf.y;
+          } else {
+            acc[g][0] += p * bf2f(vbase[t * D + lane]);
+          }
         }
       }
     }
+    // rotate the pipeline
+    blk_cur = blk_next;
+    #pragma unroll
+    for (int p = 0; p < PASSES; ++p) kt[p] = kt_next[p];
   }
 
   // ---- cross-wave merge via LDS ----------------------------------------

@@ -121,7 +121,14 @@ class GraphDecodeRunner:
         if b in self.graphs:
             return self.graphs[b]
         torch.cuda.synchronize()
-        # warmup on a side stream (required before capture)
+        # Warmup executes REAL steps on the live state (tokens/pos/lens/slots
+        # advance; KV slots for the next 2 positions get scribbled) — save
+        # and restore the small state. The scribbled KV positions are always
+        # rewritten by rope_kv_append before attention ever reads them (a
+        # position's KV is written in the same step that first includes it),
+        # so the cache needs no restore.
+        saved = [t.clone() for t in (self.tokens, self.positions,
+                                     self.seq_lens, self.slots)]
         s = torch.cuda.Stream()
         s.wait_stream(torch.cuda.current_stream())
         with torch.cuda.stream(s):
@@ -132,6 +139,10 @@ class GraphDecodeRunner:
         g = torch.cuda.CUDAGraph()
         with torch.cuda.graph(g):
             self._step_body(b)
+        for dst, src in zip((self.tokens, self.positions, self.seq_lens,
+                             self.slots), saved):
+            dst.copy_(src)
+        torch.cuda.synchronize()
         self.graphs[b] = g
         return g
 
