@@ -89,19 +89,19 @@ void paged_attn_decode_kernel(
   // K/V stream is read-once per step (each (seq, kv-head) block is owned by
   // one workgroup): non-temporal loads keep it out of L1 (guide nt-weights:
   // -18% issued-to-landed on one-reader streams).
-  uint4 kt[PASSES];
+  uint4v kt[PASSES];
   int blk_cur = (wave < nblk) ? bt[wave] : 0;
   if (wave < nblk) {
     const bf16_t* kbase = k_cache + (long)blk_cur * blk_stride + head_off;
     #pragma unroll
     for (int p = 0; p < PASSES; ++p)
-      kt[p] = __builtin_nontemporal_load(
+      kt[p] = nt_load_u4(
           reinterpret_cast<const uint4*>(kbase + (p * TPP + sub) * D) + ll);
   }
 
   for (int bi = wave; bi < nblk; bi += NW) {
     // issue next tile's loads before any compute on the current tile
-    uint4 kt_next[PASSES];
+    uint4v kt_next[PASSES];
     const int bnext = bi + NW;
     int blk_next = 0;
     if (bnext < nblk) {
@@ -109,7 +109,7 @@ void paged_attn_decode_kernel(
       const bf16_t* kb = k_cache + (long)blk_next * blk_stride + head_off;
       #pragma unroll
       for (int p = 0; p < PASSES; ++p)
-        kt_next[p] = __builtin_nontemporal_load(
+        kt_next[p] = nt_load_u4(
             reinterpret_cast<const uint4*>(kb + (p * TPP + sub) * D) + ll);
     }
     const int blk = blk_cur;
@@ -119,8 +119,8 @@ void paged_attn_decode_kernel(
     for (int p = 0; p < PASSES; ++p) {
       const int t = p * TPP + sub;
       const int tok_idx = bi * BS + t;
-      float2 f0 = unpack_bf16x2(kt[p].x), f1 = unpack_bf16x2(kt[p].y);
-      float2 f2 = unpack_bf16x2(kt[p].z), f3 = unpack_bf16x2(kt[p].w);
+      float2 f0 = unpack_bf16x2(kt[p][0]), f1 = unpack_bf16x2(kt[p][1]);
+      float2 f2 = unpack_bf16x2(kt[p][2]), f3 = unpack_bf16x2(kt[p][3]);
       #pragma unroll
       for (int g = 0; g < GC; ++g) {
         float s = qf[g][0] * f0.x + qf[g][1] * f0.y + qf[g][2] * f1.x +

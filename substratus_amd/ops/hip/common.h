@@ -13,6 +13,14 @@
 
 typedef __hip_bfloat16 bf16_t;
 
+// clang ext_vector (not HIP_vector_type) — required by
+// __builtin_nontemporal_load, which rejects class types.
+typedef __attribute__((ext_vector_type(4))) unsigned int uint4v;
+
+DEV_INLINE uint4v nt_load_u4(const void* p) {
+  return __builtin_nontemporal_load(reinterpret_cast<const uint4v*>(p));
+}
+
 DEV_INLINE float bf2f(bf16_t x) { return __bfloat162float(x); }
 DEV_INLINE bf16_t f2bf(float x) { return __float2bfloat16(x); }
 

@@ -48,6 +48,22 @@ def state() -> ParallelState:
     return _STATE
 
 
+class single_process:
+    """Context manager: build modules as if world_size==1 (e.g. construct a
+    FULL reference model inside a TP process before sharding it)."""
+
+    def __enter__(self):
+        global _STATE
+        self._saved = _STATE
+        _STATE = ParallelState()
+        return self
+
+    def __exit__(self, *exc):
+        global _STATE
+        _STATE = self._saved
+        return False
+
+
 def init_distributed(tp_size: Optional[int] = None,
                      backend: Optional[str] = None) -> ParallelState:
     """Initialize from torchrun env vars. tp_size defaults to world_size
