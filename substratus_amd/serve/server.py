@@ -55,9 +55,14 @@ def resolve_model_name(params: dict, model_dir: str | None = None) -> str:
 
 
 class InferenceWorker:
-    """Owns the engine; steps continuously while there is work."""
+    """Owns the engine; steps continuously while there is work. Under TP
+    (torchrun, tp_size>1) the worker is the scheduling leader: it broadcasts
+    each iteration's admissions to follower ranks before stepping
+    (serve/tp.py)."""
 
     def __init__(self, engine: Engine):
+        from .tp import TPCoordinator
+        self.coord = TPCoordinator()
         self.engine = engine
         self.inbox: queue.Queue = queue.Queue()
         self.outputs: dict[int, queue.Queue] = {}
@@ -81,12 +86,19 @@ class InferenceWorker:
     def _run(self) -> None:
         while not self._stop:
             moved = False
+            pending = []
             while True:
                 try:
-                    (pt, mnt, temp, tk, tp, eos, out_q, holder,
-                     done) = self.inbox.get_nowait()
+                    pending.append(self.inbox.get_nowait())
                 except queue.Empty:
                     break
+            if self.coord.active and (pending or self.engine.has_work()):
+                # lockstep TP: publish this iteration's admissions first
+                self.coord.publish([
+                    dict(prompt_tokens=pt, max_new_tokens=mnt,
+                         temperature=temp, top_k=tk, top_p=tp, eos_token=eos)
+                    for (pt, mnt, temp, tk, tp, eos, *_rest) in pending])
+            for (pt, mnt, temp, tk, tp, eos, out_q, holder, done) in pending:
                 sid = self.engine.add_request(pt, max_new_tokens=mnt,
                                               temperature=temp, top_k=tk,
                                               top_p=tp, eos_token=eos)
@@ -263,13 +275,35 @@ def _maybe_load_weights(engine: Engine,
                 path = os.path.join(model_dir, ckpts[-1], "model.safetensors")
     if os.path.exists(path):
         from safetensors.torch import load_file
+        from ..parallel.comm import state as pstate
         sd = load_file(path)
+        st = pstate()
+        if st.tp_size > 1:
+            from ..models.sharding import shard_state_dict
+            sd = shard_state_dict(sd, engine.mcfg, st.tp_rank, st.tp_size)
         engine.model.load_state_dict(
             {k: v.to(engine.device, engine.cfg.dtype) for k, v in sd.items()})
 
 
 def main() -> None:
     import uvicorn
+    from ..parallel.comm import init_distributed, state as pstate
+    st = init_distributed()  # tp over all ranks of the Pod (amd.com/gpu: N)
+    if st.initialized and st.tp_rank != 0:
+        # follower rank: build the engine, mirror the leader's schedule
+        params = load_params()
+        name = resolve_model_name(params)
+        ecfg = EngineConfig(
+            model=name,
+            dtype=torch.bfloat16 if torch.cuda.is_available() else torch.float32,
+            max_batch=int(params.get("max_batch", 64)),
+            max_seq_len=int(params.get("max_length", 2048)),
+        )
+        engine = Engine(ecfg)
+        _maybe_load_weights(engine)
+        from .tp import follower_loop
+        follower_loop(engine)
+        return
     app = create_app()
     uvicorn.run(app, host="0.0.0.0", port=int(os.environ.get("PORT", 8080)))
 

@@ -1,0 +1,61 @@
+"""Tensor-parallel serving coordination.
+
+A multi-GPU Server Pod (amd.com/gpu: N, reference
+examples/llama2-70b/server.yaml) runs torchrun with one rank per GPU; rank 0
+owns the HTTP endpoint and is the scheduling leader. Every engine iteration
+the leader broadcasts the list of newly admitted requests (torch.distributed
+broadcast_object_list over the TP group), then ALL ranks execute the
+identical engine step — the model's row-parallel layers all-reduce inside the
+forward, so schedules must be lockstep. Sampling is deterministic across
+ranks: logits are replicated after the final all-reduce and the sampling
+draw comes from identically seeded generators.
+"""
+
+from __future__ import annotations
+
+from typing import Optional
+
+import torch.distributed as dist
+
+from ..parallel.comm import state
+
+
+class TPCoordinator:
+    def __init__(self):
+        self.st = state()
+
+    @property
+    def is_leader(self) -> bool:
+        return self.st.tp_rank == 0
+
+    @property
+    def active(self) -> bool:
+        return self.st.initialized and self.st.tp_size > 1
+
+    def publish(self, adds: list[dict], stop: bool = False) -> None:
+        """Leader: broadcast this iteration's admissions to followers."""
+        if not self.active:
+            return
+        obj = [{"adds": adds, "stop": stop}]
+        dist.broadcast_object_list(obj, src=self.st.rank - self.st.tp_rank,
+                                   group=self.st.tp_group)
+
+    def receive(self) -> dict:
+        """Follower: receive the leader's iteration command."""
+        obj = [None]
+        dist.broadcast_object_list(obj, src=self.st.rank - self.st.tp_rank,
+                                   group=self.st.tp_group)
+        return obj[0]
+
+
+def follower_loop(engine) -> None:
+    """Ranks 1..tp-1: mirror the leader's admissions and step in lockstep."""
+    coord = TPCoordinator()
+    while True:
+        cmd = coord.receive()
+        if cmd.get("stop"):
+            return
+        for add in cmd["adds"]:
+            engine.add_request(**add)
+        if engine.has_work():
+            engine.step()
