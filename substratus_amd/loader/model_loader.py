@@ -48,6 +48,9 @@ def run(artifacts_dir: str = "/content/artifacts",
 
     if _try_hub_download(name, files_filter, artifacts_dir):
         pass
+    elif str(params.get("format", "")).lower() == "gguf" or \
+            ".gguf" in files_filter:
+        _write_random_gguf(name, artifacts_dir, params)
     else:
         _write_random_init(name, artifacts_dir, params)
 
@@ -88,6 +91,24 @@ def _write_random_init(name: str, out: str, params: dict) -> None:
     save_file(tensors, os.path.join(out, "model.safetensors"))
     with open(os.path.join(out, "config.json"), "w") as f:
         json.dump({"model": cfg.name, "dtype": str(dtype).split(".")[-1],
+                   "source": "random-init"}, f)
+
+
+def _write_random_gguf(name: str, out: str, params: dict) -> None:
+    """GGUF export of a random-init model (offline stand-in for the
+    reference's llama2-13b-chat-gguf example)."""
+    from ..models.config import get_config
+    from ..models.gguf import export_model_gguf
+    from ..models.model import CausalLM
+    cfg = get_config(name)
+    model = CausalLM(cfg, device="meta", dtype=torch.float32)
+    model = model.to_empty(device="cpu")
+    model.init_random(seed=int(params.get("seed", 0)))
+    quant = params.get("quant", "q4_0")
+    export_model_gguf(os.path.join(out, f"{cfg.name}.{quant}.gguf"),
+                      model.state_dict(), cfg.name, quant)
+    with open(os.path.join(out, "config.json"), "w") as f:
+        json.dump({"model": cfg.name, "format": "gguf", "quant": quant,
                    "source": "random-init"}, f)
 
 

@@ -86,7 +86,7 @@ register(ModelConfig(
 # A tiny llama for CPU tests and CI.
 # Tiny llama for CPU tests/CI and GPU numerics tests (head_dim=64, GQA=2).
 register(ModelConfig(
-    name="llama-tiny", family="llama", hidden_size=256, intermediate_size=688,
+    name="llama-tiny", family="llama", hidden_size=256, intermediate_size=704,
     num_layers=4, num_heads=4, num_kv_heads=2, vocab_size=512,
     max_position_embeddings=2048, rms_norm_eps=1e-5))
 

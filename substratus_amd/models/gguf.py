@@ -1,0 +1,269 @@
+"""GGUF quantized model format: reader, writer, and quantization codecs.
+
+The reference serves GGUF models through `substratusai/model-server-llama-cpp`
+(examples/llama2-13b-chat-gguf/server-gpu.yaml — SURVEY.md §2.5); this module
+is the first-party replacement's format layer. Supported tensor codecs:
+F32, F16, Q8_0 (fp16 scale + 32×int8 per block) and Q4_0 (fp16 scale +
+32×4-bit per block) — the llama.cpp baseline quantizations.
+
+The GPU compute path does NOT dequantize weights to bf16 at load: QuantLinear
+(substratus_amd/models/quant.py) keeps the packed blocks resident and decode
+runs a fused block-dequant GEMV HIP kernel (SURVEY.md §2.5 ledger item 8).
+The offline loader can also EXPORT random-init models to GGUF so the whole
+path is testable with no network.
+"""
+
+from __future__ import annotations
+
+import struct
+from dataclasses import dataclass
+from typing import BinaryIO, Optional
+
+import torch
+
+GGUF_MAGIC = 0x46554747  # "GGUF"
+GGUF_VERSION = 3
+
+# ggml tensor types (subset)
+GGML_F32 = 0
+GGML_F16 = 1
+GGML_Q4_0 = 2
+GGML_Q8_0 = 8
+
+QK = 32  # block size (weights per quant block)
+Q4_0_BLOCK_BYTES = 2 + QK // 2   # fp16 scale + 16 nibbles
+Q8_0_BLOCK_BYTES = 2 + QK        # fp16 scale + 32 int8
+
+# gguf metadata value types
+_T_U8, _T_I8, _T_U16, _T_I16, _T_U32, _T_I32 = 0, 1, 2, 3, 4, 5
+_T_F32, _T_BOOL, _T_STR, _T_ARR, _T_U64, _T_I64, _T_F64 = 6, 7, 8, 9, 10, 11, 12
+
+
+# ---------------------------------------------------------------- quantize
+def quantize_q8_0(w: torch.Tensor) -> torch.Tensor:
+    """w: [..., K] float -> packed uint8 [..., K/32 * 34]."""
+    shape = w.shape
+    assert shape[-1] % QK == 0
+    blocks = w.float().reshape(-1, QK)
+    amax = blocks.abs().amax(dim=1, keepdim=True)
+    scale = (amax / 127.0).clamp(min=1e-12)
+    qs = torch.round(blocks / scale).clamp(-127, 127).to(torch.int8)
+    out = torch.empty(blocks.shape[0], Q8_0_BLOCK_BYTES, dtype=torch.uint8)
+    out[:, :2] = scale.to(torch.float16).view(torch.uint8).reshape(-1, 2)
+    out[:, 2:] = qs.view(torch.uint8)
+    return out.reshape(*shape[:-1], (shape[-1] // QK) * Q8_0_BLOCK_BYTES)
+
+
+def dequantize_q8_0(packed: torch.Tensor, k: int) -> torch.Tensor:
+    rows = packed.reshape(-1, Q8_0_BLOCK_BYTES)
+    scale = rows[:, :2].contiguous().view(torch.float16).float()
+    qs = rows[:, 2:].contiguous().view(torch.int8).float()
+    vals = qs * scale
+    return vals.reshape(*packed.shape[:-1], k)
+
+
+def quantize_q4_0(w: torch.Tensor) -> torch.Tensor:
+    """llama.cpp Q4_0: per 32-block, scale = max|w|/(-8); q = round(w/scale)+8
+    in [0,15]; nibble i packs weights i and i+16 (lo=i, hi=i+16)."""
+    shape = w.shape
+    assert shape[-1] % QK == 0
+    blocks = w.float().reshape(-1, QK)
+    absmax_idx = blocks.abs().argmax(dim=1)
+    maxval = blocks.gather(1, absmax_idx[:, None]).squeeze(1)
+    scale = (maxval / -8.0)
+    scale = torch.where(scale.abs() < 1e-12,
+                        torch.full_like(scale, 1e-12), scale)
+    q = torch.round(blocks / scale[:, None]) + 8
+    q = q.clamp(0, 15).to(torch.uint8)
+    lo = q[:, :16]
+    hi = q[:, 16:]
+    packed = (lo | (hi << 4))
+    out = torch.empty(blocks.shape[0], Q4_0_BLOCK_BYTES, dtype=torch.uint8)
+    out[:, :2] = scale.to(torch.float16).view(torch.uint8).reshape(-1, 2)
+    out[:, 2:] = packed
+    return out.reshape(*shape[:-1], (shape[-1] // QK) * Q4_0_BLOCK_BYTES)
+
+
+def dequantize_q4_0(packed: torch.Tensor, k: int) -> torch.Tensor:
+    rows = packed.reshape(-1, Q4_0_BLOCK_BYTES)
+    scale = rows[:, :2].contiguous().view(torch.float16).float()
+    nib = rows[:, 2:]
+    lo = (nib & 0xF).float() - 8.0
+    hi = (nib >> 4).float() - 8.0
+    vals = torch.cat([lo, hi], dim=1) * scale
+    return vals.reshape(*packed.shape[:-1], k)
+
+
+QUANT_CODECS = {
+    "q4_0": (GGML_Q4_0, Q4_0_BLOCK_BYTES, quantize_q4_0, dequantize_q4_0),
+    "q8_0": (GGML_Q8_0, Q8_0_BLOCK_BYTES, quantize_q8_0, dequantize_q8_0),
+}
+
+
+# ------------------------------------------------------------------ writer
+def _w_str(f: BinaryIO, s: str) -> None:
+    b = s.encode()
+    f.write(struct.pack("<Q", len(b)))
+    f.write(b)
+
+
+def _w_kv(f: BinaryIO, key: str, vtype: int, packfmt: str, value) -> None:
+    _w_str(f, key)
+    f.write(struct.pack("<I", vtype))
+    if vtype == _T_STR:
+        _w_str(f, value)
+    else:
+        f.write(struct.pack(packfmt, value))
+
+
+def write_gguf(path: str, tensors: dict[str, tuple[torch.Tensor, int, list[int]]],
+               metadata: Optional[dict] = None) -> None:
+    """tensors: name -> (raw bytes tensor uint8 or float tensor, ggml_type,
+    logical_shape). metadata: str -> (str|int|float)."""
+    metadata = metadata or {}
+    with open(path, "wb") as f:
+        f.write(struct.pack("<IIQQ", GGUF_MAGIC, GGUF_VERSION,
+                            len(tensors), len(metadata)))
+        for k, v in metadata.items():
+            if isinstance(v, str):
+                _w_kv(f, k, _T_STR, "", v)
+            elif isinstance(v, bool):
+                _w_kv(f, k, _T_BOOL, "<?", v)
+            elif isinstance(v, int):
+                _w_kv(f, k, _T_U64, "<Q", v)
+            elif isinstance(v, float):
+                _w_kv(f, k, _T_F32, "<f", v)
+            else:
+                raise TypeError(f"unsupported metadata type for {k}")
+        # tensor infos
+        offset = 0
+        infos = []
+        for name, (data, gtype, shape) in tensors.items():
+            nbytes = data.numel() * data.element_size()
+            infos.append((name, gtype, shape, offset, data))
+            offset += (nbytes + 31) // 32 * 32  # 32-byte alignment
+        for name, gtype, shape, off, _ in infos:
+            _w_str(f, name)
+            f.write(struct.pack("<I", len(shape)))
+            # gguf dims are innermost-first
+            for dim in reversed(shape):
+                f.write(struct.pack("<Q", dim))
+            f.write(struct.pack("<IQ", gtype, off))
+        # align data section
+        pos = f.tell()
+        pad = (32 - pos % 32) % 32
+        f.write(b"\0" * pad)
+        base = f.tell()
+        for name, gtype, shape, off, data in infos:
+            cur = f.tell() - base
+            if cur < off:
+                f.write(b"\0" * (off - cur))
+            f.write(data.contiguous().numpy().tobytes())
+
+
+# ------------------------------------------------------------------ reader
+def _r_str(f: BinaryIO) -> str:
+    (n,) = struct.unpack("<Q", f.read(8))
+    return f.read(n).decode()
+
+
+_SCALAR_FMT = {_T_U8: "<B", _T_I8: "<b", _T_U16: "<H", _T_I16: "<h",
+               _T_U32: "<I", _T_I32: "<i", _T_F32: "<f", _T_BOOL: "<?",
+               _T_U64: "<Q", _T_I64: "<q", _T_F64: "<d"}
+
+
+def _r_value(f: BinaryIO, vtype: int):
+    if vtype == _T_STR:
+        return _r_str(f)
+    if vtype == _T_ARR:
+        (etype, n) = struct.unpack("<IQ", f.read(12))
+        return [_r_value(f, etype) for _ in range(n)]
+    fmt = _SCALAR_FMT[vtype]
+    return struct.unpack(fmt, f.read(struct.calcsize(fmt)))[0]
+
+
+@dataclass
+class GGUFTensor:
+    name: str
+    shape: list[int]       # logical shape, outermost-first
+    ggml_type: int
+    data: torch.Tensor     # raw uint8 (quantized) or float tensor
+
+    def dequantize(self) -> torch.Tensor:
+        k = self.shape[-1]
+        if self.ggml_type == GGML_F32:
+            return self.data
+        if self.ggml_type == GGML_F16:
+            return self.data.float()
+        if self.ggml_type == GGML_Q4_0:
+            return dequantize_q4_0(self.data, k)
+        if self.ggml_type == GGML_Q8_0:
+            return dequantize_q8_0(self.data, k)
+        raise ValueError(f"unsupported ggml type {self.ggml_type}")
+
+
+def read_gguf(path: str) -> tuple[dict, dict[str, GGUFTensor]]:
+    with open(path, "rb") as f:
+        magic, version, n_tensors, n_kv = struct.unpack("<IIQQ", f.read(24))
+        if magic != GGUF_MAGIC:
+            raise ValueError(f"not a GGUF file: {path}")
+        metadata = {}
+        for _ in range(n_kv):
+            key = _r_str(f)
+            (vtype,) = struct.unpack("<I", f.read(4))
+            metadata[key] = _r_value(f, vtype)
+        infos = []
+        for _ in range(n_tensors):
+            name = _r_str(f)
+            (ndim,) = struct.unpack("<I", f.read(4))
+            dims = [struct.unpack("<Q", f.read(8))[0] for _ in range(ndim)]
+            shape = list(reversed(dims))  # to outermost-first
+            gtype, off = struct.unpack("<IQ", f.read(12))
+            infos.append((name, shape, gtype, off))
+        pos = f.tell()
+        base = pos + (32 - pos % 32) % 32
+        tensors = {}
+        for name, shape, gtype, off in infos:
+            f.seek(base + off)
+            numel = 1
+            for dim in shape:
+                numel *= dim
+            if gtype == GGML_F32:
+                raw = f.read(numel * 4)
+                data = torch.frombuffer(bytearray(raw), dtype=torch.float32)\
+                    .reshape(shape).clone()
+            elif gtype == GGML_F16:
+                raw = f.read(numel * 2)
+                data = torch.frombuffer(bytearray(raw), dtype=torch.float16)\
+                    .reshape(shape).clone()
+            elif gtype in (GGML_Q4_0, GGML_Q8_0):
+                bb = Q4_0_BLOCK_BYTES if gtype == GGML_Q4_0 else Q8_0_BLOCK_BYTES
+                nbytes = numel // QK * bb
+                raw = f.read(nbytes)
+                data = torch.frombuffer(bytearray(raw), dtype=torch.uint8)\
+                    .reshape(*shape[:-1], shape[-1] // QK * bb).clone()
+            else:
+                raise ValueError(f"unsupported ggml type {gtype} for {name}")
+            tensors[name] = GGUFTensor(name, shape, gtype, data)
+        return metadata, tensors
+
+
+def export_model_gguf(path: str, state_dict: dict, model_name: str,
+                      quant: str = "q4_0") -> None:
+    """Write a CausalLM state dict as GGUF: 2-D weights quantized with
+    `quant`, 1-D tensors as F32."""
+    gtype, _bb, qfn, _ = QUANT_CODECS[quant]
+    tensors: dict = {}
+    for name, t in state_dict.items():
+        t = t.detach().float().cpu()
+        # projections quantize; embeddings/norms stay F32 (matches the
+        # in-memory quantize_model policy)
+        if t.dim() >= 2 and t.shape[-1] % QK == 0 and "embed" not in name:
+            tensors[name] = (qfn(t), gtype, list(t.shape))
+        else:
+            tensors[name] = (t, GGML_F32, list(t.shape))
+    write_gguf(path, tensors, metadata={
+        "general.architecture": "substratus-causallm",
+        "general.name": model_name,
+        "substratus.quant": quant,
+    })

@@ -246,6 +246,11 @@ class CausalLM(nn.Module):
                                   device=device, dtype=dtype)
         self.n_kv_heads = self.blocks[0].n_kv_heads  # per-rank
 
+    def _logits(self, y: torch.Tensor) -> torch.Tensor:
+        if self.lm_head is None:
+            return F.linear(y, self.embed_tokens.weight)
+        return self.lm_head(y)  # module call: works for Linear AND QuantLinear
+
     @torch.no_grad()
     def init_random(self, seed: int = 0) -> None:
         """Initialize EVERY parameter (models are built on meta + to_empty,
@@ -278,8 +283,7 @@ class CausalLM(nn.Module):
             nxt = (self.blocks[i + 1].norm1 if i + 1 < len(self.blocks)
                    else self.final_norm)
             y, h = fused_norm_residual(delta, h, nxt, self.eps)
-        w = self.embed_tokens.weight if self.lm_head is None else self.lm_head.weight
-        return F.linear(y, w)
+        return self._logits(y)
 
     # -- prefill -------------------------------------------------------------
     def forward_prefill(self, tokens: torch.Tensor, positions: torch.Tensor,
@@ -294,8 +298,7 @@ class CausalLM(nn.Module):
             h = blk.forward_prefill(h, positions, seq_start, kc, vc, slots)
         last = (seq_start[1:] - 1).long()
         h = self.final_norm(h[last], self.eps)
-        w = self.embed_tokens.weight if self.lm_head is None else self.lm_head.weight
-        return F.linear(h, w)
+        return self._logits(h)
 
     # -- train ---------------------------------------------------------------
     def forward_train(self, tokens: torch.Tensor) -> torch.Tensor:
@@ -320,5 +323,4 @@ class CausalLM(nn.Module):
             h = F.layer_norm(h.float(), (self.cfg.hidden_size,),
                              self.final_norm.weight.float(),
                              self.final_norm.bias.float(), self.eps).to(h.dtype)
-        w = self.embed_tokens.weight if self.lm_head is None else self.lm_head.weight
-        return F.linear(h, w)
+        return self._logits(h)

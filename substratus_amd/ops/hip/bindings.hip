@@ -23,6 +23,8 @@ void launch_softmax_sample(long*, const bf16_t*, const float*, const int*, const
 void launch_cross_entropy(float*, bf16_t*, const bf16_t*, const long*, long, int, long, float, hipStream_t);
 void launch_adamw(bf16_t*, const bf16_t*, float*, float*, long, float, float, float, float, float, int, hipStream_t);
 void launch_adamw_f32(float*, const float*, float*, float*, long, float, float, float, float, float, int, hipStream_t);
+void launch_quant_gemv(bf16_t*, const bf16_t*, const void*, const unsigned char*, int, int, int, bool, hipStream_t);
+void launch_dequant(bf16_t*, const void*, const unsigned char*, long, int, bool, hipStream_t);
 
 namespace {
 
@@ -219,9 +221,36 @@ void adamw_step(torch::Tensor p, torch::Tensor g, torch::Tensor m,
   post_launch_check("adamw_step");
 }
 
+void quant_gemv(torch::Tensor y, torch::Tensor x, torch::Tensor scales,
+                torch::Tensor qs, bool q8) {
+  check_bf16(y, "y"); check_bf16(x, "x");
+  TORCH_CHECK(scales.scalar_type() == torch::kHalf, "scales must be fp16");
+  TORCH_CHECK(qs.scalar_type() == torch::kByte, "qs must be uint8");
+  TORCH_CHECK(scales.is_contiguous() && qs.is_contiguous());
+  int m = x.size(0), k = x.size(1), n = y.size(1);
+  TORCH_CHECK(m >= 1 && m <= 8, "quant_gemv supports M in [1,8], got ", m);
+  TORCH_CHECK(k % 32 == 0, "K must be a multiple of 32");
+  launch_quant_gemv(bf(y), bfc(x), scales.data_ptr(),
+                    qs.data_ptr<unsigned char>(), m, n, k, q8, stream());
+  post_launch_check("quant_gemv");
+}
+
+void dequant(torch::Tensor out, torch::Tensor scales, torch::Tensor qs,
+             bool q8) {
+  check_bf16(out, "out");
+  TORCH_CHECK(scales.scalar_type() == torch::kHalf && qs.scalar_type() == torch::kByte);
+  long n = out.size(0);
+  int k = out.size(1);
+  launch_dequant(bf(out), scales.data_ptr(), qs.data_ptr<unsigned char>(),
+                 n, k, q8, stream());
+  post_launch_check("dequant");
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("quant_gemv", &quant_gemv);
+  m.def("dequant", &dequant);
   m.def("rms_norm", &rms_norm);
   m.def("rms_norm_residual", &rms_norm_residual);
   m.def("layer_norm", &layer_norm);

@@ -41,6 +41,7 @@ class EngineConfig:
     gpu_mem_fraction: float = 0.85         # of free HBM after weights
     prefill_chunk: int = 8192              # max packed prefill tokens per step
     use_graphs: bool = True                # hipGraph-captured decode on GPU
+    quant: Optional[str] = None            # "q4_0" | "q8_0" GGUF weights
     seed: int = 0
 
 
@@ -65,6 +66,12 @@ class Engine:
                 model = CausalLM(self.mcfg, device="meta", dtype=cfg.dtype)
                 model = model.to_empty(device=self.device)
                 model.init_random(cfg.seed)
+        if cfg.quant:
+            from ..models.quant import quantize_model
+            from ..parallel.comm import state as _pstate
+            assert _pstate().tp_size <= 1, \
+                "quantized serving is single-rank for now"
+            quantize_model(model, cfg.quant)
         self.model = model.eval()
         n_kv_local = self.model.n_kv_heads
         if cfg.kv_blocks is not None:

@@ -136,14 +136,28 @@ def create_app(engine: Optional[Engine] = None,
     if engine is None:
         name = model_name or resolve_model_name(params)
         on_gpu = torch.cuda.is_available()
+        quant = None
+        # reference knob: MODEL_LOAD_IN_8BIT / _4BIT env
+        # (examples/llama2-70b/server.yaml)
+        if params.get("load_in_4bit") or os.environ.get("MODEL_LOAD_IN_4BIT"):
+            quant = "q4_0"
+        if params.get("load_in_8bit") or os.environ.get("MODEL_LOAD_IN_8BIT"):
+            quant = "q8_0"
+        gguf_path = _find_gguf()
         ecfg = EngineConfig(
             model=name,
             dtype=torch.bfloat16 if on_gpu else torch.float32,
             max_batch=int(params.get("max_batch", 64)),
             max_seq_len=int(params.get("max_length", 2048)),
+            quant=None if gguf_path else quant,
         )
         engine = Engine(ecfg)
-        _maybe_load_weights(engine)
+        if gguf_path:
+            from ..models.quant import load_gguf_into_model
+            load_gguf_into_model(engine.model, gguf_path, engine.device,
+                                 engine.cfg.dtype)
+        else:
+            _maybe_load_weights(engine)
     worker = InferenceWorker(engine)
     app = FastAPI(title="substratus-amd model server")
     app.state.worker = worker
@@ -259,6 +273,16 @@ def create_app(engine: Optional[Engine] = None,
         })
 
     return app
+
+
+def _find_gguf(model_dir: str | None = None) -> Optional[str]:
+    model_dir = model_dir or os.environ.get("MODEL_DIR", "/content/model")
+    if not os.path.isdir(model_dir):
+        return None
+    for f in sorted(os.listdir(model_dir)):
+        if f.endswith(".gguf"):
+            return os.path.join(model_dir, f)
+    return None
 
 
 def _maybe_load_weights(engine: Engine,
