@@ -35,7 +35,10 @@ void paged_attn_decode_kernel(
     const bf16_t* __restrict__ v_cache,
     const int* __restrict__ block_table, // [B, max_blocks]
     const int* __restrict__ seq_lens,    // [B]
-    int hq, int hkv, int max_blocks, float scale) {
+    int hq, int hkv, int max_blocks, float scale,
+    int nsplit,                           // flash-decoding sequence splits
+    float* __restrict__ part_o,           // [B, HQ, NS, D]  (nsplit > 1)
+    float* __restrict__ part_ml) {        // [B, HQ, NS, 2]
   constexpr int LPT = D / 8;        // lanes per token for K rows (uint4 each)
   constexpr int TPP = WAVE / LPT;   // tokens per pass
   constexpr int PASSES = BS / TPP;
@@ -44,7 +47,8 @@ void paged_attn_decode_kernel(
 
   const int b = blockIdx.x;
   const int h = blockIdx.y;           // kv head
-  const int gchunk = blockIdx.z;      // q-head chunk within the GQA group
+  const int split = blockIdx.z % nsplit;   // sequence split (flash-decoding)
+  const int gchunk = blockIdx.z / nsplit;  // q-head chunk within the group
   const int group = hq / hkv;
   const int g0 = gchunk * GC;
   const int ng = min(GC, group - g0);
@@ -89,9 +93,11 @@ void paged_attn_decode_kernel(
   // K/V stream is read-once per step (each (seq, kv-head) block is owned by
   // one workgroup): non-temporal loads keep it out of L1 (guide nt-weights:
   // -18% issued-to-landed on one-reader streams).
+  const int wstride = NW * nsplit;        // total concurrent waves per (b,h,g)
+  const int w0 = split * NW + wave;       // this wave's first block
   uint4v kt[PASSES];
-  int blk_cur = (wave < nblk) ? bt[wave] : 0;
-  if (wave < nblk) {
+  int blk_cur = (w0 < nblk) ? bt[w0] : 0;
+  if (w0 < nblk) {
     const bf16_t* kbase = k_cache + (long)blk_cur * blk_stride + head_off;
     #pragma unroll
     for (int p = 0; p < PASSES; ++p)
@@ -99,10 +105,10 @@ void paged_attn_decode_kernel(
           reinterpret_cast<const uint4*>(kbase + (p * TPP + sub) * D) + ll);
   }
 
-  for (int bi = wave; bi < nblk; bi += NW) {
+  for (int bi = w0; bi < nblk; bi += wstride) {
     // issue next tile's loads before any compute on the current tile
     uint4v kt_next[PASSES];
-    const int bnext = bi + NW;
+    const int bnext = bi + wstride;
     int blk_next = 0;
     if (bnext < nblk) {
       blk_next = bt[bnext];
@@ -239,23 +245,86 @@ void paged_attn_decode_kernel(
         for (int j = 0; j < DPL; ++j)
           o[j] += f * red_o[w][g][lane * DPL + j];
       }
-      const float invl = (L > 0.f) ? 1.f / L : 0.f;
-      bf16_t* orow = out + ((long)b * hq + h * group + g0 + g) * D;
-      if (DPL == 2) {
-        reinterpret_cast<unsigned int*>(orow)[lane] =
-            pack_bf16x2(o[0] * invl, o[1] * invl);
+      const int qhead = h * group + g0 + g;
+      if (nsplit == 1) {
+        const float invl = (L > 0.f) ? 1.f / L : 0.f;
+        bf16_t* orow = out + ((long)b * hq + qhead) * D;
+        if (DPL == 2) {
+          reinterpret_cast<unsigned int*>(orow)[lane] =
+              pack_bf16x2(o[0] * invl, o[1] * invl);
+        } else {
+          orow[lane] = f2bf(o[0] * invl);
+        }
       } else {
-        orow[lane] = f2bf(o[0] * invl);
+        float* po = part_o + (((long)b * hq + qhead) * nsplit + split) * D;
+        #pragma unroll
+        for (int j = 0; j < DPL; ++j) po[lane * DPL + j] = o[j];
+        if (lane == 0) {
+          float* pm = part_ml + (((long)b * hq + qhead) * nsplit + split) * 2;
+          pm[0] = M;
+          pm[1] = L;
+        }
       }
     }
   }
+}
+
+// flash-decoding combine: one wave per (b, qhead), merge nsplit partials.
+template <int D>
+__global__ __launch_bounds__(WAVE)
+void paged_attn_reduce_kernel(bf16_t* __restrict__ out,      // [B, HQ, D]
+                              const float* __restrict__ part_o,
+                              const float* __restrict__ part_ml,
+                              int hq, int nsplit) {
+  constexpr int DPL = D / WAVE;
+  const int b = blockIdx.x;
+  const int qh = blockIdx.y;
+  const int lane = threadIdx.x;
+  const long base = ((long)b * hq + qh) * nsplit;
+  float M = -INFINITY;
+  for (int s = 0; s < nsplit; ++s) M = fmaxf(M, part_ml[(base + s) * 2]);
+  float L = 0.f;
+  float o[DPL];
+  #pragma unroll
+  for (int j = 0; j < DPL; ++j) o[j] = 0.f;
+  for (int s = 0; s < nsplit; ++s) {
+    const float pm = part_ml[(base + s) * 2];
+    const float f = (pm == -INFINITY) ? 0.f : __expf(pm - M);
+    L += f * part_ml[(base + s) * 2 + 1];
+    const float* po = part_o + (base + s) * D;
+    #pragma unroll
+    for (int j = 0; j < DPL; ++j) o[j] += f * po[lane * DPL + j];
+  }
+  const float invl = (L > 0.f) ? 1.f / L : 0.f;
+  bf16_t* orow = out + ((long)b * hq + qh) * D;
+  if (DPL == 2) {
+    reinterpret_cast<unsigned int*>(orow)[lane] =
+        pack_bf16x2(o[0] * invl, o[1] * invl);
+  } else {
+    orow[lane] = f2bf(o[0] * invl);
+  }
+}
+
+int paged_attn_nsplit(int bsz, int hq, int hkv) {
+  // flash-decoding split factor: fill >=1024 workgroups (256 CUs x 4
+  // resident WGs) when the (B, Hkv, group-chunk) grid alone is too small
+  // (llama2-70b at TP=8: B x 1 x 1). Shape-only -> hipGraph-stable.
+  const int group = hq / hkv;
+  const int gc = (group % 8 == 0) ? 8 : 1;
+  const int gchunks = (group + gc - 1) / gc;
+  const long base = (long)bsz * hkv * gchunks;
+  if (base >= 1024) return 1;
+  int ns = (int)((1024 + base - 1) / base);
+  if (ns > 16) ns = 16;
+  return ns;
 }
 
 void launch_paged_attn_decode(bf16_t* out, const bf16_t* q,
                               const bf16_t* k_cache, const bf16_t* v_cache,
                               const int* block_table, const int* seq_lens,
                               int bsz, int hq, int hkv, int d, int max_blocks,
-                              int block_size, float scale,
+                              int block_size, float scale, int nsplit,
+                              float* part_o, float* part_ml,
                               hipStream_t stream) {
   const int group = hq / hkv;
   // GC=8 when the GQA group tiles evenly by 8 (llama2-70b, llama3-style);
@@ -263,17 +332,27 @@ void launch_paged_attn_decode(bf16_t* out, const bf16_t* q,
   // tiny and L3-resident, so per-head K re-reads are cheap).
   const int gc = (group % 8 == 0) ? 8 : 1;
   const int gchunks = (group + gc - 1) / gc;
-  dim3 grid(bsz, hkv, gchunks);
+  dim3 grid(bsz, hkv, gchunks * nsplit);
   dim3 block(NW * WAVE);
   #define LAUNCH(DD, GG) \
     paged_attn_decode_kernel<DD, GG><<<grid, block, 0, stream>>>( \
         out, q, k_cache, v_cache, block_table, seq_lens, hq, hkv, \
-        max_blocks, scale)
+        max_blocks, scale, nsplit, part_o, part_ml)
   if (d == 128 && gc == 8) LAUNCH(128, 8);
   else if (d == 128) LAUNCH(128, 1);
   else if (d == 64 && gc == 8) LAUNCH(64, 8);
   else if (d == 64) LAUNCH(64, 1);
   #undef LAUNCH
+  if (nsplit > 1) {
+    dim3 rgrid(bsz, hq);
+    if (d == 128) {
+      paged_attn_reduce_kernel<128><<<rgrid, dim3(WAVE), 0, stream>>>(
+          out, part_o, part_ml, hq, nsplit);
+    } else {
+      paged_attn_reduce_kernel<64><<<rgrid, dim3(WAVE), 0, stream>>>(
+          out, part_o, part_ml, hq, nsplit);
+    }
+  }
 }
 
 bool paged_attn_decode_supported(int d, int block_size) {

@@ -14,7 +14,8 @@ void launch_layer_norm(bf16_t*, const bf16_t*, const bf16_t*, const bf16_t*, lon
 void launch_silu_mul(bf16_t*, const bf16_t*, long, int, hipStream_t);
 void launch_gelu(bf16_t*, const bf16_t*, long, hipStream_t);
 void launch_rope_kv_append(bf16_t*, const bf16_t*, const bf16_t*, const int*, bf16_t*, bf16_t*, const int*, int, int, int, int, int, float, hipStream_t);
-void launch_paged_attn_decode(bf16_t*, const bf16_t*, const bf16_t*, const bf16_t*, const int*, const int*, int, int, int, int, int, int, float, hipStream_t);
+void launch_paged_attn_decode(bf16_t*, const bf16_t*, const bf16_t*, const bf16_t*, const int*, const int*, int, int, int, int, int, int, float, int, float*, float*, hipStream_t);
+int paged_attn_nsplit(int, int, int);
 bool paged_attn_decode_supported(int, int);
 void launch_attn_prefill(bf16_t*, const bf16_t*, const bf16_t*, const bf16_t*, const int*, int, int, int, int, int, float, hipStream_t);
 bool attn_prefill_supported(int);
@@ -134,10 +135,21 @@ void paged_attn_decode(torch::Tensor out, torch::Tensor q,
               "paged_attn_decode: unsupported head_dim ", d, " / block_size ",
               block_size);
   TORCH_CHECK(hq % hkv == 0, "hq must be divisible by hkv");
+  const int nsplit = paged_attn_nsplit(bsz, hq, hkv);
+  float *po = nullptr, *pml = nullptr;
+  torch::Tensor part_o, part_ml;
+  if (nsplit > 1) {
+    auto opts = q.options().dtype(torch::kFloat);
+    part_o = torch::empty({bsz, hq, nsplit, d}, opts);
+    part_ml = torch::empty({bsz, hq, nsplit, 2}, opts);
+    po = part_o.data_ptr<float>();
+    pml = part_ml.data_ptr<float>();
+  }
   launch_paged_attn_decode(bf(out), bfc(q), bfc(k_cache), bfc(v_cache),
                            block_table.data_ptr<int>(),
                            seq_lens.data_ptr<int>(), bsz, hq, hkv, d,
-                           max_blocks, block_size, (float)scale, stream());
+                           max_blocks, block_size, (float)scale, nsplit,
+                           po, pml, stream());
   post_launch_check("paged_attn_decode");
 }
 
