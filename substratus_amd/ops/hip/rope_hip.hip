@@ -19,8 +19,9 @@
 
 #define RW 4  // waves per workgroup
 
-// grid: (ntok); block: RW*64. head_dim/2 <= 64 assumed per lane pass
+// grid: (ntok, role_chunks); block: RW*64. head_dim/2 <= 64 per lane pass
 // (d=128 -> 1 pair per lane; d=64 -> lanes 0..31 active per pass).
+// role chunking keeps >=2048 WGs in flight at any batch size.
 __global__ __launch_bounds__(RW * WAVE)
 void rope_kv_append_kernel(
     bf16_t* __restrict__ q,              // [ntok, hq, d]
@@ -49,7 +50,8 @@ void rope_kv_append_kernel(
   }
 
   const int nroles = hq + 2 * hkv;
-  for (int role = wave; role < nroles; role += RW) {
+  for (int role = blockIdx.y * RW + wave; role < nroles;
+       role += gridDim.y * RW) {
     if (role >= hq + hkv) {  // V copy
       const int h = role - hq - hkv;
       const uint4* src = reinterpret_cast<const uint4*>(
@@ -93,7 +95,10 @@ void launch_rope_kv_append(bf16_t* q, const bf16_t* k, const bf16_t* v,
                            int hq, int hkv, int d, int block_size,
                            float theta, hipStream_t stream) {
   float ln_theta = theta > 0.f ? logf(theta) : 0.f;
- hipLaunchKernelGGL(( rope_kv_append_kernel), dim3(dim3(ntok)), dim3(dim3(RW * WAVE)), 0, stream, 
+  const int nroles = hq + 2 * hkv;
+  int chunks = 1;
+  while (ntok * chunks < 2048 && chunks * RW < nroles) chunks *= 2;
+ hipLaunchKernelGGL(( rope_kv_append_kernel), dim3(dim3(ntok, chunks)), dim3(dim3(RW * WAVE)), 0, stream, 
       q, k, v, positions, k_cache, v_cache, slots, hq, hkv, d, block_size,
       ln_theta, theta > 0.f);
 }
