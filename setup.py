@@ -24,6 +24,10 @@ setup(
     name="substratus_amd",
     version="0.1.0",
     packages=["substratus_amd"],
+    entry_points={"console_scripts": [
+        "sub = substratus_amd.cli.main:main",
+        "nbwatch = substratus_amd.containertools.nbwatch:watch",
+    ]},
     ext_modules=[
         CUDAExtension(
             name="substratus_amd._C",

@@ -1,0 +1,259 @@
+"""`sub` — the substratus CLI (reference cmd/sub + internal/cli, cobra).
+
+Commands (reference internal/cli/*.go):
+  sub up                      start the local platform daemon
+  sub run [dir]               tar dir → Model/Dataset with build.upload →
+                              upload → wait ready (-i increment, -r replace)
+  sub serve [dir]             apply the Server manifest → wait → print URL
+  sub notebook [dir] | nb     convert manifest to Notebook, apply, wait
+  sub apply -f FILE           server-side apply any manifest
+  sub get [scope]             list kinds with ready state (models, models/x)
+  sub delete kind/name        delete
+  sub infer                   completion request against a ready Server
+
+Rendering uses rich tables/spinners instead of the reference's bubbletea TUI
+(internal/tui; same information architecture: ready checkmarks, conditions,
+endpoints).
+"""
+
+from __future__ import annotations
+
+import glob
+import json
+import os
+import subprocess
+import sys
+import time
+from typing import Optional
+
+import typer
+import yaml
+from rich.console import Console
+from rich.table import Table
+
+from ..api import types as apiv1
+from . import client as libclient
+
+app = typer.Typer(help="substratus-amd: ML platform CLI", no_args_is_help=True)
+console = Console()
+
+API_URL = os.environ.get("SUBSTRATUS_API", "http://127.0.0.1:8070")
+
+
+def _client() -> libclient.HTTPClient:
+    c = libclient.HTTPClient(API_URL)
+    try:
+        c.http.get("/healthz").raise_for_status()
+    except Exception:
+        console.print("[red]cannot reach the platform API at "
+                      f"{API_URL} — run `sub up` first[/red]")
+        raise typer.Exit(1)
+    return c
+
+
+def _find_manifests(path: str, kinds: tuple[str, ...]) -> list[dict]:
+    """Find *.yaml manifests of the given kinds (reference
+    internal/tui/manifests.go:130-263)."""
+    out = []
+    files = [path] if os.path.isfile(path) else \
+        sorted(glob.glob(os.path.join(path, "*.yaml")) +
+               glob.glob(os.path.join(path, "*.yml")))
+    for f in files:
+        with open(f) as fh:
+            for doc in yaml.safe_load_all(fh):
+                if isinstance(doc, dict) and doc.get("kind") in kinds:
+                    out.append(doc)
+    return out
+
+
+@app.command()
+def up(foreground: bool = typer.Option(False, "--foreground", "-f")):
+    """Start the local platform daemon (apiserver + controllers + runtime +
+    SCI — the kind-cluster equivalent, install/kind/up.sh)."""
+    if foreground:
+        from ..kube.apiserver import main as serve_main
+        serve_main()
+        return
+    log = os.path.expanduser("~/.substratus/daemon.log")
+    os.makedirs(os.path.dirname(log), exist_ok=True)
+    with open(log, "a") as lf:
+        subprocess.Popen(
+            [sys.executable, "-m", "substratus_amd.kube.apiserver"],
+            stdout=lf, stderr=lf, start_new_session=True)
+    for _ in range(50):
+        try:
+            import httpx
+            httpx.get(f"{API_URL}/healthz", timeout=1).raise_for_status()
+            console.print(f"[green]platform up at {API_URL}[/green]")
+            return
+        except Exception:
+            time.sleep(0.2)
+    console.print(f"[red]daemon did not come up; see {log}[/red]")
+    raise typer.Exit(1)
+
+
+@app.command()
+def run(directory: str = typer.Argument("."),
+        increment: bool = typer.Option(False, "-i"),
+        replace: bool = typer.Option(False, "-r"),
+        wait: bool = typer.Option(True)):
+    """Build-and-run a Model/Dataset from a directory
+    (reference internal/cli/run.go:16-104)."""
+    c = _client()
+    manifests = _find_manifests(directory, ("Model", "Dataset"))
+    if not manifests:
+        console.print("[red]no Model/Dataset manifest found[/red]")
+        raise typer.Exit(1)
+    for obj in manifests:
+        kind, name = obj["kind"], obj["metadata"]["name"]
+        ns = obj["metadata"].setdefault("namespace", "default")
+        if replace and c.get(kind, name, ns) is not None:
+            c.delete(kind, name, ns)
+        elif increment:
+            obj["metadata"]["name"] = libclient.increment_name(c, kind, name, ns)
+        has_docker = os.path.exists(os.path.join(directory, "Dockerfile"))
+        if has_docker and not (obj.get("spec") or {}).get("image"):
+            with console.status(f"uploading build context for {kind}/{name}"):
+                obj = libclient.upload_build(c, obj, directory)
+        else:
+            obj = c.apply(obj)
+        if wait:
+            _wait_and_report(c, obj)
+
+
+def _wait_and_report(c, obj: dict) -> None:
+    kind, name = obj["kind"], obj["metadata"]["name"]
+    ns = obj["metadata"].get("namespace", "default")
+    with console.status(f"waiting for {kind}/{name} to become ready") as st:
+        def prog(o):
+            conds = o.get("status", {}).get("conditions", []) or []
+            if conds:
+                last = conds[-1]
+                st.update(f"{kind}/{name}: {last.get('type')}="
+                          f"{last.get('status')} ({last.get('reason', '')})")
+        libclient.wait_ready(c, kind, name, ns, progress=prog)
+    console.print(f"[green]✓ {kind}/{name} ready[/green]")
+
+
+@app.command()
+def serve(directory: str = typer.Argument(".")):
+    """Apply the Server manifest in the directory and wait for it
+    (reference internal/cli/serve.go:17-96)."""
+    c = _client()
+    manifests = _find_manifests(directory, ("Server",))
+    if not manifests:
+        console.print("[red]no Server manifest found[/red]")
+        raise typer.Exit(1)
+    for obj in manifests:
+        obj = c.apply(obj)
+        _wait_and_report(c, obj)
+        name = obj["metadata"]["name"]
+        ns = obj["metadata"].get("namespace", "default")
+        svc = c.get("Service", f"{name}-server", ns)
+        ep = (svc or {}).get("metadata", {}).get("annotations", {}).get(
+            "substratus.ai/local-endpoint", "")
+        if ep:
+            console.print(f"serving at [bold]{ep}[/bold] "
+                          f"(POST {ep}/v1/completions)")
+
+
+@app.command()
+def notebook(directory: str = typer.Argument(".")):
+    """Convert the directory's manifest to a Notebook and open it
+    (reference internal/cli/notebook.go:16-107)."""
+    c = _client()
+    manifests = _find_manifests(
+        directory, ("Model", "Dataset", "Server", "Notebook"))
+    if not manifests:
+        console.print("[red]no manifest found[/red]")
+        raise typer.Exit(1)
+    nb = manifests[0] if manifests[0]["kind"] == "Notebook" else \
+        libclient.notebook_for_object(manifests[0])
+    nb = c.apply(nb)
+    _wait_and_report(c, nb)
+    console.print(f"notebook [bold]{nb['metadata']['name']}[/bold] ready "
+                  "(port 8888, token: default)")
+
+
+app.command(name="nb")(notebook)
+
+
+@app.command()
+def apply(file: str = typer.Option(..., "-f", "--filename")):
+    """Server-side apply any manifest (reference internal/cli/apply.go)."""
+    c = _client()
+    with open(file) as fh:
+        for doc in yaml.safe_load_all(fh):
+            if not isinstance(doc, dict):
+                continue
+            out = c.apply(doc)
+            console.print(f"{out['kind'].lower()}/"
+                          f"{out['metadata']['name']} applied")
+
+
+@app.command()
+def get(scope: str = typer.Argument("all")):
+    """List objects: `sub get models`, `sub get models/name`, `sub get all`
+    (reference internal/cli/get.go; scope grammar tui/get.go:219-287)."""
+    c = _client()
+    kind_filter, name_filter = None, None
+    if scope != "all":
+        parts = scope.split("/")
+        kind_filter = parts[0].rstrip("s").capitalize()
+        if len(parts) > 1:
+            name_filter = parts[1]
+    table = Table()
+    for col in ("KIND", "NAME", "READY", "CONDITION"):
+        table.add_column(col)
+    for kind in apiv1.KINDS:
+        if kind_filter and kind != kind_filter:
+            continue
+        for obj in c.list(kind):
+            name = obj["metadata"]["name"]
+            if name_filter and name != name_filter:
+                continue
+            ready = "✓" if apiv1.get_status_ready(obj) else "…"
+            conds = obj.get("status", {}).get("conditions", []) or []
+            last = (f"{conds[-1].get('type')}={conds[-1].get('status')}"
+                    f" ({conds[-1].get('reason', '')})") if conds else ""
+            table.add_row(kind, name, ready, last)
+    console.print(table)
+
+
+@app.command()
+def delete(target: str):
+    """Delete kind/name (reference internal/cli/delete.go)."""
+    c = _client()
+    kind, name = target.split("/", 1)
+    kind = kind.rstrip("s").capitalize()
+    c.delete(kind, name)
+    console.print(f"{kind.lower()}/{name} deleted")
+
+
+@app.command()
+def infer(server: str = typer.Option(..., "--server", "-s"),
+          prompt: str = typer.Argument(...),
+          max_tokens: int = typer.Option(32)):
+    """One completion request against a ready Server
+    (reference internal/cli/infer.go — the chat TUI stub, made functional)."""
+    import httpx
+    c = _client()
+    svc = c.get("Service", f"{server}-server")
+    ep = (svc or {}).get("metadata", {}).get("annotations", {}).get(
+        "substratus.ai/local-endpoint")
+    if not ep:
+        console.print(f"[red]server {server} has no local endpoint[/red]")
+        raise typer.Exit(1)
+    r = httpx.post(f"{ep}/v1/completions",
+                   json={"prompt": prompt, "max_tokens": max_tokens},
+                   timeout=300)
+    r.raise_for_status()
+    console.print(r.json()["choices"][0]["text"])
+
+
+def main() -> None:
+    app()
+
+
+if __name__ == "__main__":
+    main()

@@ -1,0 +1,145 @@
+"""HTTP API server over the in-memory Kubernetes API + platform daemon.
+
+This is what makes the platform real without a cluster: `sub up` starts one
+process serving the substratus.ai/v1 API over HTTP, with the controller
+manager, the local runtime (kubelet-equivalent) and the kind SCI all running
+inside — the moral equivalent of the reference's
+install/kubernetes/kind/system.yaml stack (namespace + CRDs + controller-
+manager Deployment + SCI + registry) collapsed into one binary for
+single-node MI355X deployments. The `sub` CLI talks to this API exactly like
+the reference CLI talks to a kube-apiserver (server-side apply + status
+polling — reference internal/client/client.go:39-152).
+"""
+
+from __future__ import annotations
+
+import os
+import threading
+import time
+from typing import Optional
+
+from fastapi import FastAPI, HTTPException, Request
+from fastapi.responses import JSONResponse
+
+from ..api import types as apiv1
+from .memapi import Conflict, MemKube, NotFound
+
+
+def create_api(kube: MemKube) -> FastAPI:
+    app = FastAPI(title="substratus-amd apiserver")
+    app.state.kube = kube
+
+    @app.get("/healthz")
+    def healthz():
+        return {"ok": True}
+
+    @app.post("/api/objects")
+    async def create(req: Request):
+        obj = await req.json()
+        errs = apiv1.validate(obj) if obj.get("kind") in apiv1.KINDS else []
+        if errs:
+            raise HTTPException(422, "; ".join(errs))
+        try:
+            return kube.create(obj)
+        except Conflict as e:
+            raise HTTPException(409, str(e))
+
+    @app.put("/api/objects")
+    async def apply(req: Request):
+        obj = await req.json()
+        errs = apiv1.validate(obj) if obj.get("kind") in apiv1.KINDS else []
+        if errs:
+            raise HTTPException(422, "; ".join(errs))
+        return kube.apply(obj)
+
+    @app.get("/api/objects/{kind}")
+    def list_objs(kind: str, namespace: str = "default"):
+        return {"items": kube.list(kind.capitalize() if kind[0].islower()
+                                   else kind, namespace)}
+
+    @app.get("/api/objects/{kind}/{namespace}/{name}")
+    def get_obj(kind: str, namespace: str, name: str):
+        try:
+            return kube.get(kind, name, namespace)
+        except NotFound:
+            raise HTTPException(404, f"{kind}/{namespace}/{name} not found")
+
+    @app.patch("/api/objects/{kind}/{namespace}/{name}")
+    async def patch_obj(kind: str, namespace: str, name: str, req: Request):
+        body = await req.json()
+        try:
+            return kube.patch(kind, name, body, namespace)
+        except NotFound:
+            raise HTTPException(404, "not found")
+
+    @app.delete("/api/objects/{kind}/{namespace}/{name}")
+    def delete_obj(kind: str, namespace: str, name: str):
+        try:
+            kube.delete(kind, name, namespace)
+        except NotFound:
+            raise HTTPException(404, "not found")
+        return {"deleted": True}
+
+    return app
+
+
+class Platform:
+    """Embedded control plane: MemKube + controllers + local runtime + SCI."""
+
+    def __init__(self, data_dir: str):
+        from ..cloud import new_cloud
+        from ..controller.manager import Manager
+        from ..runtime.local import LocalRuntime
+        from ..sci.kind_server import KindSCIService
+        os.makedirs(data_dir, exist_ok=True)
+        self.bucket_dir = os.path.join(data_dir, "bucket")
+        os.makedirs(self.bucket_dir, exist_ok=True)
+        self.kube = MemKube()
+        self.cloud = new_cloud({"CLOUD": "kind", "CLUSTER_NAME": "local",
+                                "BUCKET_DIR": self.bucket_dir})
+        self.sci_http_port = int(os.environ.get("SCI_HTTP_PORT", "8071"))
+        from ..sci.kind_server import serve as sci_serve
+        self._sci_server, self._sci_httpd = sci_serve(
+            self.bucket_dir, grpc_port=int(os.environ.get("SCI_GRPC_PORT",
+                                                          "10080")),
+            http_port=self.sci_http_port,
+            url_host=f"http://127.0.0.1:{self.sci_http_port}", block=False)
+        from ..sci.client import GrpcSCIClient
+        sci_client = GrpcSCIClient(
+            f"localhost:{os.environ.get('SCI_GRPC_PORT', '10080')}")
+        self.mgr = Manager(self.kube, self.cloud, sci_client)
+        self.runtime = LocalRuntime(self.kube, self.bucket_dir)
+        self._stop = False
+        self._thread = threading.Thread(target=self._loop, daemon=True)
+        self._thread.start()
+
+    def _loop(self) -> None:
+        while not self._stop:
+            try:
+                self.mgr.run_once()
+                self.mgr._pump_delayed()
+                self.runtime.sync()
+            except Exception:  # keep the control loop alive
+                import traceback
+                traceback.print_exc()
+            time.sleep(0.25)
+
+    def stop(self) -> None:
+        self._stop = True
+        self.runtime.stop()
+        self._sci_server.stop(grace=None)
+        self._sci_httpd.shutdown()
+
+
+def main() -> None:
+    import uvicorn
+    data_dir = os.environ.get("SUBSTRATUS_DATA",
+                              os.path.expanduser("~/.substratus"))
+    platform = Platform(data_dir)
+    app = create_api(platform.kube)
+    port = int(os.environ.get("SUBSTRATUS_API_PORT", "8070"))
+    uvicorn.run(app, host="127.0.0.1", port=port, log_level="warning")
+
+
+if __name__ == "__main__":
+    main()

@@ -1,0 +1,163 @@
+"""CLI end-to-end tests against a live embedded platform daemon — the
+reference's CLI behaviors (run with upload, serve, get, delete, apply,
+notebook conversion) over the real HTTP API."""
+
+import json
+import os
+import socket
+import threading
+import time
+
+import pytest
+import yaml
+from typer.testing import CliRunner
+
+import substratus_amd.cli.main as cli
+from substratus_amd.kube.apiserver import Platform, create_api
+
+runner = CliRunner()
+
+
+def _free_port():
+    s = socket.socket()
+    s.bind(("127.0.0.1", 0))
+    p = s.getsockname()[1]
+    s.close()
+    return p
+
+
+@pytest.fixture(scope="module")
+def daemon(tmp_path_factory):
+    import uvicorn
+    data = tmp_path_factory.mktemp("platform")
+    os.environ["SCI_GRPC_PORT"] = str(_free_port())
+    os.environ["SCI_HTTP_PORT"] = str(_free_port())
+    platform = Platform(str(data))
+    app = create_api(platform.kube)
+    port = _free_port()
+    config = uvicorn.Config(app, host="127.0.0.1", port=port,
+                            log_level="error")
+    server = uvicorn.Server(config)
+    t = threading.Thread(target=server.run, daemon=True)
+    t.start()
+    url = f"http://127.0.0.1:{port}"
+    for _ in range(100):
+        try:
+            import httpx
+            httpx.get(f"{url}/healthz", timeout=1).raise_for_status()
+            break
+        except Exception:
+            time.sleep(0.1)
+    old = cli.API_URL
+    cli.API_URL = url
+    yield platform, url
+    cli.API_URL = old
+    server.should_exit = True
+    platform.stop()
+
+
+def test_apply_get_delete(daemon, tmp_path):
+    manifest = tmp_path / "model.yaml"
+    manifest.write_text(yaml.dump({
+        "apiVersion": "substratus.ai/v1", "kind": "Model",
+        "metadata": {"name": "cli-m1"},
+        "spec": {"image": "substratus-amd/model-loader",
+                 "params": {"name": "llama-tiny"}},
+    }))
+    r = runner.invoke(cli.app, ["apply", "-f", str(manifest)])
+    assert r.exit_code == 0, r.output
+    assert "model/cli-m1 applied" in r.output
+    r = runner.invoke(cli.app, ["get", "models"])
+    assert "cli-m1" in r.output
+    r = runner.invoke(cli.app, ["get", "models/cli-m1"])
+    assert "cli-m1" in r.output
+    r = runner.invoke(cli.app, ["delete", "models/cli-m1"])
+    assert r.exit_code == 0
+    r = runner.invoke(cli.app, ["get", "models"])
+    assert "cli-m1" not in r.output
+
+
+def test_run_with_upload_builds(daemon, tmp_path):
+    platform, url = daemon
+    ctx = tmp_path / "ctx"
+    ctx.mkdir()
+    (ctx / "Dockerfile").write_text("FROM scratch\n")
+    (ctx / "dataset.yaml").write_text(yaml.dump({
+        "apiVersion": "substratus.ai/v1", "kind": "Dataset",
+        "metadata": {"name": "cli-ds"},
+        "spec": {"params": {"synthetic_rows": 5}},
+    }))
+    # fake the build job completing (no kaniko locally: the runtime marks
+    # unknown images complete; the data-loader Job then runs for real)
+    r = runner.invoke(cli.app, ["run", str(ctx)])
+    assert r.exit_code == 0, r.output
+    assert "ready" in r.output
+    obj = platform.kube.get("Dataset", "cli-ds")
+    assert obj["status"]["ready"] is True
+    assert obj["spec"]["image"]  # image stamped by the build reconciler
+    # -i increments the name
+    r = runner.invoke(cli.app, ["run", str(ctx), "-i"])
+    assert r.exit_code == 0, r.output
+    assert platform.kube.try_get("Dataset", "cli-ds-2") is not None
+
+
+def test_serve_and_infer_end_to_end(daemon, tmp_path):
+    platform, url = daemon
+    d = tmp_path / "srv"
+    d.mkdir()
+    (d / "base-model.yaml").write_text(yaml.dump({
+        "apiVersion": "substratus.ai/v1", "kind": "Model",
+        "metadata": {"name": "cli-base"},
+        "spec": {"image": "substratus-amd/model-loader",
+                 "params": {"name": "llama-tiny"}},
+    }))
+    r = runner.invoke(cli.app, ["run", str(d)])
+    assert r.exit_code == 0, r.output
+    (d / "server.yaml").write_text(yaml.dump({
+        "apiVersion": "substratus.ai/v1", "kind": "Server",
+        "metadata": {"name": "cli-srv"},
+        "spec": {"image": "substratus-amd/model-server",
+                 "model": {"name": "cli-base"},
+                 "params": {"max_length": 128, "max_batch": 2}},
+    }))
+    r = runner.invoke(cli.app, ["serve", str(d)])
+    assert r.exit_code == 0, r.output
+    assert "serving at" in r.output
+    r = runner.invoke(cli.app, ["infer", "--server", "cli-srv",
+                                "hello world"])
+    assert r.exit_code == 0, r.output
+
+
+def test_notebook_conversion(daemon, tmp_path):
+    from substratus_amd.cli.client import notebook_for_object
+    from substratus_amd.api import types as apiv1
+    obj = apiv1.new_object("Server", "x", {
+        "image": "i", "model": {"name": "m"}, "params": {"a": 1}})
+    nb = notebook_for_object(obj)
+    assert nb["kind"] == "Notebook"
+    assert nb["spec"]["model"] == {"name": "m"}
+    assert nb["spec"]["params"] == {"a": 1}
+
+
+def test_nbwatch_events(tmp_path):
+    import subprocess, sys
+    root = tmp_path / "content"
+    root.mkdir()
+    (root / "keep.py").write_text("x = 1\n")
+    (root / "data").mkdir()
+    proc = subprocess.Popen(
+        [sys.executable, "-c",
+         f"from substratus_amd.containertools.nbwatch import watch; "
+         f"watch({str(root)!r}, interval=0.2, iterations=12)"],
+        stdout=subprocess.PIPE, cwd=os.path.dirname(os.path.dirname(
+            os.path.abspath(__file__))))
+    time.sleep(0.5)
+    (root / "new.py").write_text("y\n")
+    (root / "keep.py").write_text("x = 2\n")
+    (root / "data" / "ignored.bin").write_text("z")  # skipped dir
+    out, _ = proc.communicate(timeout=30)
+    events = [json.loads(l) for l in out.decode().splitlines()]
+    ops = {(e["op"], os.path.basename(e["path"])) for e in events}
+    assert ("CREATE", "new.py") in ops
+    assert ("WRITE", "keep.py") in ops
+    assert not any("ignored" in p for _, p in ops)
