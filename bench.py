@@ -39,6 +39,8 @@ def parse_args():
     p.add_argument("--batch", type=int, default=256)
     p.add_argument("--prompt-len", type=int, default=512)
     p.add_argument("--seq-len", type=int, default=None)
+    p.add_argument("--quant", default=None,
+                   help="GGUF weight quantization: q4_0 | q8_0 (config 4)")
     p.add_argument("--tp", type=int, default=1,
                    help="tensor-parallel degree (serve); default DP replicas")
     return p.parse_args()
@@ -65,6 +67,7 @@ def bench_serve(args) -> None:
     cfg = EngineConfig(model=model_name, dtype=dtype, max_batch=args.batch,
                        max_seq_len=max_seq,
                        prefill_chunk=args.batch * args.prompt_len,
+                       quant=args.quant,
                        seed=1234 + st.dp_rank)
     log(f"[bench] building {model_name} engine on {device} ...")
     t_build = time.time()
@@ -134,6 +137,7 @@ def bench_serve(args) -> None:
             "prompt_len": args.prompt_len,
             "parallelism": parallelism,
             "p50_decode_latency_ms": round(ms_per_step, 3),
+            **({"weights": args.quant} if args.quant else {}),
         },
     }
     if st.rank == 0:

@@ -48,7 +48,8 @@ def quantize_q8_0(w: torch.Tensor) -> torch.Tensor:
     amax = blocks.abs().amax(dim=1, keepdim=True)
     scale = (amax / 127.0).clamp(min=1e-12)
     qs = torch.round(blocks / scale).clamp(-127, 127).to(torch.int8)
-    out = torch.empty(blocks.shape[0], Q8_0_BLOCK_BYTES, dtype=torch.uint8)
+    out = torch.empty(blocks.shape[0], Q8_0_BLOCK_BYTES, dtype=torch.uint8,
+                      device=w.device)
     out[:, :2] = scale.to(torch.float16).view(torch.uint8).reshape(-1, 2)
     out[:, 2:] = qs.view(torch.uint8)
     return out.reshape(*shape[:-1], (shape[-1] // QK) * Q8_0_BLOCK_BYTES)
@@ -78,7 +79,8 @@ def quantize_q4_0(w: torch.Tensor) -> torch.Tensor:
     lo = q[:, :16]
     hi = q[:, 16:]
     packed = (lo | (hi << 4))
-    out = torch.empty(blocks.shape[0], Q4_0_BLOCK_BYTES, dtype=torch.uint8)
+    out = torch.empty(blocks.shape[0], Q4_0_BLOCK_BYTES, dtype=torch.uint8,
+                      device=w.device)
     out[:, :2] = scale.to(torch.float16).view(torch.uint8).reshape(-1, 2)
     out[:, 2:] = packed
     return out.reshape(*shape[:-1], (shape[-1] // QK) * Q4_0_BLOCK_BYTES)

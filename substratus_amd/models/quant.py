@@ -20,7 +20,8 @@ import torch
 import torch.nn as nn
 import torch.nn.functional as F
 
-from .gguf import QK, dequantize_q4_0, dequantize_q8_0, quantize_q4_0, quantize_q8_0
+from .gguf import (QK, dequantize_q4_0, dequantize_q8_0, quantize_q4_0,
+                   quantize_q8_0)
 
 
 def repack(packed: torch.Tensor, k: int, quant: str) -> tuple[torch.Tensor, torch.Tensor]:
@@ -56,7 +57,9 @@ class QuantLinear(nn.Module):
 
     @classmethod
     def from_linear(cls, lin: nn.Module, quant: str) -> "QuantLinear":
-        w = lin.weight.detach().float().cpu()
+        # quantize on the weight's own device (GPU for big models: a 13b
+        # CPU quantize would take minutes)
+        w = lin.weight.detach().float()
         qfn = quantize_q4_0 if quant == "q4_0" else quantize_q8_0
         packed = qfn(w)
         scales, qs = repack(packed, w.shape[1], quant)
