@@ -1,0 +1,43 @@
+# substratus-amd dev loop (reference Makefile:98-156 equivalents)
+
+PY ?= python
+
+.PHONY: build
+build:  ## build the gfx950 HIP extension in-tree
+	PYTORCH_ROCM_ARCH=gfx950 $(PY) setup.py build_ext --inplace
+
+.PHONY: test
+test:  ## CPU test suite (unit + integration + system; no GPU needed)
+	$(PY) -m pytest tests/ -q -m "not gpu"
+
+.PHONY: test-unit
+test-unit:
+	$(PY) -m pytest tests/test_controllers.py tests/test_quant_cpu.py -q
+
+.PHONY: test-integration
+test-integration:  ## controllers against the in-memory API + gRPC SCI
+	$(PY) -m pytest tests/test_controllers.py tests/test_sci.py tests/test_cli.py -q
+
+.PHONY: test-system
+test-system:  ## end-to-end Model+Server -> completion request (CPU)
+	$(PY) -m pytest tests/test_system_local.py -q
+
+.PHONY: test-gpu
+test-gpu:  ## kernel parity + engine tests (run on MI355X)
+	$(PY) -m pytest tests/ -q -m gpu
+
+.PHONY: bench
+bench: build
+	$(PY) bench.py
+
+.PHONY: bench-train
+bench-train: build
+	$(PY) bench.py --mode train
+
+.PHONY: crds
+crds:  ## regenerate config/crd/bases
+	$(PY) scripts/gen_crds.py
+
+.PHONY: up
+up:  ## start the local platform daemon
+	$(PY) -m substratus_amd.kube.apiserver
