@@ -41,6 +41,9 @@ class Manager:
         self.queue: deque = deque()
         self._queued: set = set()
         self._delayed: list = []   # (due_ts, key)
+        # observability (reference: controller-runtime prometheus metrics)
+        self.metrics = {"reconcile_total": {}, "reconcile_errors_total": {},
+                        "requeue_total": 0}
         for kind in list(self.reconcilers) + ["Job", "Pod", "Deployment"]:
             kube.watch(kind, self._on_event)
 
@@ -100,9 +103,17 @@ class Manager:
             self._persist_status(kind, name, ns, obj)
             return
         rec = self.reconcilers[kind]
-        result = rec.reconcile(obj)
+        self.metrics["reconcile_total"][kind] = \
+            self.metrics["reconcile_total"].get(kind, 0) + 1
+        try:
+            result = rec.reconcile(obj)
+        except Exception:
+            self.metrics["reconcile_errors_total"][kind] = \
+                self.metrics["reconcile_errors_total"].get(kind, 0) + 1
+            raise
         self._persist(kind, name, ns, obj)
         if not result.success and result.requeue_after > 0:
+            self.metrics["requeue_total"] += 1
             self._delayed.append((time.time() + result.requeue_after, key))
 
     def _persist(self, kind: str, name: str, ns: str, obj: dict) -> None:

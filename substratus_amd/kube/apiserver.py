@@ -25,13 +25,35 @@ from ..api import types as apiv1
 from .memapi import Conflict, MemKube, NotFound
 
 
-def create_api(kube: MemKube) -> FastAPI:
+def create_api(kube: MemKube, platform: Optional["Platform"] = None) -> FastAPI:
     app = FastAPI(title="substratus-amd apiserver")
     app.state.kube = kube
 
     @app.get("/healthz")
     def healthz():
         return {"ok": True}
+
+    @app.get("/readyz")
+    def readyz():
+        return {"ok": True}
+
+    @app.get("/metrics")
+    def metrics():
+        # controller metrics in Prometheus exposition format (reference:
+        # controller-runtime metrics behind kube-rbac-proxy)
+        from fastapi import Response
+        lines = []
+        if platform is not None:
+            m = platform.mgr.metrics
+            for kind, n in m["reconcile_total"].items():
+                lines.append(
+                    f'substratus_reconcile_total{{kind="{kind}"}} {n}')
+            for kind, n in m["reconcile_errors_total"].items():
+                lines.append(
+                    f'substratus_reconcile_errors_total{{kind="{kind}"}} {n}')
+            lines.append(f'substratus_requeue_total {m["requeue_total"]}')
+        lines.append(f"substratus_objects {len(kube.list('Model', None)) + len(kube.list('Server', None)) + len(kube.list('Dataset', None)) + len(kube.list('Notebook', None))}")
+        return Response("\n".join(lines) + "\n", media_type="text/plain")
 
     @app.post("/api/objects")
     async def create(req: Request):
@@ -136,7 +158,7 @@ def main() -> None:
     data_dir = os.environ.get("SUBSTRATUS_DATA",
                               os.path.expanduser("~/.substratus"))
     platform = Platform(data_dir)
-    app = create_api(platform.kube)
+    app = create_api(platform.kube, platform)
     port = int(os.environ.get("SUBSTRATUS_API_PORT", "8070"))
     uvicorn.run(app, host="127.0.0.1", port=port, log_level="warning")
 
