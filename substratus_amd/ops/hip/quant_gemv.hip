@@ -36,31 +36,30 @@ void quant_gemv_kernel(bf16_t* __restrict__ y,         // [M, N]
   for (int m = 0; m < M; ++m) acc[m] = 0.f;
 
   if (!Q8) {
-    // 2-deep software pipeline: block b+WAVE's qs/scale/x loads are issued
-    // before block b's dequant-dot (same issue-early pattern that took the
-    // decode attention kernel from 4.2 to 6.1 TB/s); the per-block fp16
-    // scale load otherwise heads a dependent latency chain.
+    // 2-deep software pipeline: block b+WAVE's qs/scale loads are issued
+    // before block b's dequant-dot. Tail handling is BRANCHLESS: indices
+    // clamp to the last block and a zero scale masks the redundant work —
+    // a per-lane `if (b < nblocks)` guard around the loads makes hipcc
+    // branch and drain vmcnt per iteration (guide §5 trap 4(c)), which
+    // measured SLOWER than no pipeline at all.
     const uint4* qrow = reinterpret_cast<const uint4*>(qs + (long)row * k / 2);
-    uint4 nib_c;
-    float s_c;
+    const int last = nblocks - 1;
     int b = lane;
-    if (b < nblocks) {
-      nib_c = qrow[b];
-      s_c = (float)srow[b];
-    }
-    for (; b < nblocks; b += WAVE) {
-      const int bn = b + WAVE;
-      uint4 nib_n;
-      float s_n = 0.f;
-      if (bn < nblocks) {
-        nib_n = qrow[bn];
-        s_n = (float)srow[bn];
-      }
+    uint4 nib_c = qrow[min(b, last)];
+    float sv0 = (float)srow[min(b, last)];
+    float s_c = (b <= last) ? sv0 : 0.f;
+    const int iters = (nblocks + WAVE - 1) / WAVE;
+    for (int it = 0; it < iters; ++it, b += WAVE) {
+      const int bn = min(b + WAVE, last);
+      const uint4 nib_n = qrow[bn];
+      const float svn = (float)srow[bn];  // load always, select after
+      const float s_n = (b + WAVE <= last) ? svn : 0.f;
+      const int bcl = min(b, last);       // clamp x address for masked lanes
       const unsigned int w[4] = {nib_c.x, nib_c.y, nib_c.z, nib_c.w};
       #pragma unroll
       for (int m = 0; m < M; ++m) {
         const uint4* xb =
-            reinterpret_cast<const uint4*>(x + (long)m * k + b * 32);
+            reinterpret_cast<const uint4*>(x + (long)m * k + bcl * 32);
         float partial = 0.f;
         #pragma unroll
         for (int c = 0; c < 4; ++c) {   // 4 bytes -> weights i..i+3, i+16..19
