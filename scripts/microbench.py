@@ -76,6 +76,21 @@ def bench_sample(B=64, V=32000):
     print(f"softmax_sample greedy B={B} V={V}: {dt*1e6:.1f} us")
 
 
+def bench_quant_gemv(n=13824, k=5120, quant="q4_0"):
+    from substratus_amd.parallel.tp import Linear
+    from substratus_amd.models.quant import QuantLinear
+    torch.manual_seed(0)
+    lin = Linear(k, n, bias=False, dtype=torch.float32)
+    lin.weight.data.normal_(0, 0.1)
+    ql = QuantLinear.from_linear(lin, quant).to("cuda")
+    for m in (1, 4):
+        x = torch.randn(m, k, device="cuda").bfloat16()
+        dt = timeit(lambda: ql(x))
+        wb = ql.qs.numel() + ql.scales.numel() * 2
+        print(f"quant_gemv {quant} N={n} K={k} M={m}: {dt*1e6:.1f} us, "
+              f"{wb/dt/1e12:.2f} TB/s (weights)")
+
+
 if __name__ == "__main__":
     ap = argparse.ArgumentParser()
     ap.add_argument("--which", default="all")
@@ -90,3 +105,6 @@ if __name__ == "__main__":
         bench_norm()
     if a.which in ("all", "sample"):
         bench_sample()
+    if a.which in ("all", "quant"):
+        bench_quant_gemv()
+        bench_quant_gemv(quant="q8_0")
