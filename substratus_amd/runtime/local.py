@@ -210,6 +210,15 @@ class LocalRuntime:
             return [sys.executable, "-m", "substratus_amd.serve.server"]
         return None
 
+    @staticmethod
+    def _gpu_count(container: dict) -> int:
+        res = container.get("resources", {}) or {}
+        lim = res.get("limits", {}) or {}
+        try:
+            return int(lim.get("amd.com/gpu", 0))
+        except (TypeError, ValueError):
+            return 0
+
     def _exec_container(self, container: dict, env: dict,
                         mounts: dict) -> int:
         cmd = self._entrypoint(container.get("image", ""))
@@ -217,6 +226,13 @@ class LocalRuntime:
             # unknown image (e.g. kaniko builder): trivially succeed — there
             # is no container runtime to build real images locally
             return 0
+        ngpu = self._gpu_count(container)
+        if ngpu > 1 and cmd[1] == "-m":
+            # multi-GPU Job: one rank per GPU over RCCL (container contract
+            # "Multi-GPU (MI355X)" — the trainer entrypoint is DDP-aware)
+            cmd = [sys.executable, "-m", "torch.distributed.run",
+                   "--standalone", "--local-addr", "127.0.0.1",
+                   f"--nproc-per-node={ngpu}", "-m", cmd[2]]
         args = []
         if "/content/artifacts" in mounts and "loader" in cmd[-1]:
             args = [mounts["/content/artifacts"]]
