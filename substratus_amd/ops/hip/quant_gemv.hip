@@ -56,16 +56,24 @@ void quant_gemv_kernel(bf16_t* __restrict__ y,         // [M, N]
       const float s_n = (b + WAVE <= last) ? svn : 0.f;
       const int bcl = min(b, last);       // clamp x address for masked lanes
       const unsigned int w[4] = {nib_c.x, nib_c.y, nib_c.z, nib_c.w};
+      // issue every row's x loads before any compute (per-m inline loads
+      // serialize the M>1 case: 0.94 TB/s vs 2.34 at M=1)
+      uint4 xall[M][4];
       #pragma unroll
       for (int m = 0; m < M; ++m) {
         const uint4* xb =
             reinterpret_cast<const uint4*>(x + (long)m * k + bcl * 32);
+        #pragma unroll
+        for (int c = 0; c < 4; ++c) xall[m][c] = xb[c];
+      }
+      #pragma unroll
+      for (int m = 0; m < M; ++m) {
         float partial = 0.f;
         #pragma unroll
         for (int c = 0; c < 4; ++c) {   // 4 bytes -> weights i..i+3, i+16..19
           unsigned int word = w[c];
-          const uint4 xlo4 = xb[c >= 2 ? 1 : 0];
-          const uint4 xhi4 = xb[c >= 2 ? 3 : 2];
+          const uint4 xlo4 = xall[m][c >= 2 ? 1 : 0];
+          const uint4 xhi4 = xall[m][c >= 2 ? 3 : 2];
           const unsigned int xlo[2] = {c % 2 == 0 ? xlo4.x : xlo4.z,
                                        c % 2 == 0 ? xlo4.y : xlo4.w};
           const unsigned int xhi[2] = {c % 2 == 0 ? xhi4.x : xhi4.z,
@@ -94,14 +102,20 @@ void quant_gemv_kernel(bf16_t* __restrict__ y,         // [M, N]
       const uint4 q0 = qrow[b * 2], q1 = qrow[b * 2 + 1];
       const unsigned int w[8] = {q0.x, q0.y, q0.z, q0.w,
                                  q1.x, q1.y, q1.z, q1.w};
+      uint4 xall8[M][4];
       #pragma unroll
       for (int m = 0; m < M; ++m) {
         const uint4* xb =
             reinterpret_cast<const uint4*>(x + (long)m * k + b * 32);
+        #pragma unroll
+        for (int c = 0; c < 4; ++c) xall8[m][c] = xb[c];
+      }
+      #pragma unroll
+      for (int m = 0; m < M; ++m) {
         float partial = 0.f;
         #pragma unroll
         for (int c = 0; c < 4; ++c) {
-          const uint4 xv = xb[c];
+          const uint4 xv = xall8[m][c];
           const unsigned int xs[4] = {xv.x, xv.y, xv.z, xv.w};
           #pragma unroll
           for (int j = 0; j < 2; ++j) {  // 2 words of 4 int8 per uint4-pair
