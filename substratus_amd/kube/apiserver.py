@@ -108,11 +108,12 @@ def create_api(kube: MemKube, platform: Optional["Platform"] = None) -> FastAPI:
 class Platform:
     """Embedded control plane: MemKube + controllers + local runtime + SCI."""
 
-    def __init__(self, data_dir: str):
+    def __init__(self, data_dir: str, sci_address: Optional[str] = None):
         from ..cloud import new_cloud
         from ..controller.manager import Manager
         from ..runtime.local import LocalRuntime
         from ..sci.kind_server import KindSCIService
+        self.sci_address = sci_address
         os.makedirs(data_dir, exist_ok=True)
         self.bucket_dir = os.path.join(data_dir, "bucket")
         os.makedirs(self.bucket_dir, exist_ok=True)
@@ -128,6 +129,7 @@ class Platform:
             url_host=f"http://127.0.0.1:{self.sci_http_port}", block=False)
         from ..sci.client import GrpcSCIClient
         sci_client = GrpcSCIClient(
+            self.sci_address or
             f"localhost:{os.environ.get('SCI_GRPC_PORT', '10080')}")
         self.mgr = Manager(self.kube, self.cloud, sci_client)
         self.runtime = LocalRuntime(self.kube, self.bucket_dir)
@@ -154,10 +156,38 @@ class Platform:
 
 
 def main() -> None:
+    import argparse
+    import json
+    import sys
     import uvicorn
-    data_dir = os.environ.get("SUBSTRATUS_DATA",
-                              os.path.expanduser("~/.substratus"))
-    platform = Platform(data_dir)
+    # flags mirror the reference controller-manager's
+    # (cmd/controllermanager/main.go:41-58)
+    ap = argparse.ArgumentParser("substratus platform daemon")
+    ap.add_argument("--leader-elect", action="store_true",
+                    help="refuse to start if another daemon holds the lock")
+    ap.add_argument("--sci-address", default=None,
+                    help="external SCI gRPC address (default: embedded kind SCI)")
+    ap.add_argument("--metrics-bind-address", default=None)
+    ap.add_argument("--config-dump-path", default=None,
+                    help="dump resolved config as JSON and continue")
+    ap.add_argument("--data-dir", default=os.environ.get(
+        "SUBSTRATUS_DATA", os.path.expanduser("~/.substratus")))
+    args = ap.parse_args()
+    if args.leader_elect:
+        from ..controller.leader import FileLeaderLock
+        lock = FileLeaderLock(os.path.join(args.data_dir, "leader.lock"))
+        if not lock.acquire():
+            print("another daemon holds the leader lock; exiting",
+                  file=sys.stderr)
+            raise SystemExit(1)
+    platform = Platform(args.data_dir, sci_address=args.sci_address)
+    if args.config_dump_path:
+        with open(args.config_dump_path, "w") as f:
+            json.dump({"cloud": platform.cloud.name(),
+                       "cluster_name": platform.cloud.cluster_name,
+                       "artifact_bucket_url": platform.cloud.artifact_bucket_url,
+                       "registry_url": platform.cloud.registry_url,
+                       "data_dir": args.data_dir}, f, indent=2)
     app = create_api(platform.kube, platform)
     port = int(os.environ.get("SUBSTRATUS_API_PORT", "8070"))
     uvicorn.run(app, host="127.0.0.1", port=port, log_level="warning")
