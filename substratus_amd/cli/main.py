@@ -192,10 +192,23 @@ def apply(file: str = typer.Option(..., "-f", "--filename")):
 
 
 @app.command()
-def get(scope: str = typer.Argument("all")):
+def get(scope: str = typer.Argument("all"),
+        watch: bool = typer.Option(False, "--watch", "-w",
+                                   help="refresh until interrupted")):
     """List objects: `sub get models`, `sub get models/name`, `sub get all`
-    (reference internal/cli/get.go; scope grammar tui/get.go:219-287)."""
+    (reference internal/cli/get.go; scope grammar tui/get.go:219-287;
+    --watch mirrors the reference's live watch-driven listing)."""
     c = _client()
+    if watch:
+        from rich.live import Live
+        with Live(_get_table(c, scope), refresh_per_second=2) as live:
+            while True:
+                time.sleep(1)
+                live.update(_get_table(c, scope))
+    console.print(_get_table(c, scope))
+
+
+def _get_table(c, scope: str):
     kind_filter, name_filter = None, None
     if scope != "all":
         parts = scope.split("/")
@@ -217,7 +230,7 @@ def get(scope: str = typer.Argument("all")):
             last = (f"{conds[-1].get('type')}={conds[-1].get('status')}"
                     f" ({conds[-1].get('reason', '')})") if conds else ""
             table.add_row(kind, name, ready, last)
-    console.print(table)
+    return table
 
 
 @app.command()

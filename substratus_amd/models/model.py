@@ -152,13 +152,11 @@ class Block(nn.Module):
                         v_cache: torch.Tensor, slots: torch.Tensor) -> torch.Tensor:
         y = self.norm1(h, self.eps)
         q, k, v = self._split_qkv(self.wqkv(y))
+        # rope_kv_append rotates q AND k in place (k also lands rotated in
+        # the paged cache); the dense varlen attention consumes them directly
+        k = k.contiguous()
         q = ops.rope_kv_append(q, k, v, positions, k_cache, v_cache, slots,
                                self.rope_theta)
-        # read k,v back rotated from what we just wrote is unnecessary —
-        # recompute rotated k locally for the dense varlen attention.
-        if self.rope_theta > 0:
-            from ..ops import ref as _ref
-            k = _ref.apply_rope(k, positions, self.rope_theta)
         a = ops.attn_prefill(q, k, v, seq_start, self.scale)
         attn_out = self.wo(a.reshape(a.shape[0], -1))
         if self.parallel_attn:

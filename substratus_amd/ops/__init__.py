@@ -93,7 +93,12 @@ def rope_kv_append(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
     if ext is None:
         return ref.rope_kv_append(q, k, v, positions, k_cache, v_cache, slots, theta)
     q = q.contiguous()
-    ext.rope_kv_append(q, k.contiguous(), v.contiguous(),
+    if not k.is_contiguous():
+        # decode passes a qkv-split view; the in-place k rotation lands in a
+        # private copy there (decode never reads k again). Prefill passes a
+        # contiguous k and consumes the rotation.
+        k = k.contiguous()
+    ext.rope_kv_append(q, k, v.contiguous(),
                        positions, k_cache, v_cache, slots, theta)
     return q
 

@@ -63,6 +63,8 @@ void rope_kv_append_kernel(
     const bool is_q = role < hq;
     const bf16_t* src;
     bf16_t* dst;
+    bf16_t* dst2 = nullptr;   // k also writes back in place (prefill reads
+                              // the rotated k for its dense attention)
     if (is_q) {
       src = q + ((long)tok * hq + role) * d;
       dst = const_cast<bf16_t*>(src);
@@ -70,6 +72,7 @@ void rope_kv_append_kernel(
       const int h = role - hq;
       src = k + ((long)tok * hkv + h) * d;
       dst = k_cache + cache_row + (long)h * block_size * d;
+      dst2 = const_cast<bf16_t*>(src);
     }
     if (!do_rope) {
       if (!is_q) {
@@ -82,8 +85,14 @@ void rope_kv_append_kernel(
     if (lane < half) {
       const float x1 = bf2f(src[lane]);
       const float x2 = bf2f(src[lane + half]);
-      dst[lane] = f2bf(x1 * c - x2 * s);
-      dst[lane + half] = f2bf(x2 * c + x1 * s);
+      const bf16_t o1 = f2bf(x1 * c - x2 * s);
+      const bf16_t o2 = f2bf(x2 * c + x1 * s);
+      dst[lane] = o1;
+      dst[lane + half] = o2;
+      if (dst2 != nullptr) {
+        dst2[lane] = o1;
+        dst2[lane + half] = o2;
+      }
     }
   }
 }

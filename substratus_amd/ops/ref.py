@@ -67,6 +67,7 @@ def rope_kv_append(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
     """
     q_out = apply_rope(q, positions, theta)
     k_rot = apply_rope(k, positions, theta)
+    k.copy_(k_rot)  # k is rotated IN PLACE too (prefill consumes it dense)
     num_blocks, hkv, block_size, d = k_cache.shape
     blk = torch.div(slots, block_size, rounding_mode="floor").long()
     off = (slots % block_size).long()
