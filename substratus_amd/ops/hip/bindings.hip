@@ -55,6 +55,7 @@ void rms_norm(torch::Tensor out, torch::Tensor x, torch::Tensor w, double eps) {
   check_bf16(out, "out"); check_bf16(x, "x"); check_bf16(w, "w");
   int hidden = x.size(-1);
   TORCH_CHECK(hidden % 8 == 0, "hidden must be a multiple of 8");
+  TORCH_CHECK(hidden <= 16384, "hidden > 16384 unsupported (register cache)");
   long rows = x.numel() / hidden;
   launch_rms_norm(bf(out), bfc(x), bfc(w), rows, hidden, (float)eps, stream());
   post_launch_check("rms_norm");
@@ -66,6 +67,7 @@ void rms_norm_residual(torch::Tensor out, torch::Tensor residual,
   check_bf16(x, "x"); check_bf16(w, "w");
   int hidden = x.size(-1);
   TORCH_CHECK(hidden % 8 == 0, "hidden must be a multiple of 8");
+  TORCH_CHECK(hidden <= 16384, "hidden > 16384 unsupported (register cache)");
   long rows = x.numel() / hidden;
   launch_rms_norm_residual(bf(out), bf(residual), bfc(x), bfc(w), rows,
                            hidden, (float)eps, stream());

@@ -12,12 +12,18 @@
 #include "common.h"
 
 // Each thread handles H/BLOCK elements; vector path requires H % 8 == 0.
+// The row is cached in REGISTERS between the reduce and the scale pass when
+// it fits (hidden <= BLOCK*8*RCAP — every production config: h <= 8192 at
+// BLOCK=256 needs 4 uint4/thread), so each norm makes ONE read pass over
+// the hidden state instead of two (the decode chain runs 2 norms/layer at
+// 1 WG/CU where the second pass is pure latency).
 template <int BLOCK, bool RESIDUAL>
 __global__ void rms_norm_kernel(bf16_t* __restrict__ out,
                                 bf16_t* __restrict__ residual,
                                 const bf16_t* __restrict__ x,
                                 const bf16_t* __restrict__ weight,
                                 int hidden, float eps) {
+  constexpr int RCAP = 8;  // max uint4 per thread held in registers
   __shared__ float scratch[BLOCK / WAVE];
   const long row = blockIdx.x;
   const bf16_t* xr = x + row * hidden;
@@ -25,12 +31,16 @@ __global__ void rms_norm_kernel(bf16_t* __restrict__ out,
   bf16_t* orow = out + row * hidden;
 
   float ss = 0.f;
-  // vectorized: 8 bf16 per iteration per thread
   const int nvec = hidden / 8;
   const uint4* xv = reinterpret_cast<const uint4*>(xr);
   uint4* rv = RESIDUAL ? reinterpret_cast<uint4*>(rr) : nullptr;
-  // cache the row in registers when it fits (hidden <= 8*BLOCK*REG_CAP)
-  for (int i = threadIdx.x; i < nvec; i += BLOCK) {
+  // statically-indexed register cache (a runtime-indexed array goes to
+  // scratch — guide §5.4 rule 20); launcher guarantees nvec <= BLOCK*RCAP
+  uint4 keep[RCAP];
+  #pragma unroll
+  for (int kk = 0; kk < RCAP; ++kk) {
+    const int i = threadIdx.x + kk * BLOCK;
+    if (i >= nvec) break;
     uint4 u = xv[i];
     float2 f0 = unpack_bf16x2(u.x), f1 = unpack_bf16x2(u.y);
     float2 f2 = unpack_bf16x2(u.z), f3 = unpack_bf16x2(u.w);
@@ -44,18 +54,22 @@ __global__ void rms_norm_kernel(bf16_t* __restrict__ out,
       w.x = pack_bf16x2(f0.x, f0.y); w.y = pack_bf16x2(f1.x, f1.y);
       w.z = pack_bf16x2(f2.x, f2.y); w.w = pack_bf16x2(f3.x, f3.y);
       rv[i] = w;  // residual stream updated in place
+      u = w;
     }
+    keep[kk] = u;
     ss += f0.x * f0.x + f0.y * f0.y + f1.x * f1.x + f1.y * f1.y +
           f2.x * f2.x + f2.y * f2.y + f3.x * f3.x + f3.y * f3.y;
   }
   float total = block_sum<BLOCK / WAVE>(ss, scratch);
   float inv = rsqrtf(total / (float)hidden + eps);
 
-  const uint4* src = RESIDUAL ? reinterpret_cast<const uint4*>(rr) : xv;
   const uint4* wv = reinterpret_cast<const uint4*>(weight);
   uint4* ov = reinterpret_cast<uint4*>(orow);
-  for (int i = threadIdx.x; i < nvec; i += BLOCK) {
-    uint4 u = src[i];
+  #pragma unroll
+  for (int kk = 0; kk < RCAP; ++kk) {
+    const int i = threadIdx.x + kk * BLOCK;
+    if (i >= nvec) break;
+    uint4 u = keep[kk];
     uint4 w = wv[i];
     float2 f0 = unpack_bf16x2(u.x), f1 = unpack_bf16x2(u.y);
     float2 f2 = unpack_bf16x2(u.z), f3 = unpack_bf16x2(u.w);
@@ -123,6 +137,7 @@ __global__ void layer_norm_kernel(bf16_t* __restrict__ out,
 void launch_rms_norm(bf16_t* out, const bf16_t* x, const bf16_t* w,
                      long rows, int hidden, float eps, hipStream_t stream) {
   constexpr int BLOCK = 256;
+  // register cache bound: hidden <= BLOCK*8*RCAP = 16384 (all configs)
  hipLaunchKernelGGL(( rms_norm_kernel<BLOCK, false>), dim3(dim3(rows)), dim3(dim3(BLOCK)), 0, stream, 
       out, nullptr, x, w, hidden, eps);
 }
