@@ -244,6 +244,20 @@ def delete(target: str):
 
 
 @app.command()
+def logs(name: str, tail: int = typer.Option(100, "--tail"),
+         namespace: str = typer.Option("default", "-n")):
+    """Logs of an object's workload: `sub logs myserver-server` /
+    `sub logs mymodel-modeller` (reference: pod log streaming in the TUI)."""
+    c = _client()
+    r = c.http.get(f"/api/logs/{namespace}/{name}", params={"tail": tail})
+    if r.status_code == 404:
+        console.print(f"[red]no logs for {name}[/red]")
+        raise typer.Exit(1)
+    for line in r.json()["lines"]:
+        console.print(line, highlight=False)
+
+
+@app.command()
 def infer(server: str = typer.Option(..., "--server", "-s"),
           prompt: str = typer.Argument(...),
           max_tokens: int = typer.Option(32)):

@@ -94,6 +94,19 @@ def create_api(kube: MemKube, platform: Optional["Platform"] = None) -> FastAPI:
         except NotFound:
             raise HTTPException(404, "not found")
 
+    @app.get("/api/logs/{namespace}/{name}")
+    def logs(namespace: str, name: str, tail: int = 200):
+        # pod-log streaming equivalent (the reference TUI streams the
+        # default container's logs — internal/tui/pods.go:222-246)
+        if platform is None:
+            raise HTTPException(404, "no runtime")
+        path = platform.runtime.log_path(namespace, name)
+        if not os.path.exists(path):
+            raise HTTPException(404, f"no logs for {namespace}/{name}")
+        with open(path, "rb") as f:
+            lines = f.read().decode(errors="replace").splitlines()
+        return {"lines": lines[-tail:]}
+
     @app.delete("/api/objects/{kind}/{namespace}/{name}")
     def delete_obj(kind: str, namespace: str, name: str):
         try:

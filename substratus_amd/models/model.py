@@ -243,6 +243,7 @@ class CausalLM(nn.Module):
             self.lm_head = Linear(cfg.hidden_size, cfg.vocab_size, bias=False,
                                   device=device, dtype=dtype)
         self.n_kv_heads = self.blocks[0].n_kv_heads  # per-rank
+        self.gradient_checkpointing = False
 
     def _logits(self, y: torch.Tensor) -> torch.Tensor:
         if self.lm_head is None:
@@ -313,7 +314,12 @@ class CausalLM(nn.Module):
         else:
             cos = sin = None
         for blk in self.blocks:
-            h = blk.forward_train(h, cos, sin)
+            if self.gradient_checkpointing and h.requires_grad is not False \
+                    and torch.is_grad_enabled():
+                h = torch.utils.checkpoint.checkpoint(
+                    blk.forward_train, h, cos, sin, use_reentrant=False)
+            else:
+                h = blk.forward_train(h, cos, sin)
         if self.final_norm.rms:
             h = F.rms_norm(h.float(), (self.cfg.hidden_size,),
                            self.final_norm.weight.float(), self.eps).to(h.dtype)

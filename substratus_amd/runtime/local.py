@@ -97,6 +97,7 @@ class LocalRuntime:
             pod_spec = job["spec"]["template"]["spec"]
             c = pod_spec["containers"][0]
             env, mounts = self._resolve(pod_spec, c, ns)
+            env["JOB_NAME"] = name
             rc = self._exec_container(c, env, mounts)
             cond = {"type": "Complete" if rc == 0 else "Failed",
                     "status": "True"}
@@ -127,7 +128,7 @@ class LocalRuntime:
         port = _free_port()
         env["PORT"] = str(port)
         self.server_ports[f"{ns}/{name}"] = port
-        proc = self._spawn_server(c, env, mounts)
+        proc = self._spawn_server(c, env, mounts, ns, name)
         self.procs[key] = proc
         # record the local endpoint on the Service for clients/tests
         svc = self.kube.try_get("Service", name, ns)
@@ -238,17 +239,27 @@ class LocalRuntime:
             args = [mounts["/content/artifacts"]]
         proc = subprocess.run(cmd + args, env=env, cwd=_repo_root(),
                               capture_output=True, timeout=600)
+        with open(self.log_path("default",
+                                env.get("JOB_NAME", "job")), "ab") as lf:
+            lf.write(proc.stdout)
+            lf.write(proc.stderr)
         if proc.returncode != 0:
             sys.stderr.write(proc.stderr.decode()[-2000:])
         return proc.returncode
 
+    def log_path(self, ns: str, name: str) -> str:
+        d = os.path.join(self.bucket_dir, "_logs")
+        os.makedirs(d, exist_ok=True)
+        return os.path.join(d, f"{ns}-{name}.log")
+
     def _spawn_server(self, container: dict, env: dict,
-                      mounts: dict) -> subprocess.Popen:
+                      mounts: dict, ns: str = "default",
+                      name: str = "server") -> subprocess.Popen:
         cmd = self._entrypoint(container.get("image", "")) or [
             sys.executable, "-m", "substratus_amd.serve.server"]
+        lf = open(self.log_path(ns, name), "ab")
         return subprocess.Popen(cmd, env=env, cwd=_repo_root(),
-                                stdout=subprocess.DEVNULL,
-                                stderr=subprocess.PIPE)
+                                stdout=lf, stderr=lf)
 
     def stop(self) -> None:
         for proc in self.procs.values():

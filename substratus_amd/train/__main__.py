@@ -40,6 +40,8 @@ def main() -> None:
         seq_len=int(params.get("seq_len", params.get("max_length", 2048))),
         num_train_epochs=int(params.get("num_train_epochs", 1)),
         save_steps=int(params.get("save_steps", 0)),
+        gradient_checkpointing=bool(int(params.get(
+            "gradient_checkpointing", 0))),
         artifacts_dir=artifacts,
         dtype=torch.bfloat16 if on_gpu else torch.float32,
     )

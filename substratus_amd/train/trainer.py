@@ -48,6 +48,7 @@ class TrainConfig:
     save_steps: int = 0         # 0 = only final checkpoint
     artifacts_dir: str = "/content/artifacts"
     resume: bool = True
+    gradient_checkpointing: bool = False   # recompute activations in backward
     seed: int = 0
     dtype: torch.dtype = torch.bfloat16
     grad_clip: float = 1.0
@@ -103,6 +104,7 @@ class Trainer:
             model = model.to_empty(device=self.device)
             model.init_random(cfg.seed)  # all DP ranks: same seed = same init
         self.model = model.train()
+        self.model.gradient_checkpointing = cfg.gradient_checkpointing
         params = [p for p in self.model.parameters()]
         for p in params:
             p.requires_grad_(True)

@@ -154,3 +154,17 @@ def test_ddp_two_ranks_match_single(tmp_path):
         # vs single-process batch mean): tolerance covers that, not a bug
         torch.testing.assert_close(results[0][k], ref[k], atol=1e-4,
                                    rtol=5e-3)
+
+
+def test_gradient_checkpointing_matches(tmp_path):
+    """Activation checkpointing must not change gradients."""
+    torch.manual_seed(7)
+    t1 = make_trainer(tmp_path)
+    t2 = make_trainer(tmp_path, gradient_checkpointing=True)
+    t2.model.load_state_dict(t1.model.state_dict())
+    g = torch.Generator().manual_seed(51)
+    batch = torch.randint(0, 512, (2, 33), generator=g)
+    t1.train_step(batch.clone())
+    t2.train_step(batch.clone())
+    for p1, p2 in zip(t1.model.parameters(), t2.model.parameters()):
+        torch.testing.assert_close(p1.data, p2.data, atol=1e-6, rtol=1e-5)
