@@ -39,6 +39,9 @@ def parse_args():
     p.add_argument("--batch", type=int, default=256)
     p.add_argument("--prompt-len", type=int, default=512)
     p.add_argument("--seq-len", type=int, default=None)
+    p.add_argument("--kv-dtype", default=None, choices=[None, "fp8"],
+                   help="KV-cache dtype override (fp8 e4m3; reduced-precision"
+                        " secondary config, NOT the headline)")
     p.add_argument("--quant", default=None,
                    help="GGUF weight quantization: q4_0 | q8_0 (config 4)")
     p.add_argument("--tp", type=int, default=1,
@@ -68,6 +71,7 @@ def bench_serve(args) -> None:
                        max_seq_len=max_seq,
                        prefill_chunk=args.batch * args.prompt_len,
                        quant=args.quant,
+                       kv_dtype=args.kv_dtype,
                        seed=1234 + st.dp_rank)
     log(f"[bench] building {model_name} engine on {device} ...")
     t_build = time.time()
@@ -138,6 +142,7 @@ def bench_serve(args) -> None:
             "parallelism": parallelism,
             "p50_decode_latency_ms": round(ms_per_step, 3),
             **({"weights": args.quant} if args.quant else {}),
+            **({"kv_cache": "fp8_e4m3"} if args.kv_dtype == "fp8" else {}),
         },
     }
     if st.rank == 0:

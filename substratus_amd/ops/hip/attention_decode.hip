@@ -5,19 +5,20 @@
 // Decode at batch B is HBM-bound on the K/V stream (e.g. llama2-7b @ctx 512,
 // B=64: ~16 GB of KV per step vs 13.5 GB of weights). Structure:
 //
-//   * grid = (num_seqs, n_kv_heads, ceil(group/GC)); 4 waves per workgroup;
-//     each wave owns KV blocks stride-4 (flash-decoding split over the
-//     sequence).
-//   * K panel row [16, D] bf16: D/8 lanes × uint4 = one full 256 B row per
-//     token. K tiles are PREFETCHED one block ahead into registers (guide
-//     T14 issue-early/consume-late: the next block's loads are issued before
-//     this block's score+PV compute, hiding the ~900-cycle HBM latency), so
-//     the score phase runs entirely from registers.
-//   * Scores reduced via shfl within the lane group (guide §5
-//     common-mistake #6: wave-parallel softmax, never serial).
-//   * Online softmax per wave; P·V with dims distributed over lanes; for
-//     full blocks all 16 V row-loads are issued before the FMA chain.
-//   * Cross-wave merge through LDS at the end.
+//   * grid = (num_seqs, n_kv_heads, ceil(group/GC) × nsplit); 4 waves per
+//     workgroup; each wave owns KV blocks stride NW×nsplit (flash-decoding
+//     split over the sequence, combined by a reduce kernel when nsplit > 1).
+//   * K panel row [16, D]: D/8 lanes per token (uint4 of bf16, or uint2 of
+//     fp8 — same fragment map at half the bytes). K tiles are PREFETCHED one
+//     block ahead into registers (guide T14 issue-early/consume-late), so
+//     the score phase runs entirely from registers; nt loads keep the
+//     read-once stream out of L1.
+//   * Scores reduced via shfl within the lane group; online softmax per
+//     wave; P·V with dims distributed over lanes; cross-wave merge via LDS.
+//
+// KV8 (opt-in fp8 e4m3 cache): gfx950's v_cvt_f32_fp8 decodes inline —
+// halves the dominant KV stream at a documented precision cost (the cache is
+// written by rope_kv_append's fp8 path; scale 1.0).
 //
 // Cache layout: [num_blocks, n_kv_heads, BS=16, D]; block_table [B, maxb]
 // int32; seq_lens [B] int32 counts INCLUDING the current token.
@@ -26,20 +27,79 @@
 #define BS 16   // tokens per KV block
 #define NW 4    // waves per workgroup
 
-template <int D, int GC>
+template <bool KV8>
+struct KTile {  // one lane's 8 dims of one token row
+  uint4v u;
+  DEV_INLINE void load(const void* base, long off_elems) {
+    u = nt_load_u4(reinterpret_cast<const bf16_t*>(base) + off_elems);
+  }
+  DEV_INLINE void unpack(float f[8]) const {
+    float2 f0 = unpack_bf16x2(u[0]), f1 = unpack_bf16x2(u[1]);
+    float2 f2 = unpack_bf16x2(u[2]), f3 = unpack_bf16x2(u[3]);
+    f[0] = f0.x; f[1] = f0.y; f[2] = f1.x; f[3] = f1.y;
+    f[4] = f2.x; f[5] = f2.y; f[6] = f3.x; f[7] = f3.y;
+  }
+};
+
+template <>
+struct KTile<true> {
+  uint2v u;
+  DEV_INLINE void load(const void* base, long off_elems) {
+    u = nt_load_u2(reinterpret_cast<const unsigned char*>(base) + off_elems);
+  }
+  DEV_INLINE void unpack(float f[8]) const {
+    f[0] = fp8_to_f32<0>(u[0]); f[1] = fp8_to_f32<1>(u[0]);
+    f[2] = fp8_to_f32<2>(u[0]); f[3] = fp8_to_f32<3>(u[0]);
+    f[4] = fp8_to_f32<0>(u[1]); f[5] = fp8_to_f32<1>(u[1]);
+    f[6] = fp8_to_f32<2>(u[1]); f[7] = fp8_to_f32<3>(u[1]);
+  }
+};
+
+// one lane's DPL V dims of one token (vbase is a BYTE pointer at the block's
+// head panel; t rows of d_total elements)
+template <bool KV8, int DPL>
+DEV_INLINE void load_v(const char* vbase, int t, int d_total, int lane,
+                       float out[DPL]) {
+  if (!KV8) {
+    const bf16_t* p = reinterpret_cast<const bf16_t*>(vbase) +
+                      (long)t * d_total;
+    if (DPL == 2) {
+      const unsigned int uv = __builtin_nontemporal_load(
+          reinterpret_cast<const unsigned int*>(p) + lane);
+      float2 vf = unpack_bf16x2(uv);
+      out[0] = vf.x;
+      out[DPL - 1] = vf.y;
+    } else {
+      out[0] = bf2f(p[lane]);
+    }
+  } else {
+    const unsigned char* p = reinterpret_cast<const unsigned char*>(vbase) +
+                             (long)t * d_total;
+    if (DPL == 2) {
+      const unsigned short us = *reinterpret_cast<const unsigned short*>(
+          p + lane * 2);
+      out[0] = fp8_to_f32<0>(us);
+      out[DPL - 1] = fp8_to_f32<1>(us);
+    } else {
+      out[0] = fp8_to_f32<0>(p[lane]);
+    }
+  }
+}
+
+template <int D, int GC, bool KV8>
 __global__ __launch_bounds__(NW * WAVE)
 void paged_attn_decode_kernel(
     bf16_t* __restrict__ out,            // [B, HQ, D]
     const bf16_t* __restrict__ q,        // [B, HQ, D]
-    const bf16_t* __restrict__ k_cache,  // [NB, HKV, BS, D]
-    const bf16_t* __restrict__ v_cache,
+    const void* __restrict__ k_cache,    // [NB, HKV, BS, D] bf16 | fp8
+    const void* __restrict__ v_cache,
     const int* __restrict__ block_table, // [B, max_blocks]
     const int* __restrict__ seq_lens,    // [B]
     int hq, int hkv, int max_blocks, float scale,
-    int nsplit,                           // flash-decoding sequence splits
+    int nsplit,
     float* __restrict__ part_o,           // [B, HQ, NS, D]  (nsplit > 1)
     float* __restrict__ part_ml) {        // [B, HQ, NS, 2]
-  constexpr int LPT = D / 8;        // lanes per token for K rows (uint4 each)
+  constexpr int LPT = D / 8;        // lanes per token (8 dims each)
   constexpr int TPP = WAVE / LPT;   // tokens per pass
   constexpr int PASSES = BS / TPP;
   constexpr int DPL = D / WAVE;     // accum dims per lane (1 or 2)
@@ -47,8 +107,8 @@ void paged_attn_decode_kernel(
 
   const int b = blockIdx.x;
   const int h = blockIdx.y;           // kv head
-  const int split = blockIdx.z % nsplit;   // sequence split (flash-decoding)
-  const int gchunk = blockIdx.z / nsplit;  // q-head chunk within the group
+  const int split = blockIdx.z % nsplit;
+  const int gchunk = blockIdx.z / nsplit;
   const int group = hq / hkv;
   const int g0 = gchunk * GC;
   const int ng = min(GC, group - g0);
@@ -86,37 +146,31 @@ void paged_attn_decode_kernel(
 
   __shared__ float s_scores[NW][GC][BS];
   const int* bt = block_table + (long)b * max_blocks;
-  const long head_off = (long)h * BS * D;
-  const long blk_stride = (long)hkv * BS * D;
+  const long head_off = (long)h * BS * D;       // elems
+  const long blk_stride = (long)hkv * BS * D;   // elems
 
-  // ---- K prefetch pipeline: tile bi held in regs while bi+NW streams ----
-  // K/V stream is read-once per step (each (seq, kv-head) block is owned by
-  // one workgroup): non-temporal loads keep it out of L1 (guide nt-weights:
-  // -18% issued-to-landed on one-reader streams).
-  const int wstride = NW * nsplit;        // total concurrent waves per (b,h,g)
-  const int w0 = split * NW + wave;       // this wave's first block
-  uint4v kt[PASSES];
+  // ---- K prefetch pipeline ---------------------------------------------
+  const int wstride = NW * nsplit;
+  const int w0 = split * NW + wave;
+  KTile<KV8> kt[PASSES];
   int blk_cur = (w0 < nblk) ? bt[w0] : 0;
   if (w0 < nblk) {
-    const bf16_t* kbase = k_cache + (long)blk_cur * blk_stride + head_off;
+    const long base = (long)blk_cur * blk_stride + head_off;
     #pragma unroll
     for (int p = 0; p < PASSES; ++p)
-      kt[p] = nt_load_u4(
-          reinterpret_cast<const uint4*>(kbase + (p * TPP + sub) * D) + ll);
+      kt[p].load(k_cache, base + (long)(p * TPP + sub) * D + ll * 8);
   }
 
   for (int bi = w0; bi < nblk; bi += wstride) {
-    // issue next tile's loads before any compute on the current tile
-    uint4v kt_next[PASSES];
+    KTile<KV8> kt_next[PASSES];
     const int bnext = bi + wstride;
     int blk_next = 0;
     if (bnext < nblk) {
       blk_next = bt[bnext];
-      const bf16_t* kb = k_cache + (long)blk_next * blk_stride + head_off;
+      const long base = (long)blk_next * blk_stride + head_off;
       #pragma unroll
       for (int p = 0; p < PASSES; ++p)
-        kt_next[p] = nt_load_u4(
-            reinterpret_cast<const uint4*>(kb + (p * TPP + sub) * D) + ll);
+        kt_next[p].load(k_cache, base + (long)(p * TPP + sub) * D + ll * 8);
     }
     const int blk = blk_cur;
 
@@ -125,13 +179,13 @@ void paged_attn_decode_kernel(
     for (int p = 0; p < PASSES; ++p) {
       const int t = p * TPP + sub;
       const int tok_idx = bi * BS + t;
-      float2 f0 = unpack_bf16x2(kt[p][0]), f1 = unpack_bf16x2(kt[p][1]);
-      float2 f2 = unpack_bf16x2(kt[p][2]), f3 = unpack_bf16x2(kt[p][3]);
+      float kf[8];
+      kt[p].unpack(kf);
       #pragma unroll
       for (int g = 0; g < GC; ++g) {
-        float s = qf[g][0] * f0.x + qf[g][1] * f0.y + qf[g][2] * f1.x +
-                  qf[g][3] * f1.y + qf[g][4] * f2.x + qf[g][5] * f2.y +
-                  qf[g][6] * f3.x + qf[g][7] * f3.y;
+        float s = qf[g][0] * kf[0] + qf[g][1] * kf[1] + qf[g][2] * kf[2] +
+                  qf[g][3] * kf[3] + qf[g][4] * kf[4] + qf[g][5] * kf[5] +
+                  qf[g][6] * kf[6] + qf[g][7] * kf[7];
         #pragma unroll
         for (int off = LPT / 2; off > 0; off >>= 1)
           s += __shfl_xor(s, off, WAVE);
@@ -142,20 +196,14 @@ void paged_attn_decode_kernel(
     __builtin_amdgcn_wave_barrier();
 
     // ---- online softmax + P·V -----------------------------------------
-    const bf16_t* vbase = v_cache + (long)blk * blk_stride + head_off;
+    const char* vbase = reinterpret_cast<const char*>(v_cache) +
+                        ((long)blk * blk_stride + head_off) * (KV8 ? 1 : 2);
     const int valid = min(n - bi * BS, BS);
     if (valid == BS) {
-      // full block: hoist all 16 V loads ahead of the FMA chains
-      unsigned int vv[BS];
-      bf16_t vs[BS];
+      float vv[BS][DPL];
       #pragma unroll
-      for (int t = 0; t < BS; ++t) {
-        if (DPL == 2)
-          vv[t] = __builtin_nontemporal_load(
-              reinterpret_cast<const unsigned int*>(vbase + t * D) + lane);
-        else
-          vs[t] = vbase[t * D + lane];
-      }
+      for (int t = 0; t < BS; ++t)
+        load_v<KV8, DPL>(vbase, t, D, lane, vv[t]);
       #pragma unroll
       for (int g = 0; g < GC; ++g) {
         float bm = -INFINITY;
@@ -171,13 +219,8 @@ void paged_attn_decode_kernel(
         for (int t = 0; t < BS; ++t) {
           const float p = __expf(s_scores[wave][g][t] - mn);
           l[g] += p;
-          if (DPL == 2) {
-            float2 vf = unpack_bf16x2(vv[t]);
-            acc[g][0] += p * vf.x;
-            acc[g][1] += p * vf.y;
-          } else {
-            acc[g][0] += p * bf2f(vs[t]);
-          }
+          #pragma unroll
+          for (int j = 0; j < DPL; ++j) acc[g][j] += p * vv[t][j];
         }
       }
     } else {
@@ -196,19 +239,13 @@ void paged_attn_decode_kernel(
         for (int t = 0; t < valid; ++t) {
           const float p = __expf(s_scores[wave][g][t] - mn);
           l[g] += p;
-          if (DPL == 2) {
-            const unsigned int uv = reinterpret_cast<const unsigned int*>(
-                vbase + t * D)[lane];
-            float2 vf = unpack_bf16x2(uv);
-            acc[g][0] += p * vf.x;
-            acc[g][1] += p * vf.y;
-          } else {
-            acc[g][0] += p * bf2f(vbase[t * D + lane]);
-          }
+          float vvt[DPL];
+          load_v<KV8, DPL>(vbase, t, D, lane, vvt);
+          #pragma unroll
+          for (int j = 0; j < DPL; ++j) acc[g][j] += p * vvt[j];
         }
       }
     }
-    // rotate the pipeline
     blk_cur = blk_next;
     #pragma unroll
     for (int p = 0; p < PASSES; ++p) kt[p] = kt_next[p];
@@ -320,11 +357,11 @@ int paged_attn_nsplit(int bsz, int hq, int hkv) {
 }
 
 void launch_paged_attn_decode(bf16_t* out, const bf16_t* q,
-                              const bf16_t* k_cache, const bf16_t* v_cache,
+                              const void* k_cache, const void* v_cache,
                               const int* block_table, const int* seq_lens,
                               int bsz, int hq, int hkv, int d, int max_blocks,
                               int block_size, float scale, int nsplit,
-                              float* part_o, float* part_ml,
+                              float* part_o, float* part_ml, bool kv8,
                               hipStream_t stream) {
   const int group = hq / hkv;
   // GC=8 when the GQA group tiles evenly by 8 (llama2-70b, llama3-style);
@@ -334,14 +371,21 @@ void launch_paged_attn_decode(bf16_t* out, const bf16_t* q,
   const int gchunks = (group + gc - 1) / gc;
   dim3 grid(bsz, hkv, gchunks * nsplit);
   dim3 block(NW * WAVE);
-  #define LAUNCH(DD, GG) \
-    paged_attn_decode_kernel<DD, GG><<<grid, block, 0, stream>>>( \
+  #define LAUNCH(DD, GG, K8) \
+    paged_attn_decode_kernel<DD, GG, K8><<<grid, block, 0, stream>>>( \
         out, q, k_cache, v_cache, block_table, seq_lens, hq, hkv, \
         max_blocks, scale, nsplit, part_o, part_ml)
-  if (d == 128 && gc == 8) LAUNCH(128, 8);
-  else if (d == 128) LAUNCH(128, 1);
-  else if (d == 64 && gc == 8) LAUNCH(64, 8);
-  else if (d == 64) LAUNCH(64, 1);
+  if (!kv8) {
+    if (d == 128 && gc == 8) LAUNCH(128, 8, false);
+    else if (d == 128) LAUNCH(128, 1, false);
+    else if (d == 64 && gc == 8) LAUNCH(64, 8, false);
+    else if (d == 64) LAUNCH(64, 1, false);
+  } else {
+    if (d == 128 && gc == 8) LAUNCH(128, 8, true);
+    else if (d == 128) LAUNCH(128, 1, true);
+    else if (d == 64 && gc == 8) LAUNCH(64, 8, true);
+    else if (d == 64) LAUNCH(64, 1, true);
+  }
   #undef LAUNCH
   if (nsplit > 1) {
     dim3 rgrid(bsz, hq);

@@ -13,8 +13,8 @@ void launch_rms_norm_residual(bf16_t*, bf16_t*, const bf16_t*, const bf16_t*, lo
 void launch_layer_norm(bf16_t*, const bf16_t*, const bf16_t*, const bf16_t*, long, int, float, hipStream_t);
 void launch_silu_mul(bf16_t*, const bf16_t*, long, int, hipStream_t);
 void launch_gelu(bf16_t*, const bf16_t*, long, hipStream_t);
-void launch_rope_kv_append(bf16_t*, const bf16_t*, const bf16_t*, const int*, bf16_t*, bf16_t*, const int*, int, int, int, int, int, float, hipStream_t);
-void launch_paged_attn_decode(bf16_t*, const bf16_t*, const bf16_t*, const bf16_t*, const int*, const int*, int, int, int, int, int, int, float, int, float*, float*, hipStream_t);
+void launch_rope_kv_append(bf16_t*, bf16_t*, const bf16_t*, const int*, void*, void*, const int*, int, int, int, int, int, float, bool, hipStream_t);
+void launch_paged_attn_decode(bf16_t*, const bf16_t*, const void*, const void*, const int*, const int*, int, int, int, int, int, int, float, int, float*, float*, bool, hipStream_t);
 int paged_attn_nsplit(int, int, int);
 bool paged_attn_decode_supported(int, int);
 void launch_attn_prefill(bf16_t*, const bf16_t*, const bf16_t*, const bf16_t*, const int*, int, int, int, int, int, float, hipStream_t);
@@ -103,11 +103,20 @@ void gelu_tanh(torch::Tensor out, torch::Tensor x) {
   post_launch_check("gelu_tanh");
 }
 
+bool cache_kv8(const torch::Tensor& t, const char* name) {
+  if (t.scalar_type() == torch::kBFloat16) return false;
+  TORCH_CHECK(t.scalar_type() == torch::kFloat8_e4m3fn, name,
+              " cache must be bf16 or float8_e4m3fn");
+  return true;
+}
+
 void rope_kv_append(torch::Tensor q, torch::Tensor k, torch::Tensor v,
                     torch::Tensor positions, torch::Tensor k_cache,
                     torch::Tensor v_cache, torch::Tensor slots, double theta) {
   check_bf16(q, "q"); check_bf16(k, "k"); check_bf16(v, "v");
-  check_bf16(k_cache, "k_cache"); check_bf16(v_cache, "v_cache");
+  const bool kv8 = cache_kv8(k_cache, "k");
+  TORCH_CHECK(cache_kv8(v_cache, "v") == kv8, "cache dtypes must match");
+  TORCH_CHECK(k_cache.is_contiguous() && v_cache.is_contiguous());
   TORCH_CHECK(positions.scalar_type() == torch::kInt, "positions must be int32");
   TORCH_CHECK(slots.scalar_type() == torch::kInt, "slots must be int32");
   int ntok = q.size(0), hq = q.size(1), d = q.size(2);
@@ -115,9 +124,10 @@ void rope_kv_append(torch::Tensor q, torch::Tensor k, torch::Tensor v,
   int block_size = k_cache.size(2);
   TORCH_CHECK(k_cache.size(1) == hkv && k_cache.size(3) == d, "cache shape");
   TORCH_CHECK(d % 8 == 0 && (d / 2) % 8 == 0, "head_dim constraints");
-  launch_rope_kv_append(bf(q), bfc(k), bfc(v), positions.data_ptr<int>(),
-                        bf(k_cache), bf(v_cache), slots.data_ptr<int>(), ntok,
-                        hq, hkv, d, block_size, (float)theta, stream());
+  launch_rope_kv_append(bf(q), bf(k), bfc(v), positions.data_ptr<int>(),
+                        k_cache.data_ptr(), v_cache.data_ptr(),
+                        slots.data_ptr<int>(), ntok,
+                        hq, hkv, d, block_size, (float)theta, kv8, stream());
   post_launch_check("rope_kv_append");
 }
 
@@ -126,7 +136,9 @@ void paged_attn_decode(torch::Tensor out, torch::Tensor q,
                        torch::Tensor block_table, torch::Tensor seq_lens,
                        double scale) {
   check_bf16(out, "out"); check_bf16(q, "q");
-  check_bf16(k_cache, "k_cache"); check_bf16(v_cache, "v_cache");
+  const bool kv8 = cache_kv8(k_cache, "k");
+  TORCH_CHECK(cache_kv8(v_cache, "v") == kv8, "cache dtypes must match");
+  TORCH_CHECK(k_cache.is_contiguous() && v_cache.is_contiguous());
   TORCH_CHECK(block_table.scalar_type() == torch::kInt, "block_table int32");
   TORCH_CHECK(seq_lens.scalar_type() == torch::kInt, "seq_lens int32");
   TORCH_CHECK(block_table.is_contiguous() && seq_lens.is_contiguous());
@@ -147,11 +159,12 @@ void paged_attn_decode(torch::Tensor out, torch::Tensor q,
     po = part_o.data_ptr<float>();
     pml = part_ml.data_ptr<float>();
   }
-  launch_paged_attn_decode(bf(out), bfc(q), bfc(k_cache), bfc(v_cache),
+  launch_paged_attn_decode(bf(out), bfc(q), k_cache.data_ptr(),
+                           v_cache.data_ptr(),
                            block_table.data_ptr<int>(),
                            seq_lens.data_ptr<int>(), bsz, hq, hkv, d,
                            max_blocks, block_size, (float)scale, nsplit,
-                           po, pml, stream());
+                           po, pml, kv8, stream());
   post_launch_check("paged_attn_decode");
 }
 

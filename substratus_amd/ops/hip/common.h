@@ -21,6 +21,24 @@ DEV_INLINE uint4v nt_load_u4(const void* p) {
   return __builtin_nontemporal_load(reinterpret_cast<const uint4v*>(p));
 }
 
+typedef __attribute__((ext_vector_type(2))) unsigned int uint2v;
+
+DEV_INLINE uint2v nt_load_u2(const void* p) {
+  return __builtin_nontemporal_load(reinterpret_cast<const uint2v*>(p));
+}
+
+// OCP fp8 e4m3 <-> f32 via the gfx950 hardware converts (OCP semantics on
+// gfx950 — matches torch.float8_e4m3fn).
+template <int SEL>
+DEV_INLINE float fp8_to_f32(unsigned int word) {
+  return __builtin_amdgcn_cvt_f32_fp8(word, SEL);  // sel must be constant
+}
+
+DEV_INLINE unsigned char f32_to_fp8(float x) {
+  return (unsigned char)(__builtin_amdgcn_cvt_pk_fp8_f32(x, x, 0, false) &
+                         0xFF);
+}
+
 DEV_INLINE float bf2f(bf16_t x) { return __bfloat162float(x); }
 DEV_INLINE bf16_t f2bf(float x) { return __float2bfloat16(x); }
 

@@ -42,6 +42,7 @@ class EngineConfig:
     prefill_chunk: int = 8192              # max packed prefill tokens per step
     use_graphs: bool = True                # hipGraph-captured decode on GPU
     quant: Optional[str] = None            # "q4_0" | "q8_0" GGUF weights
+    kv_dtype: Optional[str] = None         # None=model dtype | "fp8" (e4m3)
     seed: int = 0
 
 
@@ -87,9 +88,15 @@ class Engine:
         else:
             num_blocks = cfg.max_batch * (
                 (cfg.max_seq_len + cfg.block_size - 1) // cfg.block_size)
+        kv_dtype = cfg.dtype
+        if cfg.kv_dtype == "fp8":
+            assert self.device.type == "cuda", "fp8 KV cache is GPU-only"
+            kv_dtype = torch.float8_e4m3fn
+        elif cfg.kv_dtype is not None:
+            raise ValueError(f"unknown kv_dtype {cfg.kv_dtype!r}")
         self.kv = PagedKVCache(self.mcfg, n_kv_local, num_blocks,
                                cfg.block_size, device=self.device,
-                               dtype=cfg.dtype)
+                               dtype=kv_dtype)
         self.max_blocks_per_seq = (cfg.max_seq_len + cfg.block_size - 1) // cfg.block_size
         self._next_id = itertools.count()
         self.waiting: list[SequenceState] = []

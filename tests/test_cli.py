@@ -33,7 +33,7 @@ def daemon(tmp_path_factory):
     os.environ["SCI_GRPC_PORT"] = str(_free_port())
     os.environ["SCI_HTTP_PORT"] = str(_free_port())
     platform = Platform(str(data))
-    app = create_api(platform.kube)
+    app = create_api(platform.kube, platform)
     port = _free_port()
     config = uvicorn.Config(app, host="127.0.0.1", port=port,
                             log_level="error")

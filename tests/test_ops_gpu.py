@@ -436,3 +436,64 @@ def test_prefill_matches_decode_continuation():
     while engB.has_work():
         toksB.extend(o.token for o in engB.step())
     assert toksA == toksB
+
+
+# ------------------------------------------------------------- fp8 KV cache
+def test_paged_attn_decode_fp8_kv():
+    """fp8 e4m3 KV cache vs an fp32 reference computed from the SAME
+    quantized values (the quantization error itself is checked separately)."""
+    from substratus_amd.ops import ref
+    from substratus_amd import ops
+    torch.manual_seed(40)
+    bsz, hq, hkv, d, bs = 4, 8, 4, 128, 16
+    lens = [1, 17, 64, 33]
+    maxb = 4
+    nb = 20
+    kc8 = (torch.randn(nb, hkv, bs, d, device=DEV) * 0.5).to(
+        torch.float8_e4m3fn)
+    vc8 = (torch.randn(nb, hkv, bs, d, device=DEV) * 0.5).to(
+        torch.float8_e4m3fn)
+    bt = torch.arange(bsz * maxb, dtype=torch.int32).reshape(bsz, maxb)
+    q = torch.randn(bsz, hq, d, device=DEV).bfloat16()
+    seq_lens = torch.tensor(lens, dtype=torch.int32)
+    got = ops.paged_attn_decode(q, kc8, vc8, bt.to(DEV), seq_lens.to(DEV),
+                                d ** -0.5)
+    want = ref.paged_attn_decode(
+        q.cpu(), kc8.cpu().to(torch.bfloat16), vc8.cpu().to(torch.bfloat16),
+        bt, seq_lens, d ** -0.5)
+    assert_close_bf16(got, want, atol=3e-2, rtol=3e-2)
+
+
+def test_rope_kv_append_fp8():
+    from substratus_amd.ops import ref
+    from substratus_amd import ops
+    torch.manual_seed(41)
+    ntok, hq, hkv, nb, bs, d = 6, 4, 2, 4, 16, 128
+    q = torch.randn(ntok, hq, d, device=DEV).bfloat16()
+    k = torch.randn(ntok, hkv, d, device=DEV).bfloat16()
+    v = torch.randn(ntok, hkv, d, device=DEV).bfloat16()
+    pos = torch.randint(0, 100, (ntok,), dtype=torch.int32, device=DEV)
+    slots = torch.randperm(nb * bs, device=DEV)[:ntok].to(torch.int32)
+    kc8 = torch.zeros(nb, hkv, bs, d, device=DEV).to(torch.float8_e4m3fn)
+    vc8 = torch.zeros_like(kc8)
+    kc_ref = torch.zeros(nb, hkv, bs, d).bfloat16()
+    vc_ref = torch.zeros_like(kc_ref)
+    ref.rope_kv_append(q.cpu().clone(), k.cpu().clone(), v.cpu(), pos.cpu(),
+                       kc_ref, vc_ref, slots.cpu(), 10000.0)
+    ops.rope_kv_append(q, k, v, pos, kc8, vc8, slots, 10000.0)
+    # fp8 storage: compare dequantized cache vs bf16 ref within e4m3 step
+    assert_close_bf16(kc8.float(), kc_ref.float(), atol=8e-2, rtol=8e-2)
+    assert_close_bf16(vc8.float(), vc_ref.float(), atol=8e-2, rtol=8e-2)
+
+
+def test_engine_fp8_kv_generates():
+    from substratus_amd.serve.engine import Engine, EngineConfig
+    torch.manual_seed(42)
+    cfg = EngineConfig(model="llama-tiny", dtype=torch.bfloat16, max_batch=2,
+                       max_seq_len=96, kv_blocks=32, kv_dtype="fp8")
+    eng = Engine(cfg, device=DEV)
+    sid = eng.add_request([1, 2, 3, 4, 5, 6, 7, 8], max_new_tokens=6)
+    toks = []
+    while eng.has_work():
+        toks.extend(o.token for o in eng.step())
+    assert len(toks) == 6
