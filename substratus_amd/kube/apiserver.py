@@ -19,8 +19,6 @@ import time
 from typing import Optional
 
 from fastapi import FastAPI, HTTPException, Request
-from fastapi.responses import JSONResponse
-
 from ..api import types as apiv1
 from .memapi import Conflict, MemKube, NotFound
 
@@ -125,7 +123,6 @@ class Platform:
         from ..cloud import new_cloud
         from ..controller.manager import Manager
         from ..runtime.local import LocalRuntime
-        from ..sci.kind_server import KindSCIService
         self.sci_address = sci_address
         os.makedirs(data_dir, exist_ok=True)
         self.bucket_dir = os.path.join(data_dir, "bucket")

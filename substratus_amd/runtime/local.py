@@ -21,7 +21,6 @@ external images):
 
 from __future__ import annotations
 
-import json
 import os
 import socket
 import subprocess

@@ -28,7 +28,6 @@ import torch
 from fastapi import FastAPI, Request
 from fastapi.responses import JSONResponse, StreamingResponse
 
-from ..models.config import get_config
 from . import tokenizer as tok
 from .engine import Engine, EngineConfig, StepOutput
 
