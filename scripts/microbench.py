@@ -76,6 +76,20 @@ def bench_sample(B=64, V=32000):
     print(f"softmax_sample greedy B={B} V={V}: {dt*1e6:.1f} us")
 
 
+def bench_prefill(B=8, hq=32, hkv=32, d=128, S=512):
+    torch.manual_seed(0)
+    t = B * S
+    q = torch.randn(t, hq, d, device="cuda").bfloat16()
+    k = torch.randn(t, hkv, d, device="cuda").bfloat16()
+    v = torch.randn(t, hkv, d, device="cuda").bfloat16()
+    ss = torch.arange(0, t + 1, S, dtype=torch.int32, device="cuda")
+    scale = d ** -0.5
+    dt = timeit(lambda: ops.attn_prefill(q, k, v, ss, scale), iters=50)
+    flops = 4.0 * B * hq * (S * (S + 1) / 2) * d
+    print(f"attn_prefill B={B} S={S} hq={hq} d={d}: {dt*1e6:.1f} us, "
+          f"{flops/dt/1e12:.0f} TF/s")
+
+
 def bench_quant_gemv(n=13824, k=5120, quant="q4_0"):
     from substratus_amd.parallel.tp import Linear
     from substratus_amd.models.quant import QuantLinear
@@ -99,6 +113,9 @@ if __name__ == "__main__":
         bench_attn()
         bench_attn(ctx=2048)
         bench_attn(hq=64, hkv=8, ctx=512)  # llama2-70b-style GQA
+    if a.which in ("all", "prefill"):
+        bench_prefill()
+        bench_prefill(S=2048, B=4)
     if a.which in ("all", "rope"):
         bench_rope()
     if a.which in ("all", "norm"):

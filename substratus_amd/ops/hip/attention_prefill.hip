@@ -3,19 +3,23 @@
 // ingestion.
 //
 // Geometry: one workgroup (4 waves) per (sequence, q-head, 64-row q tile);
-// wave w owns q rows [tile*64 + 16w, +16). Keys stream in chunks of 32:
-//   * QK^T: Q A-fragments preloaded in registers (D/32 chunks × short8);
-//     K^T B-fragments loaded straight from global (k is dense [T, Hkv, D]).
-//     D/32 MFMAs per 16-key subchunk → S[16q × 16k] fp32 C-fragments.
-//   * Online softmax in C-layout registers (row = (lane>>4)*4+i, col =
-//     lane&15): 16-lane shfl row-max/row-sum, exp, rescale (guide §5
-//     common-mistake #6: wave-parallel softmax).
-//   * P transposed C-layout → A-fragment through a per-wave LDS tile;
-//     V[32, D] staged cooperatively into LDS with coalesced uint4 loads;
-//     P·V accumulates O[16q × D] as D/16 C-fragments.
+// wave w owns q rows [tile*64 + 16w, +16). Keys stream in chunks of 32.
 //
-// MFMA fragment maps (gfx950, 16x16x32 bf16; verified by the mfma_probe
-// GPU test against torch.matmul on asymmetric data):
+// V2 data movement (V1 measured 141 TF — dominated by scalar LDS reads and
+// unswizzled layouts):
+//   * K and V chunks are staged cooperatively into LDS with coalesced uint4
+//     global loads; V is TRANSPOSED at staging (vT[d][k]) so every MFMA
+//     B-fragment — K^T for QK^T, V for P·V — is ONE contiguous 16-byte
+//     ds_read_b128 per lane instead of 8 scalar reads.
+//   * All LDS tiles are XOR-swizzled at 16-byte-chunk granularity
+//     (chunk ^= row & (row_chunks-1), guide T2): fragment reads put 16 lanes
+//     on 16 DIFFERENT rows at the same column range, which is the textbook
+//     16-way ds_read_b128 bank conflict on a linear layout.
+//   * P transposes C-layout → A-layout through a swizzled LDS tile the same
+//     way (scalar writes, one uint4 read per fragment).
+//
+// MFMA fragment maps (gfx950, 16x16x32 bf16; verified by the mfma_probe GPU
+// test against torch.matmul on asymmetric data):
 //   A: lane l holds A[row = l&15][k = (l>>4)*8 + j]  (j = 0..7, short8)
 //   B: lane l holds B[k = (l>>4)*8 + j][col = l&15]
 //   C/D: lane l holds C[row = (l>>4)*4 + i][col = l&15]  (i = 0..3, float4)
@@ -27,6 +31,14 @@ typedef __attribute__((ext_vector_type(4))) float float4v;
 #define PF_WAVES 4
 #define QTILE 64   // q rows per workgroup (16 per wave)
 #define KCHUNK 32  // keys per iteration
+
+// XOR-swizzled 16-B-chunk address inside a row-major LDS tile.
+// row_bytes must be a power-of-two multiple of 16.
+template <int ROW_BYTES>
+DEV_INLINE int swz_chunk(int row, int chunk) {
+  constexpr int NCH = ROW_BYTES / 16;
+  return row * NCH + (chunk ^ (row & (NCH - 1)));
+}
 
 template <int D>
 __global__ __launch_bounds__(PF_WAVES * WAVE)
@@ -64,12 +76,10 @@ void attn_prefill_kernel(bf16_t* __restrict__ out,      // [T, HQ, D]
     const bf16_t* qp = q + (long)(base + qrow) * qrow_stride + (long)head * D;
     #pragma unroll
     for (int c = 0; c < DC32; ++c) {
-      // 8 contiguous bf16 at d = c*32 + lhi*8
       qa[c] = *reinterpret_cast<const short8*>(qp + c * 32 + lhi * 8);
     }
   }
 
-  // ---- state ------------------------------------------------------------
   float m[4], l[4];             // per held q row (i = 0..3 → q = lhi*4+i)
   float4v o_acc[DC16];
   #pragma unroll
@@ -77,8 +87,13 @@ void attn_prefill_kernel(bf16_t* __restrict__ out,      // [T, HQ, D]
   #pragma unroll
   for (int c = 0; c < DC16; ++c) o_acc[c] = (float4v){0.f, 0.f, 0.f, 0.f};
 
-  __shared__ bf16_t v_tile[KCHUNK][D];                 // cooperative stage
-  __shared__ bf16_t p_tile[PF_WAVES][16][KCHUNK];      // per-wave P transpose
+  // LDS tiles, all uint4-granular and XOR-swizzled:
+  //   k_tile: [KCHUNK rows][D cols]        — QK^T B-fragments
+  //   vT:     [D rows][KCHUNK cols]        — P·V B-fragments (transposed!)
+  //   p_t:    per wave [16 rows][KCHUNK]   — P C→A transpose
+  __shared__ uint4 k_tile[KCHUNK * D * 2 / 16];
+  __shared__ uint4 vT[D * KCHUNK * 2 / 16];
+  __shared__ uint4 p_t[PF_WAVES][16 * KCHUNK * 2 / 16];
 
   const int kmax_tile = min(q0_tile + QTILE, n);       // causal bound of tile
   const int nchunks = (kmax_tile + KCHUNK - 1) / KCHUNK;
@@ -86,37 +101,51 @@ void attn_prefill_kernel(bf16_t* __restrict__ out,      // [T, HQ, D]
 
   for (int kc = 0; kc < nchunks; ++kc) {
     const int kbase = kc * KCHUNK;
-    // ---- cooperative V stage: 32 rows × D, uint4 per lane --------------
     __syncthreads();
+    // ---- cooperative K+V stage -----------------------------------------
     {
-      constexpr int VLOADS = KCHUNK * D / 8 / (PF_WAVES * WAVE);  // uint4 each
+      constexpr int LOADS = KCHUNK * D / 8 / (PF_WAVES * WAVE);  // per thread
       #pragma unroll
-      for (int i = 0; i < VLOADS; ++i) {
-        const int flat = (i * PF_WAVES * WAVE + threadIdx.x) * 8;  // bf16 idx
+      for (int i = 0; i < LOADS; ++i) {
+        const int flat = (i * PF_WAVES * WAVE + threadIdx.x) * 8;  // elem idx
         const int row = flat / D, col = flat % D;
-        uint4 val = make_uint4(0, 0, 0, 0);
-        if (kbase + row < n) {
-          val = *reinterpret_cast<const uint4*>(
-              v + (long)(base + kbase + row) * krow_stride + (long)kvh * D + col);
+        const bool ok = kbase + row < n;
+        const long goff = (long)(base + kbase + row) * krow_stride +
+                          (long)kvh * D + col;
+        // K: row-major, swizzled by 16-B chunk
+        uint4 kv4 = ok ? *reinterpret_cast<const uint4*>(k + goff)
+                       : make_uint4(0, 0, 0, 0);
+        k_tile[swz_chunk<D * 2>(row, col * 2 / 16)] = kv4;
+        // V: transpose into vT[d][k] (8 scalar writes of this uint4's dims)
+        uint4 vv4 = ok ? *reinterpret_cast<const uint4*>(v + goff)
+                       : make_uint4(0, 0, 0, 0);
+        const unsigned short* vs =
+            reinterpret_cast<const unsigned short*>(&vv4);
+        #pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          const int d = col + j;
+          // byte address inside vT row d: k index `row` -> byte row*2
+          unsigned short* vt16 = reinterpret_cast<unsigned short*>(vT);
+          const int chunk = swz_chunk<KCHUNK * 2>(d, (row * 2) / 16);
+          vt16[chunk * 8 + (row % 8)] = vs[j];
         }
-        *reinterpret_cast<uint4*>(&v_tile[row][col]) = val;
       }
     }
     __syncthreads();
 
     const bool compute = wave_live && kbase < kmax_wave;
     if (compute) {
-      // ---- QK^T: two 16-key subchunks -------------------------------
+      // ---- QK^T: two 16-key subchunks (B-frags from swizzled k_tile) --
       float4v sfrag[2];
       #pragma unroll
       for (int sc = 0; sc < 2; ++sc) {
         float4v acc = {0.f, 0.f, 0.f, 0.f};
-        const int krow = kbase + sc * 16 + llo;  // B col = llo
-        const bf16_t* kp = k + (long)(base + min(krow, n - 1)) * krow_stride +
-                           (long)kvh * D;
+        const int key = sc * 16 + llo;   // B col = this lane's key row
         #pragma unroll
         for (int c = 0; c < DC32; ++c) {
-          short8 kb = *reinterpret_cast<const short8*>(kp + c * 32 + lhi * 8);
+          // kdim byte offset = (c*32 + lhi*8)*2 -> chunk c*4 + lhi
+          const uint4 kb4 = k_tile[swz_chunk<D * 2>(key, c * 4 + lhi)];
+          const short8 kb = *reinterpret_cast<const short8*>(&kb4);
           acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(qa[c], kb, acc, 0, 0, 0);
         }
         sfrag[sc] = acc;
@@ -138,7 +167,6 @@ void attn_prefill_kernel(bf16_t* __restrict__ out,      // [T, HQ, D]
           rowmax[i] = fmaxf(rowmax[i], sv);
         }
       }
-      // reduce rowmax across the 16 lanes holding this q row
       #pragma unroll
       for (int i = 0; i < 4; ++i) {
         #pragma unroll
@@ -152,8 +180,9 @@ void attn_prefill_kernel(bf16_t* __restrict__ out,      // [T, HQ, D]
         alpha[i] = (m[i] == -INFINITY) ? 0.f : __expf(m[i] - mn);
         m[i] = mn;
       }
-      // P = exp(S - m); row-sum; write transposed into p_tile
+      // P = exp(S - m); row-sum; write transposed into swizzled p_t
       float rowsum[4] = {0.f, 0.f, 0.f, 0.f};
+      unsigned short* p16 = reinterpret_cast<unsigned short*>(p_t[wave]);
       #pragma unroll
       for (int sc = 0; sc < 2; ++sc) {
         #pragma unroll
@@ -161,7 +190,12 @@ void attn_prefill_kernel(bf16_t* __restrict__ out,      // [T, HQ, D]
           float pv = (pm[sc][i] == -INFINITY || m[i] == -INFINITY)
                          ? 0.f : __expf(pm[sc][i] - m[i]);
           rowsum[i] += pv;
-          p_tile[wave][lhi * 4 + i][sc * 16 + llo] = f2bf(pv);
+          const int qrow = lhi * 4 + i;          // P row
+          const int kk = sc * 16 + llo;          // P col
+          bf16_t pb = f2bf(pv);
+          const int chunk = swz_chunk<KCHUNK * 2>(qrow, (kk * 2) / 16);
+          p16[chunk * 8 + (kk % 8)] =
+              *reinterpret_cast<unsigned short*>(&pb);
         }
       }
       #pragma unroll
@@ -177,19 +211,14 @@ void attn_prefill_kernel(bf16_t* __restrict__ out,      // [T, HQ, D]
         #pragma unroll
         for (int i = 0; i < 4; ++i) o_acc[c][i] *= alpha[i];
       }
-      // ---- P·V: A = P [16q × 32k] from LDS, B = V [32k × 16d] from LDS
-      short8 pa;  // A-fragment of P: row llo, k = lhi*8 + j
-      #pragma unroll
-      for (int j = 0; j < 8; ++j)
-        ((short*)&pa)[j] = *reinterpret_cast<const short*>(
-            &p_tile[wave][llo][lhi * 8 + j]);
+      // ---- P·V: A-frag = one uint4 from p_t; B-frag = one uint4 from vT
+      const uint4 pa4 = p_t[wave][swz_chunk<KCHUNK * 2>(llo, lhi)];
+      const short8 pa = *reinterpret_cast<const short8*>(&pa4);
       #pragma unroll
       for (int c = 0; c < DC16; ++c) {
-        short8 vb;  // B-fragment: k = lhi*8 + j, col = llo (d = c*16 + llo)
-        #pragma unroll
-        for (int j = 0; j < 8; ++j)
-          ((short*)&vb)[j] = *reinterpret_cast<const short*>(
-              &v_tile[lhi * 8 + j][c * 16 + llo]);
+        const int drow = c * 16 + llo;   // B col = output dim
+        const uint4 vb4 = vT[swz_chunk<KCHUNK * 2>(drow, lhi)];
+        const short8 vb = *reinterpret_cast<const short8*>(&vb4);
         o_acc[c] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(pa, vb, o_acc[c],
                                                            0, 0, 0);
       }

@@ -146,9 +146,7 @@ __global__ void dequant_kernel(bf16_t* __restrict__ out,   // [N, K]
                                const f16_t* __restrict__ scales,
                                const unsigned char* __restrict__ qs,
                                long nblocks_total, int k) {
-  const long b = blockIdx.x * (long)blockDim.x / WAVE +
-                 threadIdx.x / WAVE;  // one 32-block per wave? no: per lane
-  // simpler: one block per thread
+  // one 32-weight block per thread
   const long blk = blockIdx.x * (long)blockDim.x + threadIdx.x;
   if (blk >= nblocks_total) return;
   const int nb_per_row = k / 32;
