@@ -90,6 +90,23 @@ def bench_prefill(B=8, hq=32, hkv=32, d=128, S=512):
           f"{flops/dt/1e12:.0f} TF/s")
 
 
+def bench_decode_gemm(m=256):
+    import torch.nn.functional as F
+    from substratus_amd import _C
+    shapes = [(12288, 4096), (4096, 4096), (22016, 4096), (4096, 11008)]
+    for n, k in shapes:
+        torch.manual_seed(0)
+        x = torch.randn(m, k, device="cuda").bfloat16()
+        w = torch.randn(n, k, device="cuda").bfloat16() * 0.02
+        y = torch.empty(m, n, device="cuda").bfloat16()
+        dt_ours = timeit(lambda: _C.decode_gemm(y, x, w), iters=100)
+        dt_blas = timeit(lambda: F.linear(x, w), iters=100)
+        wb = n * k * 2
+        print(f"decode_gemm M={m} N={n} K={k}: ours {dt_ours*1e6:.1f} us "
+              f"({wb/dt_ours/1e12:.2f} TB/s W) vs hipBLASLt "
+              f"{dt_blas*1e6:.1f} us ({wb/dt_blas/1e12:.2f} TB/s)")
+
+
 def bench_quant_gemv(n=13824, k=5120, quant="q4_0"):
     from substratus_amd.parallel.tp import Linear
     from substratus_amd.models.quant import QuantLinear
@@ -122,6 +139,9 @@ if __name__ == "__main__":
         bench_norm()
     if a.which in ("all", "sample"):
         bench_sample()
+    if a.which in ("all", "dgemm"):
+        bench_decode_gemm()
+        bench_decode_gemm(m=64)
     if a.which in ("all", "quant"):
         bench_quant_gemv()
         bench_quant_gemv(quant="q8_0")

@@ -26,6 +26,9 @@ void launch_adamw(bf16_t*, const bf16_t*, float*, float*, long, float, float, fl
 void launch_adamw_f32(float*, const float*, float*, float*, long, float, float, float, float, float, int, hipStream_t);
 void launch_quant_gemv(bf16_t*, const bf16_t*, const void*, const unsigned char*, int, int, int, bool, hipStream_t);
 void launch_dequant(bf16_t*, const void*, const unsigned char*, long, int, bool, hipStream_t);
+int decode_gemm_splitk(int, int);
+bool decode_gemm_supported(int, int, int);
+void launch_decode_gemm(bf16_t*, const bf16_t*, const bf16_t*, float*, int, int, int, int, hipStream_t);
 
 namespace {
 
@@ -273,9 +276,29 @@ void dequant(torch::Tensor out, torch::Tensor scales, torch::Tensor qs,
   post_launch_check("dequant");
 }
 
+void decode_gemm(torch::Tensor y, torch::Tensor x, torch::Tensor w) {
+  check_bf16(y, "y"); check_bf16(x, "x"); check_bf16(w, "w");
+  int m = x.size(0), k = x.size(1), n = w.size(0);
+  TORCH_CHECK(w.size(1) == k && y.size(0) == m && y.size(1) == n);
+  TORCH_CHECK(decode_gemm_supported(m, n, k),
+              "decode_gemm unsupported shape M=", m, " N=", n, " K=", k);
+  const int splitk = decode_gemm_splitk(n, k);
+  auto part = torch::empty({splitk, 256, n},
+                           x.options().dtype(torch::kFloat));
+  launch_decode_gemm(bf(y), bfc(x), bfc(w), part.data_ptr<float>(), m, n, k,
+                     splitk, stream());
+  post_launch_check("decode_gemm");
+}
+
+bool decode_gemm_ok(long m, long n, long k) {
+  return decode_gemm_supported((int)m, (int)n, (int)k);
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("decode_gemm", &decode_gemm);
+  m.def("decode_gemm_ok", &decode_gemm_ok);
   m.def("quant_gemv", &quant_gemv);
   m.def("dequant", &dequant);
   m.def("rms_norm", &rms_norm);

@@ -497,3 +497,17 @@ def test_engine_fp8_kv_generates():
     while eng.has_work():
         toks.extend(o.token for o in eng.step())
     assert len(toks) == 6
+
+
+# ----------------------------------------------------------- decode GEMM
+@pytest.mark.parametrize("m,n,k", [(256, 12288, 4096), (256, 4096, 4096),
+                                   (64, 4096, 11008), (17, 22016, 4096)])
+def test_decode_gemm_matches_linear(m, n, k):
+    from substratus_amd import _C
+    torch.manual_seed(33)
+    x = torch.randn(m, k, device=DEV).bfloat16()
+    w = (torch.randn(n, k, device=DEV) * 0.02).bfloat16()
+    y = torch.empty(m, n, device=DEV).bfloat16()
+    _C.decode_gemm(y, x, w)
+    want = torch.nn.functional.linear(x.float().cpu(), w.float().cpu())
+    assert_close_bf16(y, want, atol=5e-2, rtol=5e-2)
