@@ -53,11 +53,13 @@ class LocalRuntime:
 
     # ------------------------------------------------------------- driving
     def sync(self) -> None:
-        """One pass: start pending Jobs/Deployments, reap finished ones."""
+        """One pass: start pending Jobs/Deployments/Pods, probe readiness."""
         for job in self.kube.list("Job"):
             self._sync_job(job)
         for dep in self.kube.list("Deployment"):
             self._sync_deployment(dep)
+        for pod in self.kube.list("Pod"):
+            self._sync_pod(pod)
 
     def run_until_quiet(self, mgr, timeout: float = 120.0,
                         poll: float = 0.2) -> None:
@@ -135,6 +137,53 @@ class LocalRuntime:
             self.kube.patch("Service", name, {"metadata": {"annotations": {
                 "substratus.ai/local-endpoint":
                     f"http://127.0.0.1:{port}"}}}, ns)
+
+    # ------------------------------------------------------------------ pods
+    def _sync_pod(self, pod: dict) -> None:
+        """Long-running Pods (Notebooks: jupyter on 8888, readiness GET
+        /api). Locally the notebook container runs the shim
+        (serve/notebook_shim.py) — same contract surface."""
+        name = pod["metadata"]["name"]
+        ns = pod["metadata"].get("namespace", "default")
+        c = pod["spec"]["containers"][0]
+        ports = [p.get("containerPort") for p in c.get("ports", []) or []]
+        if 8888 not in ports:
+            return  # only notebook-shaped pods run locally
+        key = ("Pod", ns, name)
+        if key in self.procs and self.procs[key].poll() is None:
+            self._probe_pod(pod, name, ns)
+            return
+        env, mounts = self._resolve(pod["spec"], c, ns)
+        port = _free_port()
+        env["PORT"] = str(port)
+        env.setdefault("CONTENT_DIR",
+                       mounts.get("/content/artifacts",
+                                  self.bucket_dir))
+        self.server_ports[f"{ns}/{name}"] = port
+        lf = open(self.log_path(ns, name), "ab")
+        self.procs[key] = subprocess.Popen(
+            [sys.executable, "-m", "substratus_amd.serve.notebook_shim"],
+            env=env, cwd=_repo_root(), stdout=lf, stderr=lf)
+        self.kube.patch("Pod", name, {"metadata": {"annotations": {
+            "substratus.ai/local-endpoint":
+                f"http://127.0.0.1:{port}"}}}, ns)
+
+    def _probe_pod(self, pod: dict, name: str, ns: str) -> None:
+        port = self.server_ports.get(f"{ns}/{name}")
+        if port is None:
+            return
+        try:
+            ready = httpx.get(f"http://127.0.0.1:{port}/api",
+                              timeout=2).status_code == 200
+        except Exception:
+            ready = False
+        conds = pod.get("status", {}).get("conditions", []) or []
+        cur = any(c.get("type") == "Ready" and c.get("status") == "True"
+                  for c in conds)
+        if ready != cur:
+            self.kube.patch("Pod", name, {"status": {"conditions": [
+                {"type": "Ready",
+                 "status": "True" if ready else "False"}]}}, ns)
 
     def _probe_deployment(self, dep: dict, name: str, ns: str) -> None:
         port = self.server_ports.get(f"{ns}/{name}")

@@ -1,0 +1,81 @@
+"""Notebook shim — the local runtime's stand-in for the Jupyter container.
+
+The production notebook image runs `notebook.sh` → JupyterLab on 8888
+(images/Dockerfile.notebook; container contract: readiness GET /api,
+NOTEBOOK_TOKEN honored). This offline environment has no jupyterlab, so the
+LocalRuntime serves the SAME contract surface with a minimal FastAPI app:
+GET /api → version info (the readiness probe), GET/PUT /files/* over the
+mounted /content tree (what the CLI's sync needs), token checked like
+Jupyter's `?token=`.
+"""
+
+from __future__ import annotations
+
+import os
+
+from fastapi import FastAPI, HTTPException, Request
+from fastapi.responses import JSONResponse, PlainTextResponse
+
+
+def create_app(content_dir: str = "/content",
+               token: str = "") -> FastAPI:
+    app = FastAPI(title="substratus notebook shim")
+
+    def check_token(req: Request) -> None:
+        if token and req.query_params.get("token") != token:
+            raise HTTPException(403, "bad token")
+
+    @app.get("/api")
+    def api():
+        # the readiness probe target (container contract)
+        return {"version": "substratus-notebook-shim"}
+
+    @app.get("/")
+    def root():
+        return {"status": "ok", "content": content_dir}
+
+    @app.get("/files")
+    def list_files(req: Request):
+        check_token(req)
+        out = []
+        for root, dirs, files in os.walk(content_dir):
+            dirs[:] = [d for d in dirs
+                       if d not in ("data", "model", "artifacts")
+                       and not d.startswith(".")]
+            for f in files:
+                out.append(os.path.relpath(os.path.join(root, f),
+                                           content_dir))
+        return {"files": sorted(out)}
+
+    @app.get("/files/{path:path}")
+    def get_file(path: str, req: Request):
+        check_token(req)
+        full = os.path.join(content_dir, path)
+        if not os.path.isfile(full):
+            raise HTTPException(404, path)
+        with open(full) as f:
+            return PlainTextResponse(f.read())
+
+    @app.put("/files/{path:path}")
+    async def put_file(path: str, req: Request):
+        check_token(req)
+        full = os.path.join(content_dir, path)
+        os.makedirs(os.path.dirname(full) or content_dir, exist_ok=True)
+        body = await req.body()
+        with open(full, "wb") as f:
+            f.write(body)
+        return JSONResponse({"written": path})
+
+    return app
+
+
+def main() -> None:
+    import uvicorn
+    app = create_app(os.environ.get("CONTENT_DIR", "/content"),
+                     os.environ.get("NOTEBOOK_TOKEN", ""))
+    uvicorn.run(app, host="0.0.0.0",
+                port=int(os.environ.get("PORT", 8888)))
+
+
+if __name__ == "__main__":
+    main()

@@ -75,3 +75,46 @@ def test_system_model_server_completion(tmp_path):
         assert body["usage"]["completion_tokens"] >= 1
     finally:
         rt.stop()
+
+
+@pytest.mark.timeout(180)
+def test_notebook_flow(tmp_path):
+    """Notebook CR → local runtime runs the notebook shim → readiness GET
+    /api → Ready; suspend kills it (reference notebook lifecycle)."""
+    bucket = str(tmp_path / "bucket")
+    os.makedirs(bucket)
+    kube = MemKube()
+    cloud = new_cloud({"CLOUD": "kind", "CLUSTER_NAME": "nb",
+                       "BUCKET_DIR": bucket})
+    mgr = Manager(kube, cloud, FakeSCIClient())
+    rt = LocalRuntime(kube, bucket)
+    kube.create(apiv1.new_object("Notebook", "nb1", {
+        "image": "substratus-amd/base",
+    }))
+    deadline = time.time() + 120
+    ready = False
+    try:
+        while time.time() < deadline:
+            rt.run_until_quiet(mgr, timeout=5)
+            nb = kube.get("Notebook", "nb1")
+            if apiv1.get_status_ready(nb):
+                ready = True
+                break
+            time.sleep(0.5)
+        assert ready, kube.get("Notebook", "nb1").get("status")
+        pod = kube.get("Pod", "nb1-notebook")
+        ep = pod["metadata"]["annotations"]["substratus.ai/local-endpoint"]
+        r = httpx.get(f"{ep}/api", timeout=5)
+        assert r.status_code == 200
+        # file write/read through the shim (NOTEBOOK_TOKEN honored like
+        # jupyter's ?token= — the controller sets it to "default")
+        httpx.put(f"{ep}/files/hello.py?token=default",
+                  content=b"print(1)\n", timeout=5).raise_for_status()
+        assert "hello.py" in httpx.get(f"{ep}/files?token=default",
+                                       timeout=5).json()["files"]
+        # suspend deletes the pod
+        kube.patch("Notebook", "nb1", {"spec": {"suspend": True}})
+        mgr.run_until_settled()
+        assert kube.try_get("Pod", "nb1-notebook") is None
+    finally:
+        rt.stop()
