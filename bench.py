@@ -116,8 +116,9 @@ def bench_serve(args) -> None:
         elapsed = float(t.item())
 
     assert len(eng.running) == args.batch, "sequences finished mid-bench"
-    per_rank_batch = args.batch
-    tokens = per_rank_batch * args.steps * n_gpus
+    # TP ranks serve ONE shared batch; DP replicas each serve their own
+    n_replicas = n_gpus // max(args.tp, 1)
+    tokens = args.batch * args.steps * n_replicas
     value = tokens / elapsed
     ms_per_step = elapsed / args.steps * 1000.0
     parallelism = (f"tp{args.tp}" if args.tp > 1 else f"dp{n_gpus}")
@@ -136,7 +137,7 @@ def bench_serve(args) -> None:
         "data": "synthetic",
         "config": {
             "model": model_name,
-            "global_batch": per_rank_batch * n_gpus,
+            "global_batch": args.batch * n_replicas,
             "seq_len": args.prompt_len + args.warmup + args.steps,
             "prompt_len": args.prompt_len,
             "parallelism": parallelism,

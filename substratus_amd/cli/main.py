@@ -171,8 +171,13 @@ def notebook(directory: str = typer.Argument(".")):
         libclient.notebook_for_object(manifests[0])
     nb = c.apply(nb)
     _wait_and_report(c, nb)
-    console.print(f"notebook [bold]{nb['metadata']['name']}[/bold] ready "
-                  "(port 8888, token: default)")
+    name = nb["metadata"]["name"]
+    ns = nb["metadata"].get("namespace", "default")
+    pod = c.get("Pod", f"{name}-notebook", ns) or {}
+    ep = pod.get("metadata", {}).get("annotations", {}).get(
+        "substratus.ai/local-endpoint", "http://localhost:8888")
+    console.print(f"notebook [bold]{name}[/bold] ready at "
+                  f"{ep}/?token=default")
 
 
 app.command(name="nb")(notebook)
