@@ -118,3 +118,44 @@ def test_notebook_flow(tmp_path):
         assert kube.try_get("Pod", "nb1-notebook") is None
     finally:
         rt.stop()
+
+
+@pytest.mark.timeout(600)
+def test_system_opt125m_like_reference(tmp_path):
+    """BASELINE config 1 with the reference's actual model: facebook-opt-125m
+    Model+Server on CPU, completion request (test/system.sh's exact shape)."""
+    bucket = str(tmp_path / "bucket")
+    os.makedirs(bucket)
+    kube = MemKube()
+    cloud = new_cloud({"CLOUD": "kind", "CLUSTER_NAME": "sys2",
+                       "BUCKET_DIR": bucket})
+    mgr = Manager(kube, cloud, FakeSCIClient())
+    rt = LocalRuntime(kube, bucket)
+    kube.create(apiv1.new_object("Model", "facebook-opt-125m", {
+        "image": "substratus-amd/model-loader",
+        "params": {"name": "opt-125m", "dtype": "fp32"},
+    }))
+    kube.create(apiv1.new_object("Server", "facebook-opt-125m", {
+        "image": "substratus-amd/model-server",
+        "model": {"name": "facebook-opt-125m"},
+        "params": {"max_length": 64, "max_batch": 2},
+    }))
+    deadline = time.time() + 480
+    try:
+        while time.time() < deadline:
+            rt.run_until_quiet(mgr, timeout=15)
+            if (apiv1.get_status_ready(kube.get("Model", "facebook-opt-125m"))
+                    and apiv1.get_status_ready(
+                        kube.get("Server", "facebook-opt-125m"))):
+                break
+            time.sleep(1)
+        assert apiv1.get_status_ready(kube.get("Server", "facebook-opt-125m"))
+        svc = kube.get("Service", "facebook-opt-125m-server")
+        ep = svc["metadata"]["annotations"]["substratus.ai/local-endpoint"]
+        r = httpx.post(f"{ep}/v1/completions", json={
+            "prompt": "Who was the first president of the United States?",
+            "max_tokens": 4}, timeout=180)
+        assert r.status_code == 200
+        assert r.json()["usage"]["completion_tokens"] >= 1
+    finally:
+        rt.stop()
