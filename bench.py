@@ -38,6 +38,8 @@ def parse_args():
     p.add_argument("--model", default=None, help="override model config")
     p.add_argument("--batch", type=int, default=256)
     p.add_argument("--prompt-len", type=int, default=512)
+    p.add_argument("--micro-batch", type=int, default=4,
+                   help="per-GPU micro batch (train mode)")
     p.add_argument("--seq-len", type=int, default=None)
     p.add_argument("--kv-dtype", default=None, choices=[None, "fp8"],
                    help="KV-cache dtype override (fp8 e4m3; reduced-precision"

@@ -81,7 +81,8 @@ class Engine:
             free, _total = torch.cuda.mem_get_info(self.device)
             budget = int(free * cfg.gpu_mem_fraction)
             num_blocks = PagedKVCache.blocks_for_memory(
-                self.mcfg, n_kv_local, budget, cfg.block_size)
+                self.mcfg, n_kv_local, budget, cfg.block_size,
+                dtype_bytes=kv_bytes_per_el)
             max_needed = (cfg.max_batch *
                           ((cfg.max_seq_len + cfg.block_size - 1) // cfg.block_size))
             num_blocks = min(num_blocks, max_needed)
@@ -94,6 +95,7 @@ class Engine:
             kv_dtype = torch.float8_e4m3fn
         elif cfg.kv_dtype is not None:
             raise ValueError(f"unknown kv_dtype {cfg.kv_dtype!r}")
+        kv_bytes_per_el = 1 if kv_dtype == torch.float8_e4m3fn else 2
         self.kv = PagedKVCache(self.mcfg, n_kv_local, num_blocks,
                                cfg.block_size, device=self.device,
                                dtype=kv_dtype)
@@ -169,6 +171,8 @@ class Engine:
 
     # ---------------------------------------------------------------- decode
     def _decode(self) -> list[StepOutput]:
+        if not self.running:
+            return []
         if self.cfg.use_graphs and self.device.type == "cuda":
             return self._decode_graph()
         return self._decode_eager()

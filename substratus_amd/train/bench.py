@@ -21,7 +21,7 @@ def run_train_bench(args) -> None:
     n_gpus = max(st.world_size, 1)
     on_gpu = torch.cuda.is_available()
     model_name = args.model or "falcon-7b-instruct"
-    micro_batch = args.batch if args.batch != 64 else 4  # train default
+    micro_batch = getattr(args, "micro_batch", 4)
     seq_len = args.seq_len or 2048
     cfg = TrainConfig(model=model_name, micro_batch=micro_batch,
                       seq_len=seq_len,
