@@ -90,6 +90,27 @@ def bench_prefill(B=8, hq=32, hkv=32, d=128, S=512):
           f"{flops/dt/1e12:.0f} TF/s")
 
 
+def bench_prefill_ablation(B=8, S=512, hq=32, hkv=32, d=128):
+    """Within-probe variant ablation (guide rule 8): where does prefill time
+    go? Variants are timing-only (wrong numerics)."""
+    from substratus_amd import _C
+    torch.manual_seed(0)
+    t = B * S
+    q = torch.randn(t, hq, d, device="cuda").bfloat16()
+    k = torch.randn(t, hkv, d, device="cuda").bfloat16()
+    v = torch.randn(t, hkv, d, device="cuda").bfloat16()
+    out = torch.empty_like(q)
+    ss = torch.arange(0, t + 1, S, dtype=torch.int32, device="cuda")
+    names = {0: "full", 1: "no-softmax-valu", 2: "no-pv-mfma",
+             3: "no-v-staging", 4: "no-qk-mfma"}
+    flops = 4.0 * B * hq * (S * (S + 1) / 2) * d
+    for variant in (0, 1, 2, 3, 4):
+        dt = timeit(lambda: _C.attn_prefill_probe(variant, out, q, k, v, ss,
+                                                  S, d ** -0.5), iters=50)
+        print(f"prefill variant {variant} ({names[variant]:16s}): "
+              f"{dt*1e6:7.1f} us  ({flops/dt/1e12:4.0f} TF-equiv)")
+
+
 def bench_decode_gemm(m=256):
     import torch.nn.functional as F
     from substratus_amd import _C
@@ -133,6 +154,8 @@ if __name__ == "__main__":
     if a.which in ("all", "prefill"):
         bench_prefill()
         bench_prefill(S=2048, B=4)
+    if a.which == "prefill_ablate":
+        bench_prefill_ablation()
     if a.which in ("all", "rope"):
         bench_rope()
     if a.which in ("all", "norm"):

@@ -41,7 +41,11 @@ DEV_INLINE int swz_chunk(int row, int chunk) {
   return row * NCH + (chunk ^ (row & (NCH - 1)));
 }
 
-template <int D>
+// VARIANT (ablation probe, guide §5.4 rules 8/17 — ablate before
+// optimizing; non-zero variants are TIMING-ONLY, numerics are wrong):
+//   0 full   1 no-softmax-VALU   2 no-PV-mfma   3 no-V-staging
+//   4 no-QK-mfma (QK loads kept alive)
+template <int D, int VARIANT = 0>
 __global__ __launch_bounds__(PF_WAVES * WAVE)
 void attn_prefill_kernel(bf16_t* __restrict__ out,      // [T, HQ, D]
                          const bf16_t* __restrict__ q,  // [T, HQ, D]
@@ -120,6 +124,10 @@ void attn_prefill_kernel(bf16_t* __restrict__ out,      // [T, HQ, D]
         // V: transpose into vT[d][k] (8 scalar writes of this uint4's dims)
         uint4 vv4 = ok ? *reinterpret_cast<const uint4*>(v + goff)
                        : make_uint4(0, 0, 0, 0);
+        if (VARIANT == 3) {  // keep the loads, skip the transpose writes
+          asm volatile("" :: "v"(vv4.x), "v"(vv4.y));
+          continue;
+        }
         const unsigned short* vs =
             reinterpret_cast<const unsigned short*>(&vv4);
         #pragma unroll
@@ -147,10 +155,34 @@ void attn_prefill_kernel(bf16_t* __restrict__ out,      // [T, HQ, D]
           // kdim byte offset = (c*32 + lhi*8)*2 -> chunk c*4 + lhi
           const uint4 kb4 = k_tile[swz_chunk<D * 2>(key, c * 4 + lhi)];
           const short8 kb = *reinterpret_cast<const short8*>(&kb4);
-          acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(qa[c], kb, acc, 0, 0, 0);
+          if (VARIANT == 4) {
+            asm volatile("" :: "v"(kb4.x));  // keep the ds_read alive
+          } else {
+            acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(qa[c], kb, acc,
+                                                          0, 0, 0);
+          }
         }
         sfrag[sc] = acc;
       }
+      if (VARIANT == 1) {
+        // skip the softmax VALU: p := raw scores (timing-only), l := +16
+        unsigned short* p16x = reinterpret_cast<unsigned short*>(p_t[wave]);
+        #pragma unroll
+        for (int sc = 0; sc < 2; ++sc)
+          #pragma unroll
+          for (int i = 0; i < 4; ++i) {
+            const int qrow = lhi * 4 + i;
+            const int kk = sc * 16 + llo;
+            bf16_t pb = f2bf(sfrag[sc][i]);
+            const int chunk = swz_chunk<KCHUNK * 2>(qrow, (kk * 2) / 16);
+            p16x[chunk * 8 + (kk % 8)] =
+                *reinterpret_cast<unsigned short*>(&pb);
+          }
+        #pragma unroll
+        for (int i = 0; i < 4; ++i) l[i] += 16.f;
+        goto pv_phase;
+      }
+      {
       // ---- mask + online softmax (C layout) -------------------------
       float pm[2][4];
       float rowmax[4];
@@ -212,6 +244,12 @@ void attn_prefill_kernel(bf16_t* __restrict__ out,      // [T, HQ, D]
         #pragma unroll
         for (int i = 0; i < 4; ++i) o_acc[c][i] *= alpha[i];
       }
+      }
+      pv_phase:
+      if (VARIANT == 2) {
+        asm volatile("" :: "v"(sfrag[0][0]));  // keep QK^T alive
+        continue;
+      }
       // ---- P·V: A-frag = one uint4 from p_t; B-frag = one uint4 from vT
       const uint4 pa4 = p_t[wave][swz_chunk<KCHUNK * 2>(llo, lhi)];
       const short8 pa = *reinterpret_cast<const short8*>(&pa4);
@@ -258,6 +296,29 @@ void launch_attn_prefill(bf16_t* out, const bf16_t* q, const bf16_t* k,
 }
 
 bool attn_prefill_supported(int d) { return d == 128 || d == 64; }
+
+// timing-only ablation probe (D=128): variant semantics in the kernel header
+void launch_attn_prefill_probe(int variant, bf16_t* out, const bf16_t* q,
+                               const bf16_t* k, const bf16_t* v,
+                               const int* seq_start, int num_seqs,
+                               int max_len, int hq, int hkv, float scale,
+                               hipStream_t stream) {
+  const int tiles = (max_len + QTILE - 1) / QTILE;
+  dim3 grid(num_seqs, hq, tiles);
+  dim3 block(PF_WAVES * WAVE);
+  switch (variant) {
+    case 1:hipLaunchKernelGGL(( attn_prefill_kernel<128, 1>), dim3(grid), dim3(block), 0, stream, 
+        out, q, k, v, seq_start, hq, hkv, scale); break;
+    case 2:hipLaunchKernelGGL(( attn_prefill_kernel<128, 2>), dim3(grid), dim3(block), 0, stream, 
+        out, q, k, v, seq_start, hq, hkv, scale); break;
+    case 3:hipLaunchKernelGGL(( attn_prefill_kernel<128, 3>), dim3(grid), dim3(block), 0, stream, 
+        out, q, k, v, seq_start, hq, hkv, scale); break;
+    case 4:hipLaunchKernelGGL(( attn_prefill_kernel<128, 4>), dim3(grid), dim3(block), 0, stream, 
+        out, q, k, v, seq_start, hq, hkv, scale); break;
+    default:hipLaunchKernelGGL(( attn_prefill_kernel<128, 0>), dim3(grid), dim3(block), 0, stream, 
+        out, q, k, v, seq_start, hq, hkv, scale); break;
+  }
+}
 
 // ---------------------------------------------------------------------------
 // mfma_probe: D[16,16] = A[16,32]·B[32,16] via one mfma_f32_16x16x32_bf16 —

@@ -19,6 +19,7 @@ int paged_attn_nsplit(int, int, int);
 bool paged_attn_decode_supported(int, int);
 void launch_attn_prefill(bf16_t*, const bf16_t*, const bf16_t*, const bf16_t*, const int*, int, int, int, int, int, float, hipStream_t);
 bool attn_prefill_supported(int);
+void launch_attn_prefill_probe(int, bf16_t*, const bf16_t*, const bf16_t*, const bf16_t*, const int*, int, int, int, int, float, hipStream_t);
 void launch_mfma_probe(float*, const bf16_t*, const bf16_t*, hipStream_t);
 void launch_softmax_sample(long*, const bf16_t*, const float*, const int*, const float*, const float*, int, int, hipStream_t);
 void launch_cross_entropy(float*, bf16_t*, const bf16_t*, const long*, long, int, long, float, hipStream_t);
@@ -187,6 +188,20 @@ void attn_prefill(torch::Tensor out, torch::Tensor q, torch::Tensor k,
   post_launch_check("attn_prefill");
 }
 
+void attn_prefill_probe(long variant, torch::Tensor out, torch::Tensor q,
+                        torch::Tensor k, torch::Tensor v,
+                        torch::Tensor seq_start, long max_len, double scale) {
+  check_bf16(out, "out"); check_bf16(q, "q"); check_bf16(k, "k");
+  check_bf16(v, "v");
+  int hq = q.size(1), hkv = k.size(1);
+  TORCH_CHECK(q.size(2) == 128, "probe is D=128 only");
+  launch_attn_prefill_probe((int)variant, bf(out), bfc(q), bfc(k), bfc(v),
+                            seq_start.data_ptr<int>(),
+                            seq_start.size(0) - 1, (int)max_len, hq, hkv,
+                            (float)scale, stream());
+  post_launch_check("attn_prefill_probe");
+}
+
 torch::Tensor mfma_probe(torch::Tensor a, torch::Tensor b) {
   check_bf16(a, "a"); check_bf16(b, "b");
   TORCH_CHECK(a.size(0) == 16 && a.size(1) == 32 && b.size(0) == 32 &&
@@ -310,6 +325,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("paged_attn_decode", &paged_attn_decode);
   m.def("attn_prefill", &attn_prefill);
   m.def("mfma_probe", &mfma_probe);
+  m.def("attn_prefill_probe", &attn_prefill_probe);
   m.def("softmax_sample", &softmax_sample);
   m.def("cross_entropy_fwd_bwd", &cross_entropy_fwd_bwd);
   m.def("adamw_step", &adamw_step);

@@ -21,6 +21,14 @@
 //
 // M is padded to 256 by clamping row indices (extra rows compute garbage
 // that the combine kernel never reads).
+//
+// STATUS: correct (parity tests) but NOT wired into the model — the
+// within-probe A/B (scripts/microbench.py --which dgemm, MI355X) measured
+// 0.43-0.70 TB/s of weight stream vs hipBLASLt's 1.75-5.95 TB/s on the same
+// shapes: the 2-barrier-per-32-K-step structure is staging/barrier-bound
+// (guide §5: that structure caps far below the 8-phase schedule, which is
+// what the library effectively has). Kept as the measured baseline for a
+// future 8-phase/glds variant; hipBLASLt remains the production path.
 #include "common.h"
 
 typedef __attribute__((ext_vector_type(8))) short short8;
