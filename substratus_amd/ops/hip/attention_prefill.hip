@@ -30,7 +30,8 @@ typedef __attribute__((ext_vector_type(4))) float float4v;
 
 #define PF_WAVES 4
 #define QTILE 64   // q rows per workgroup (16 per wave)
-#define KCHUNK 32  // keys per iteration
+#define KCHUNK 64  // keys per iteration (64: half the barriers of 32,
+                   // double the MFMA per phase — guide: BK 32->64 +7-16%)
 
 // XOR-swizzled 16-B-chunk address inside a row-major LDS tile.
 // row_bytes must be a power-of-two multiple of 16.
@@ -143,10 +144,11 @@ void attn_prefill_kernel(bf16_t* __restrict__ out,      // [T, HQ, D]
 
     const bool compute = wave_live && kbase < kmax_wave;
     if (compute) {
-      // ---- QK^T: two 16-key subchunks (B-frags from swizzled k_tile) --
-      float4v sfrag[2];
+      // ---- QK^T: KCHUNK/16 16-key subchunks (B-frags from k_tile) -----
+      constexpr int NSC = KCHUNK / 16;
+      float4v sfrag[NSC];
       #pragma unroll
-      for (int sc = 0; sc < 2; ++sc) {
+      for (int sc = 0; sc < NSC; ++sc) {
         float4v acc = {0.f, 0.f, 0.f, 0.f};
         const int key = sc * 16 + llo;   // B col = this lane's key row
         #pragma unroll
@@ -167,7 +169,7 @@ void attn_prefill_kernel(bf16_t* __restrict__ out,      // [T, HQ, D]
         // skip the softmax VALU: p := raw scores (timing-only), l := +16
         unsigned short* p16x = reinterpret_cast<unsigned short*>(p_t[wave]);
         #pragma unroll
-        for (int sc = 0; sc < 2; ++sc)
+        for (int sc = 0; sc < NSC; ++sc)
           #pragma unroll
           for (int i = 0; i < 4; ++i) {
             const int qrow = lhi * 4 + i;
@@ -183,12 +185,12 @@ void attn_prefill_kernel(bf16_t* __restrict__ out,      // [T, HQ, D]
       }
       {
       // ---- mask + online softmax (C layout) -------------------------
-      float pm[2][4];
+      float pm[NSC][4];
       float rowmax[4];
       #pragma unroll
       for (int i = 0; i < 4; ++i) rowmax[i] = -INFINITY;
       #pragma unroll
-      for (int sc = 0; sc < 2; ++sc) {
+      for (int sc = 0; sc < NSC; ++sc) {
         #pragma unroll
         for (int i = 0; i < 4; ++i) {
           const int qpos = q0 + lhi * 4 + i;
@@ -216,7 +218,7 @@ void attn_prefill_kernel(bf16_t* __restrict__ out,      // [T, HQ, D]
       float rowsum[4] = {0.f, 0.f, 0.f, 0.f};
       unsigned short* p16 = reinterpret_cast<unsigned short*>(p_t[wave]);
       #pragma unroll
-      for (int sc = 0; sc < 2; ++sc) {
+      for (int sc = 0; sc < NSC; ++sc) {
         #pragma unroll
         for (int i = 0; i < 4; ++i) {
           float pv = (pm[sc][i] == -INFINITY || m[i] == -INFINITY)
@@ -249,16 +251,21 @@ void attn_prefill_kernel(bf16_t* __restrict__ out,      // [T, HQ, D]
         asm volatile("" :: "v"(sfrag[0][0]));  // keep QK^T alive
         continue;
       }
-      // ---- P·V: A-frag = one uint4 from p_t; B-frag = one uint4 from vT
-      const uint4 pa4 = p_t[wave][swz_chunk<KCHUNK * 2>(llo, lhi)];
-      const short8 pa = *reinterpret_cast<const short8*>(&pa4);
+      // ---- P·V over KCHUNK keys in 32-k MFMA steps; A/B frags are one
+      // uint4 each from the swizzled p_t / vT tiles
       #pragma unroll
-      for (int c = 0; c < DC16; ++c) {
-        const int drow = c * 16 + llo;   // B col = output dim
-        const uint4 vb4 = vT[swz_chunk<KCHUNK * 2>(drow, lhi)];
-        const short8 vb = *reinterpret_cast<const short8*>(&vb4);
-        o_acc[c] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(pa, vb, o_acc[c],
-                                                           0, 0, 0);
+      for (int ks = 0; ks < KCHUNK / 32; ++ks) {
+        const uint4 pa4 =
+            p_t[wave][swz_chunk<KCHUNK * 2>(llo, ks * 4 + lhi)];
+        const short8 pa = *reinterpret_cast<const short8*>(&pa4);
+        #pragma unroll
+        for (int c = 0; c < DC16; ++c) {
+          const int drow = c * 16 + llo;   // B col = output dim
+          const uint4 vb4 = vT[swz_chunk<KCHUNK * 2>(drow, ks * 4 + lhi)];
+          const short8 vb = *reinterpret_cast<const short8*>(&vb4);
+          o_acc[c] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              pa, vb, o_acc[c], 0, 0, 0);
+        }
       }
     }
   }
