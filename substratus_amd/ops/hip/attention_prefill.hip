@@ -26,7 +26,16 @@
 #include "common.h"
 
 typedef __attribute__((ext_vector_type(8))) short short8;
+typedef __attribute__((ext_vector_type(4))) short short4v;
 typedef __attribute__((ext_vector_type(4))) float float4v;
+
+// ds_read_b64_tr_b16 (guide T10): given per-lane LDS address
+// base + ((l>>4)*64 + (l&15)) elems into a [16][16] bf16 image, lane l
+// receives column (l&15) of rows (l>>4)*4 + j — a free 4x16 transpose.
+DEV_INLINE short4v tr16_read(const short* p) {
+  return __builtin_amdgcn_ds_read_tr16_b64_v4i16(
+      (__attribute__((address_space(3))) short4v*)p);
+}
 
 #define PF_WAVES 4
 #define QTILE 64   // q rows per workgroup (16 per wave)
@@ -92,12 +101,17 @@ void attn_prefill_kernel(bf16_t* __restrict__ out,      // [T, HQ, D]
   #pragma unroll
   for (int c = 0; c < DC16; ++c) o_acc[c] = (float4v){0.f, 0.f, 0.f, 0.f};
 
-  // LDS tiles, all uint4-granular and XOR-swizzled:
-  //   k_tile: [KCHUNK rows][D cols]        — QK^T B-fragments
-  //   vT:     [D rows][KCHUNK cols]        — P·V B-fragments (transposed!)
-  //   p_t:    per wave [16 rows][KCHUNK]   — P C→A transpose
+  // LDS tiles:
+  //   k_tile: [KCHUNK rows][D cols], XOR-swizzled — QK^T B-fragments
+  //   v_img:  row-major [16][16] tr-read images, rows k-interleaved
+  //           ((k/8)*4 + k%4, k%8>=4 selects the sibling image) so TWO
+  //           ds_read_b64_tr_b16 deliver a P·V B-fragment — staging stays
+  //           ONE coalesced uint4 write per loaded uint4 (the 8-scalar-write
+  //           transpose was 35% of kernel time, profiles/
+  //           prefill_ablation_r01.txt)
+  //   p_t:    per wave [16 rows][KCHUNK], XOR-swizzled — P C→A transpose
   __shared__ uint4 k_tile[KCHUNK * D * 2 / 16];
-  __shared__ uint4 vT[D * KCHUNK * 2 / 16];
+  __shared__ short v_img[DC16][KCHUNK / 32][2][16][16];
   __shared__ uint4 p_t[PF_WAVES][16 * KCHUNK * 2 / 16];
 
   const int kmax_tile = min(q0_tile + QTILE, n);       // causal bound of tile
@@ -124,19 +138,17 @@ void attn_prefill_kernel(bf16_t* __restrict__ out,      // [T, HQ, D]
         // V: transpose into vT[d][k] (8 scalar writes of this uint4's dims)
         uint4 vv4 = ok ? *reinterpret_cast<const uint4*>(v + goff)
                        : make_uint4(0, 0, 0, 0);
-        if (VARIANT == 3) {  // keep the loads, skip the transpose writes
+        if (VARIANT == 3) {  // keep the loads, skip the image writes
           asm volatile("" :: "v"(vv4.x), "v"(vv4.y));
           continue;
         }
-        const unsigned short* vs =
-            reinterpret_cast<const unsigned short*>(&vv4);
-        #pragma unroll
-        for (int j = 0; j < 8; ++j) {
-          const int d = col + j;
-          // byte address inside vT row d: k index `row` -> byte row*2
-          unsigned short* vt16 = reinterpret_cast<unsigned short*>(vT);
-          const int chunk = swz_chunk<KCHUNK * 2>(d, (row * 2) / 16);
-          vt16[chunk * 8 + (row % 8)] = vs[j];
+        // one uint4 into the tr-read image: c = col/16, half = col%16
+        {
+          const int c = col / 16, off = col % 16;   // off in {0, 8}
+          const int ks = row / 32, kk = row % 32;
+          const int sel = (kk % 8) >= 4;
+          const int r = (kk / 8) * 4 + (kk % 4);
+          *reinterpret_cast<uint4*>(&v_img[c][ks][sel][r][off]) = vv4;
         }
       }
     }
@@ -260,9 +272,14 @@ void attn_prefill_kernel(bf16_t* __restrict__ out,      // [T, HQ, D]
         const short8 pa = *reinterpret_cast<const short8*>(&pa4);
         #pragma unroll
         for (int c = 0; c < DC16; ++c) {
-          const int drow = c * 16 + llo;   // B col = output dim
-          const uint4 vb4 = vT[swz_chunk<KCHUNK * 2>(drow, ks * 4 + lhi)];
-          const short8 vb = *reinterpret_cast<const short8*>(&vb4);
+          // two hardware-transpose reads assemble the B-frag (k = lhi*8+j)
+          const short4v lo =
+              tr16_read(&v_img[c][ks][0][lhi * 4][llo]);
+          const short4v hi =
+              tr16_read(&v_img[c][ks][1][lhi * 4][llo]);
+          short8 vb;
+          vb[0] = lo[0]; vb[1] = lo[1]; vb[2] = lo[2]; vb[3] = lo[3];
+          vb[4] = hi[0]; vb[5] = hi[1]; vb[6] = hi[2]; vb[7] = hi[3];
           o_acc[c] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
               pa, vb, o_acc[c], 0, 0, 0);
         }
