@@ -273,11 +273,17 @@ void attn_prefill_kernel(bf16_t* __restrict__ out,      // [T, HQ, D]
         const short8 pa = *reinterpret_cast<const short8*>(&pa4);
         #pragma unroll
         for (int c = 0; c < DC16; ++c) {
-          // two hardware-transpose reads assemble the B-frag (k = lhi*8+j)
+          // two hardware-transpose reads assemble the B-frag (k = lhi*8+j).
+          // ds_read_b64_tr_b16 lane mapping (probed on gfx950, see
+          // gpurun_out/tr16_probe.txt): with per-lane address base+lane*4
+          // shorts into a row-major [16][16] image, out[l][j] =
+          // img[(l>>4)*4 + j][l&15] — so the address is lane*4 FLAT, not
+          // [lhi*4][llo] (whose 2-byte llo offset the 8B-aligned b64 load
+          // would silently mask off).
           const short4v lo =
-              tr16_read(&v_img[c][ks][0][lhi * 4][llo]);
+              tr16_read(&v_img[c][ks][0][0][0] + lane * 4);
           const short4v hi =
-              tr16_read(&v_img[c][ks][1][lhi * 4][llo]);
+              tr16_read(&v_img[c][ks][1][0][0] + lane * 4);
           short8 vb;
           vb[0] = lo[0]; vb[1] = lo[1]; vb[2] = lo[2]; vb[3] = lo[3];
           vb[4] = hi[0]; vb[5] = hi[1]; vb[6] = hi[2]; vb[7] = hi[3];

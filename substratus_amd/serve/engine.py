@@ -75,6 +75,13 @@ class Engine:
             quantize_model(model, cfg.quant)
         self.model = model.eval()
         n_kv_local = self.model.n_kv_heads
+        kv_dtype = cfg.dtype
+        if cfg.kv_dtype == "fp8":
+            assert self.device.type == "cuda", "fp8 KV cache is GPU-only"
+            kv_dtype = torch.float8_e4m3fn
+        elif cfg.kv_dtype is not None:
+            raise ValueError(f"unknown kv_dtype {cfg.kv_dtype!r}")
+        kv_bytes_per_el = 1 if kv_dtype == torch.float8_e4m3fn else 2
         if cfg.kv_blocks is not None:
             num_blocks = cfg.kv_blocks
         elif self.device.type == "cuda":
@@ -89,13 +96,6 @@ class Engine:
         else:
             num_blocks = cfg.max_batch * (
                 (cfg.max_seq_len + cfg.block_size - 1) // cfg.block_size)
-        kv_dtype = cfg.dtype
-        if cfg.kv_dtype == "fp8":
-            assert self.device.type == "cuda", "fp8 KV cache is GPU-only"
-            kv_dtype = torch.float8_e4m3fn
-        elif cfg.kv_dtype is not None:
-            raise ValueError(f"unknown kv_dtype {cfg.kv_dtype!r}")
-        kv_bytes_per_el = 1 if kv_dtype == torch.float8_e4m3fn else 2
         self.kv = PagedKVCache(self.mcfg, n_kv_local, num_blocks,
                                cfg.block_size, device=self.device,
                                dtype=kv_dtype)
