@@ -48,7 +48,38 @@ def parse_args():
                    help="GGUF weight quantization: q4_0 | q8_0 (config 4)")
     p.add_argument("--tp", type=int, default=1,
                    help="tensor-parallel degree (serve); default DP replicas")
+    p.add_argument("--profile", default=None, metavar="PATH",
+                   help="profile the timed region with torch.profiler and "
+                        "write a kernel table + chrome trace to PATH "
+                        "(SURVEY.md §5: bake a --profile mode into the data "
+                        "plane; complements rocprofv3)")
     return p.parse_args()
+
+
+def profiled(path):
+    """Context wrapping the timed loop in torch.profiler when --profile is
+    set (no-op otherwise). Writes PATH.txt (kernel table by device time) and
+    PATH.json (chrome trace) on exit — rank 0 only."""
+    import contextlib
+    if path is None or int(os.environ.get("RANK", "0")) != 0:
+        return contextlib.nullcontext()
+
+    @contextlib.contextmanager
+    def _ctx():
+        acts = [torch.profiler.ProfilerActivity.CPU]
+        if torch.cuda.is_available():
+            acts.append(torch.profiler.ProfilerActivity.CUDA)
+        with torch.profiler.profile(activities=acts) as prof:
+            yield
+        table = prof.key_averages().table(
+            sort_by="self_cuda_time_total" if torch.cuda.is_available()
+            else "self_cpu_time_total", row_limit=40)
+        with open(path + ".txt", "w") as f:
+            f.write(table)
+        prof.export_chrome_trace(path + ".json")
+        print(f"[bench] profile written to {path}.txt / {path}.json",
+              flush=True)
+    return _ctx()
 
 
 def log(msg: str) -> None:
@@ -103,10 +134,11 @@ def bench_serve(args) -> None:
         torch.cuda.synchronize()
     barrier()
     t0 = time.time()
-    for _ in range(args.steps):
-        eng.decode_only_step()
-    if on_gpu:
-        torch.cuda.synchronize()
+    with profiled(args.profile):
+        for _ in range(args.steps):
+            eng.decode_only_step()
+        if on_gpu:
+            torch.cuda.synchronize()
     barrier()
     elapsed = time.time() - t0
     # max over ranks
