@@ -197,21 +197,40 @@ void attn_prefill_kernel(bf16_t* __restrict__ out,      // [T, HQ, D]
         goto pv_phase;
       }
       {
-      // ---- mask + online softmax (C layout) -------------------------
+      // ---- mask + online softmax (C layout), log2 domain --------------
+      // All exponentials run as exp2 with scale*log2(e) folded into the
+      // score once (saves a VALU mul per element before every v_exp_f32);
+      // chunks entirely below the causal diagonal (kbase+KCHUNK-1 <= q0,
+      // in-bounds) skip the per-element mask compare+select.
+      const float scale2 = scale * 1.44269504088896340736f;  // log2(e)
       float pm[NSC][4];
       float rowmax[4];
       #pragma unroll
       for (int i = 0; i < 4; ++i) rowmax[i] = -INFINITY;
-      #pragma unroll
-      for (int sc = 0; sc < NSC; ++sc) {
+      if (kbase + KCHUNK - 1 <= q0 && kbase + KCHUNK <= n) {
         #pragma unroll
-        for (int i = 0; i < 4; ++i) {
-          const int qpos = q0 + lhi * 4 + i;
+        for (int sc = 0; sc < NSC; ++sc) {
+          #pragma unroll
+          for (int i = 0; i < 4; ++i) {
+            const float sv = sfrag[sc][i] * scale2;
+            pm[sc][i] = sv;
+            rowmax[i] = fmaxf(rowmax[i], sv);
+          }
+        }
+      } else {
+        int klim[4];
+        #pragma unroll
+        for (int i = 0; i < 4; ++i)
+          klim[i] = min(q0 + lhi * 4 + i, n - 1);
+        #pragma unroll
+        for (int sc = 0; sc < NSC; ++sc) {
           const int kpos = kbase + sc * 16 + llo;
-          float sv = (kpos <= qpos && kpos < n && qpos < n)
-                         ? sfrag[sc][i] * scale : -INFINITY;
-          pm[sc][i] = sv;
-          rowmax[i] = fmaxf(rowmax[i], sv);
+          #pragma unroll
+          for (int i = 0; i < 4; ++i) {
+            float sv = (kpos <= klim[i]) ? sfrag[sc][i] * scale2 : -INFINITY;
+            pm[sc][i] = sv;
+            rowmax[i] = fmaxf(rowmax[i], sv);
+          }
         }
       }
       #pragma unroll
@@ -224,7 +243,7 @@ void attn_prefill_kernel(bf16_t* __restrict__ out,      // [T, HQ, D]
       #pragma unroll
       for (int i = 0; i < 4; ++i) {
         const float mn = fmaxf(m[i], rowmax[i]);
-        alpha[i] = (m[i] == -INFINITY) ? 0.f : __expf(m[i] - mn);
+        alpha[i] = (m[i] == -INFINITY) ? 0.f : __builtin_amdgcn_exp2f(m[i] - mn);
         m[i] = mn;
       }
       // P = exp(S - m); row-sum; write transposed into swizzled p_t
@@ -235,7 +254,7 @@ void attn_prefill_kernel(bf16_t* __restrict__ out,      // [T, HQ, D]
         #pragma unroll
         for (int i = 0; i < 4; ++i) {
           float pv = (pm[sc][i] == -INFINITY || m[i] == -INFINITY)
-                         ? 0.f : __expf(pm[sc][i] - m[i]);
+                         ? 0.f : __builtin_amdgcn_exp2f(pm[sc][i] - m[i]);
           rowsum[i] += pv;
           const int qrow = lhi * 4 + i;          // P row
           const int kk = sc * 16 + llo;          // P col
