@@ -109,10 +109,17 @@ void attn_prefill_kernel(bf16_t* __restrict__ out,      // [T, HQ, D]
   //           ONE coalesced uint4 write per loaded uint4 (the 8-scalar-write
   //           transpose was 35% of kernel time, profiles/
   //           prefill_ablation_r01.txt)
-  //   p_t:    per wave [16 rows][KCHUNK], XOR-swizzled — P C→A transpose
+  //   p_img:  per wave, the SAME k-interleaved tr-read image layout as
+  //           v_img but storing P TRANSPOSED (img[r(k)][qrow]): a lane's 4
+  //           C-fragment rows (qrow = lhi*4+i) are CONSECUTIVE shorts, so
+  //           the whole P store is ONE packed b64 write per 16-key subchunk
+  //           (was 16 scalar ds_write_b16), and the P·V A-fragment is two
+  //           ds_read_b64_tr_b16 — same probed mapping as the V B-frags
+  //           (tr16 out[l][j] = img[(l>>4)*4+j][l&15] = P^T[k][qrow] gives
+  //           exactly A[row=llo][k=lhi*8+sel*4+j])
   __shared__ uint4 k_tile[KCHUNK * D * 2 / 16];
   __shared__ short v_img[DC16][KCHUNK / 32][2][16][16];
-  __shared__ uint4 p_t[PF_WAVES][16 * KCHUNK * 2 / 16];
+  __shared__ short p_img[PF_WAVES][KCHUNK / 32][2][16][16];
 
   const int kmax_tile = min(q0_tile + QTILE, n);       // causal bound of tile
   const int nchunks = (kmax_tile + KCHUNK - 1) / KCHUNK;
@@ -179,18 +186,19 @@ void attn_prefill_kernel(bf16_t* __restrict__ out,      // [T, HQ, D]
       }
       if (VARIANT == 1) {
         // skip the softmax VALU: p := raw scores (timing-only), l := +16
-        unsigned short* p16x = reinterpret_cast<unsigned short*>(p_t[wave]);
         #pragma unroll
-        for (int sc = 0; sc < NSC; ++sc)
-          #pragma unroll
-          for (int i = 0; i < 4; ++i) {
-            const int qrow = lhi * 4 + i;
-            const int kk = sc * 16 + llo;
-            bf16_t pb = f2bf(sfrag[sc][i]);
-            const int chunk = swz_chunk<KCHUNK * 2>(qrow, (kk * 2) / 16);
-            p16x[chunk * 8 + (kk % 8)] =
-                *reinterpret_cast<unsigned short*>(&pb);
-          }
+        for (int sc = 0; sc < NSC; ++sc) {
+          unsigned plo_, phi_;
+          asm("v_cvt_pk_bf16_f32 %0, %1, %2"
+              : "=v"(plo_) : "v"(sfrag[sc][0]), "v"(sfrag[sc][1]));
+          asm("v_cvt_pk_bf16_f32 %0, %1, %2"
+              : "=v"(phi_) : "v"(sfrag[sc][2]), "v"(sfrag[sc][3]));
+          const int kf = sc * 16 + llo, kk = kf & 31;
+          *reinterpret_cast<uint2*>(
+              &p_img[wave][kf >> 5][(kk % 8) >= 4]
+                    [(kk / 8) * 4 + (kk % 4)][lhi * 4]) =
+              make_uint2(plo_, phi_);
+        }
         #pragma unroll
         for (int i = 0; i < 4; ++i) l[i] += 16.f;
         goto pv_phase;
@@ -245,23 +253,28 @@ void attn_prefill_kernel(bf16_t* __restrict__ out,      // [T, HQ, D]
         alpha[i] = (m[i] == -INFINITY) ? 0.f : __builtin_amdgcn_exp2f(m[i] - mn);
         m[i] = mn;
       }
-      // P = exp(S - m); row-sum; write transposed into swizzled p_t
+      // P = exp2(S - m); row-sum; one packed b64 write per subchunk into
+      // the transposed tr-read image (lane's 4 rows are consecutive shorts)
       float rowsum[4] = {0.f, 0.f, 0.f, 0.f};
-      unsigned short* p16 = reinterpret_cast<unsigned short*>(p_t[wave]);
       #pragma unroll
       for (int sc = 0; sc < NSC; ++sc) {
+        float pv4[4];
         #pragma unroll
         for (int i = 0; i < 4; ++i) {
-          float pv = (pm[sc][i] == -INFINITY || m[i] == -INFINITY)
-                         ? 0.f : __builtin_amdgcn_exp2f(pm[sc][i] - m[i]);
-          rowsum[i] += pv;
-          const int qrow = lhi * 4 + i;          // P row
-          const int kk = sc * 16 + llo;          // P col
-          bf16_t pb = f2bf(pv);
-          const int chunk = swz_chunk<KCHUNK * 2>(qrow, (kk * 2) / 16);
-          p16[chunk * 8 + (kk % 8)] =
-              *reinterpret_cast<unsigned short*>(&pb);
+          pv4[i] = (pm[sc][i] == -INFINITY || m[i] == -INFINITY)
+                       ? 0.f : __builtin_amdgcn_exp2f(pm[sc][i] - m[i]);
+          rowsum[i] += pv4[i];
         }
+        unsigned plo_, phi_;
+        asm("v_cvt_pk_bf16_f32 %0, %1, %2"
+            : "=v"(plo_) : "v"(pv4[0]), "v"(pv4[1]));
+        asm("v_cvt_pk_bf16_f32 %0, %1, %2"
+            : "=v"(phi_) : "v"(pv4[2]), "v"(pv4[3]));
+        const int kf = sc * 16 + llo, kk = kf & 31;
+        *reinterpret_cast<uint2*>(
+            &p_img[wave][kf >> 5][(kk % 8) >= 4]
+                  [(kk / 8) * 4 + (kk % 4)][lhi * 4]) =
+            make_uint2(plo_, phi_);
       }
       #pragma unroll
       for (int i = 0; i < 4; ++i) {
@@ -282,13 +295,16 @@ void attn_prefill_kernel(bf16_t* __restrict__ out,      // [T, HQ, D]
         asm volatile("" :: "v"(sfrag[0][0]));  // keep QK^T alive
         continue;
       }
-      // ---- P·V over KCHUNK keys in 32-k MFMA steps; A/B frags are one
-      // uint4 each from the swizzled p_t / vT tiles
+      // ---- P·V over KCHUNK keys in 32-k MFMA steps; A and B frags both
+      // come from tr-read images (two ds_read_b64_tr_b16 each)
       #pragma unroll
       for (int ks = 0; ks < KCHUNK / 32; ++ks) {
-        const uint4 pa4 =
-            p_t[wave][swz_chunk<KCHUNK * 2>(llo, ks * 4 + lhi)];
-        const short8 pa = *reinterpret_cast<const short8*>(&pa4);
+        // A-frag: P^T image → out[l][j] = P[qrow=llo][k=lhi*8+sel*4+j]
+        const short4v alo = tr16_read(&p_img[wave][ks][0][0][0] + lane * 4);
+        const short4v ahi = tr16_read(&p_img[wave][ks][1][0][0] + lane * 4);
+        short8 pa;
+        pa[0] = alo[0]; pa[1] = alo[1]; pa[2] = alo[2]; pa[3] = alo[3];
+        pa[4] = ahi[0]; pa[5] = ahi[1]; pa[6] = ahi[2]; pa[7] = ahi[3];
         #pragma unroll
         for (int c = 0; c < DC16; ++c) {
           // two hardware-transpose reads assemble the B-frag (k = lhi*8+j).
