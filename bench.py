@@ -36,7 +36,7 @@ def parse_args():
     p.add_argument("--warmup", type=int, default=16)
     p.add_argument("--mode", choices=["serve", "train"], default="serve")
     p.add_argument("--model", default=None, help="override model config")
-    p.add_argument("--batch", type=int, default=256)
+    p.add_argument("--batch", type=int, default=512)
     p.add_argument("--prompt-len", type=int, default=512)
     p.add_argument("--micro-batch", type=int, default=4,
                    help="per-GPU micro batch (train mode)")
