@@ -128,6 +128,39 @@ def bench_decode_gemm(m=256):
               f"{dt_blas*1e6:.1f} us ({wb/dt_blas/1e12:.2f} TB/s)")
 
 
+def bench_wstream(m=256):
+    """wstream GEMM v2 vs hipBLASLt, interleaved rounds (guide rule 24),
+    plus a parity check against fp32 matmul. Shapes: llama2-7b decode
+    projections + the logits GEMM."""
+    import torch.nn.functional as F
+    from substratus_amd import ops
+    from substratus_amd import _C
+    shapes = [(12288, 4096), (4096, 4096), (22016, 4096), (4096, 11008),
+              (32000, 4096)]
+    for n, k in shapes:
+        torch.manual_seed(0)
+        x = (torch.randn(m, k, device="cuda") * 0.3).bfloat16()
+        w = (torch.randn(n, k, device="cuda") * 0.02).bfloat16()
+        wp = ops.wstream_repack(w)
+        y = ops.wstream_linear(x, wp, n, k)
+        refy = (x.float() @ w.float().T)
+        err = (y.float() - refy).abs().max().item()
+        scale = refy.abs().max().item()
+        ok = err <= 0.02 * max(scale, 1.0)
+        # interleaved timing rounds
+        ours, blas = [], []
+        for _ in range(5):
+            ours.append(timeit(lambda: ops.wstream_linear(x, wp, n, k),
+                               iters=50))
+            blas.append(timeit(lambda: F.linear(x, w), iters=50))
+        do, db = min(ours), min(blas)
+        wb = n * k * 2
+        print(f"wstream M={m} N={n} K={k}: ours {do*1e6:6.1f} us "
+              f"({wb/do/1e12:.2f} TB/s W) vs hipBLASLt {db*1e6:6.1f} us "
+              f"({wb/db/1e12:.2f} TB/s)  x{db/do:.2f}  "
+              f"parity={'OK' if ok else f'FAIL err={err:.4f}/{scale:.1f}'}")
+
+
 def bench_quant_gemv(n=13824, k=5120, quant="q4_0"):
     from substratus_amd.parallel.tp import Linear
     from substratus_amd.models.quant import QuantLinear
@@ -165,6 +198,9 @@ if __name__ == "__main__":
     if a.which in ("all", "dgemm"):
         bench_decode_gemm()
         bench_decode_gemm(m=64)
+    if a.which in ("all", "wstream"):
+        bench_wstream()
+        bench_wstream(m=64)
     if a.which in ("all", "quant"):
         bench_quant_gemv()
         bench_quant_gemv(quant="q8_0")

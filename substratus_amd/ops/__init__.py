@@ -177,3 +177,32 @@ def adamw_step(param: torch.Tensor, grad: torch.Tensor, exp_avg: torch.Tensor,
         return
     ext.adamw_step(param, grad, exp_avg, exp_avg_sq, lr, beta1, beta2,
                    eps, weight_decay, step)
+
+
+# ---- weight-streaming decode GEMM (ops/hip/wstream_gemm.hip) -------------
+
+def wstream_repack(w: torch.Tensor) -> torch.Tensor:
+    """Repack a torch Linear weight W[N, K] (bf16) into the MFMA
+    B-fragment-linear layout [N/16][K/32][64 lanes][8 bf16] the wstream
+    decode GEMM streams sequentially. Lane l = lhi*16+llo holds
+    W[n16*16 + llo][t*32 + lhi*8 + j] — one contiguous uint4 per fragment."""
+    n, k = w.shape
+    assert n % 16 == 0 and k % 32 == 0, (n, k)
+    return (w.view(n // 16, 16, k // 32, 4, 8)
+             .permute(0, 2, 3, 1, 4).contiguous().view(-1))
+
+
+def wstream_supported(m: int, n: int, k: int) -> bool:
+    return m <= 256 and n % 128 == 0 and k % 64 == 0
+
+
+def wstream_linear(x: torch.Tensor, wp: torch.Tensor, n: int,
+                   k: int) -> torch.Tensor:
+    """Y[M, N] = X[M, K] @ W^T with W pre-repacked by wstream_repack.
+    GPU-only (decode path); callers fall back to F.linear when unsupported."""
+    ext = _dispatch(x)
+    assert ext is not None, "wstream_linear is GPU-only"
+    m = x.shape[0]
+    y = torch.empty(m, n, dtype=x.dtype, device=x.device)
+    ext.wstream_gemm(y, x.contiguous(), wp, n, k)
+    return y

@@ -30,6 +30,9 @@ void launch_dequant(bf16_t*, const void*, const unsigned char*, long, int, bool,
 int decode_gemm_splitk(int, int);
 bool decode_gemm_supported(int, int, int);
 void launch_decode_gemm(bf16_t*, const bf16_t*, const bf16_t*, float*, int, int, int, int, hipStream_t);
+int wstream_splitk(int, int);
+bool wstream_supported(int, int, int);
+void launch_wstream_gemm(bf16_t*, const bf16_t*, const bf16_t*, float*, int, int, int, int, hipStream_t);
 
 namespace {
 
@@ -309,11 +312,38 @@ bool decode_gemm_ok(long m, long n, long k) {
   return decode_gemm_supported((int)m, (int)n, (int)k);
 }
 
+// Y[M, N] = X[M, K] · W^T with W repacked fragment-linear
+// (ops.wstream_repack); wstream_gemm.hip
+void wstream_gemm(torch::Tensor y, torch::Tensor x, torch::Tensor wp,
+                  long n_, long k_) {
+  check_bf16(y, "y"); check_bf16(x, "x"); check_bf16(wp, "wp");
+  const int m = x.size(0), k = (int)k_, n = (int)n_;
+  TORCH_CHECK(x.size(1) == k && y.size(0) == m && y.size(1) == n);
+  TORCH_CHECK(wp.numel() == (long)n * k, "packed W size mismatch");
+  TORCH_CHECK(wstream_supported(m, n, k),
+              "wstream_gemm unsupported shape M=", m, " N=", n, " K=", k);
+  const int splitk = wstream_splitk(n, k);
+  torch::Tensor part;
+  float* pp = nullptr;
+  if (splitk > 1) {
+    part = torch::empty({splitk, 256, n}, x.options().dtype(torch::kFloat));
+    pp = part.data_ptr<float>();
+  }
+  launch_wstream_gemm(bf(y), bfc(x), bfc(wp), pp, m, n, k, splitk, stream());
+  post_launch_check("wstream_gemm");
+}
+
+bool wstream_ok(long m, long n, long k) {
+  return wstream_supported((int)m, (int)n, (int)k);
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("decode_gemm", &decode_gemm);
   m.def("decode_gemm_ok", &decode_gemm_ok);
+  m.def("wstream_gemm", &wstream_gemm);
+  m.def("wstream_ok", &wstream_ok);
   m.def("quant_gemv", &quant_gemv);
   m.def("dequant", &dequant);
   m.def("rms_norm", &rms_norm);
