@@ -126,3 +126,51 @@ def test_falcon_and_opt_families_run():
         while eng.has_work():
             toks.extend(o.token for o in eng.step() if o.seq_id == sid)
         assert len(toks) == 3, name
+
+
+def test_engine_churn_cpu():
+    """CPU version of the serving soak: staggered admissions, mixed
+    prompt/output lengths, every request yields exactly max_new tokens."""
+    import random
+    import torch
+    from substratus_amd.serve.engine import Engine, EngineConfig
+    rng = random.Random(5)
+    eng = Engine(EngineConfig(model="llama-tiny", dtype=torch.float32,
+                              max_batch=4, max_seq_len=96, kv_blocks=64,
+                              seed=11))
+    want, got, pending = {}, {}, 10
+    while pending > 0 or eng.has_work():
+        if pending > 0 and len(eng.waiting) + len(eng.running) < 4:
+            plen = rng.randint(1, 24)
+            mnew = rng.randint(1, 8)
+            sid = eng.add_request(
+                [rng.randrange(eng.mcfg.vocab_size) for _ in range(plen)],
+                max_new_tokens=mnew, temperature=rng.choice([0.0, 0.8]))
+            want[sid] = mnew
+            pending -= 1
+        for out in eng.step():
+            got[out.seq_id] = got.get(out.seq_id, 0) + 1
+    assert len(want) == 10
+    for sid, mnew in want.items():
+        assert got.get(sid, 0) == mnew
+
+
+def test_wstream_repack_layout():
+    """wstream_repack puts W[n16*16+llo][t*32+lhi*8+j] at packed index
+    ((n16*(K/32)+t)*64 + lhi*16+llo)*8 + j — the exact fragment order
+    ops/hip/wstream_gemm.hip streams (lane = lhi*16+llo)."""
+    import torch
+    from substratus_amd import ops
+    N, K = 32, 64
+    w = torch.arange(N * K, dtype=torch.float32).view(N, K).bfloat16()
+    p = ops.wstream_repack(w)
+    k32 = K // 32
+    for n16 in range(N // 16):
+        for t in range(k32):
+            for lhi in range(4):
+                for llo in range(16):
+                    lane = lhi * 16 + llo
+                    base = ((n16 * k32 + t) * 64 + lane) * 8
+                    for j in range(8):
+                        assert p[base + j] == w[n16 * 16 + llo][
+                            t * 32 + lhi * 8 + j]

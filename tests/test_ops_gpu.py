@@ -526,3 +526,51 @@ def test_wstream_gemm_matches_linear(m, n, k):
     y = ops.wstream_linear(x, wp, n, k)
     want = torch.nn.functional.linear(x.float().cpu(), w.float().cpu())
     assert_close_bf16(y, want, atol=5e-2, rtol=5e-2)
+
+
+def test_engine_kv_autosize_gpu():
+    """kv_blocks=None on GPU sizes the paged cache from free HBM
+    (regression: the auto-size branch once read kv dtype state that was
+    computed after it — engine.py)."""
+    from substratus_amd.serve.engine import Engine, EngineConfig
+    eng = Engine(EngineConfig(model="llama-tiny", dtype=torch.bfloat16,
+                              max_batch=4, max_seq_len=128, kv_blocks=None))
+    assert eng.kv.num_blocks >= 4 * (128 // 16)
+    sid = eng.add_request([1, 2, 3], max_new_tokens=4)
+    while eng.has_work():
+        eng.step()
+    assert sid == 0
+
+
+def test_engine_churn_gpu():
+    """Serving soak: staggered admissions with mixed lengths through the
+    graph-captured decode path — every request completes with the right
+    token count (continuous batching + admission + completion recycling)."""
+    import random
+    from substratus_amd.serve.engine import Engine, EngineConfig
+    rng = random.Random(5)
+    eng = Engine(EngineConfig(model="llama-tiny", dtype=torch.bfloat16,
+                              max_batch=8, max_seq_len=256, kv_blocks=256,
+                              seed=11))
+    want: dict[int, int] = {}
+    got: dict[int, int] = {}
+    pending = 40
+    while pending > 0 or eng.has_work():
+        if pending > 0 and len(eng.waiting) + len(eng.running) < 8:
+            n_new = rng.randint(1, 2)
+            for _ in range(n_new):
+                if pending == 0:
+                    break
+                plen = rng.randint(1, 48)
+                mnew = rng.randint(1, 24)
+                sid = eng.add_request(
+                    [rng.randrange(eng.mcfg.vocab_size) for _ in range(plen)],
+                    max_new_tokens=mnew,
+                    temperature=rng.choice([0.0, 0.8]))
+                want[sid] = mnew
+                pending -= 1
+        for out in eng.step():
+            got[out.seq_id] = got.get(out.seq_id, 0) + 1
+    assert len(want) == 40
+    for sid, mnew in want.items():
+        assert got.get(sid, 0) == mnew, (sid, mnew, got.get(sid))
