@@ -85,6 +85,12 @@ def layer_norm(x: torch.Tensor, weight: torch.Tensor, bias: torch.Tensor,
     return out
 
 
+def _rows_contiguous(t: torch.Tensor) -> bool:
+    """[n, h, d] with contiguous inner rows; the leading (token) stride may
+    be larger — a qkv-split view."""
+    return t.stride(2) == 1 and t.stride(1) == t.shape[2]
+
+
 def rope_kv_append(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
                    positions: torch.Tensor, k_cache: torch.Tensor,
                    v_cache: torch.Tensor, slots: torch.Tensor,
@@ -92,14 +98,17 @@ def rope_kv_append(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
     ext = _dispatch(q)
     if ext is None:
         return ref.rope_kv_append(q, k, v, positions, k_cache, v_cache, slots, theta)
-    q = q.contiguous()
-    if not k.is_contiguous():
-        # decode passes a qkv-split view; the in-place k rotation lands in a
-        # private copy there (decode never reads k again). Prefill passes a
-        # contiguous k and consumes the rotation.
+    # The kernel takes a token-row stride, so qkv-split VIEWS flow straight
+    # through (decode saves three elementwise copies per layer; rotation
+    # lands in-place inside the fused qkv buffer, which nothing re-reads).
+    # Only reshape if the inner [h, d] rows themselves aren't contiguous.
+    if not _rows_contiguous(q):
+        q = q.contiguous()
+    if not _rows_contiguous(k):
         k = k.contiguous()
-    ext.rope_kv_append(q, k, v.contiguous(),
-                       positions, k_cache, v_cache, slots, theta)
+    if not _rows_contiguous(v):
+        v = v.contiguous()
+    ext.rope_kv_append(q, k, v, positions, k_cache, v_cache, slots, theta)
     return q
 
 
@@ -109,8 +118,10 @@ def paged_attn_decode(q: torch.Tensor, k_cache: torch.Tensor,
     ext = _dispatch(q)
     if ext is None:
         return ref.paged_attn_decode(q, k_cache, v_cache, block_table, seq_lens, scale)
-    out = torch.empty_like(q)
-    ext.paged_attn_decode(out, q.contiguous(), k_cache, v_cache,
+    if not _rows_contiguous(q):
+        q = q.contiguous()
+    out = torch.empty(q.shape, dtype=q.dtype, device=q.device)
+    ext.paged_attn_decode(out, q, k_cache, v_cache,
                           block_table, seq_lens, scale)
     return out
 

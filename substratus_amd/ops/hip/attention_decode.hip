@@ -96,6 +96,7 @@ void paged_attn_decode_kernel(
     const int* __restrict__ block_table, // [B, max_blocks]
     const int* __restrict__ seq_lens,    // [B]
     int hq, int hkv, int max_blocks, float scale,
+    long q_rstride,                       // token-row stride in elements
     int nsplit,
     float* __restrict__ part_o,           // [B, HQ, NS, D]  (nsplit > 1)
     float* __restrict__ part_ml) {        // [B, HQ, NS, 2]
@@ -128,7 +129,7 @@ void paged_attn_decode_kernel(
     for (int j = 0; j < 8; ++j) qf[g][j] = 0.f;
     if (g < ng) {
       const uint4 u = reinterpret_cast<const uint4*>(
-          q + ((long)b * hq + h * group + g0 + g) * D)[ll];
+          q + (long)b * q_rstride + (long)(h * group + g0 + g) * D)[ll];
       float2 f0 = unpack_bf16x2(u.x), f1 = unpack_bf16x2(u.y);
       float2 f2 = unpack_bf16x2(u.z), f3 = unpack_bf16x2(u.w);
       qf[g][0] = f0.x; qf[g][1] = f0.y; qf[g][2] = f1.x; qf[g][3] = f1.y;
@@ -360,9 +361,9 @@ void launch_paged_attn_decode(bf16_t* out, const bf16_t* q,
                               const void* k_cache, const void* v_cache,
                               const int* block_table, const int* seq_lens,
                               int bsz, int hq, int hkv, int d, int max_blocks,
-                              int block_size, float scale, int nsplit,
-                              float* part_o, float* part_ml, bool kv8,
-                              hipStream_t stream) {
+                              int block_size, float scale, long q_rstride,
+                              int nsplit, float* part_o, float* part_ml,
+                              bool kv8, hipStream_t stream) {
   const int group = hq / hkv;
   // GC=8 when the GQA group tiles evenly by 8 (llama2-70b, llama3-style);
   // GC=1 otherwise (MHA group=1; falcon's prime group=71 — its MQA KV is
@@ -374,7 +375,7 @@ void launch_paged_attn_decode(bf16_t* out, const bf16_t* q,
   #define LAUNCH(DD, GG, K8) \
     paged_attn_decode_kernel<DD, GG, K8><<<grid, block, 0, stream>>>( \
         out, q, k_cache, v_cache, block_table, seq_lens, hq, hkv, \
-        max_blocks, scale, nsplit, part_o, part_ml)
+        max_blocks, scale, q_rstride, nsplit, part_o, part_ml)
   if (!kv8) {
     if (d == 128 && gc == 8) LAUNCH(128, 8, false);
     else if (d == 128) LAUNCH(128, 1, false);

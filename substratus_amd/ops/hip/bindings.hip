@@ -13,8 +13,8 @@ void launch_rms_norm_residual(bf16_t*, bf16_t*, const bf16_t*, const bf16_t*, lo
 void launch_layer_norm(bf16_t*, const bf16_t*, const bf16_t*, const bf16_t*, long, int, float, hipStream_t);
 void launch_silu_mul(bf16_t*, const bf16_t*, long, int, hipStream_t);
 void launch_gelu(bf16_t*, const bf16_t*, long, hipStream_t);
-void launch_rope_kv_append(bf16_t*, bf16_t*, const bf16_t*, const int*, void*, void*, const int*, int, int, int, int, int, float, bool, hipStream_t);
-void launch_paged_attn_decode(bf16_t*, const bf16_t*, const void*, const void*, const int*, const int*, int, int, int, int, int, int, float, int, float*, float*, bool, hipStream_t);
+void launch_rope_kv_append(bf16_t*, bf16_t*, const bf16_t*, const int*, void*, void*, const int*, int, int, int, int, int, float, bool, long, long, long, hipStream_t);
+void launch_paged_attn_decode(bf16_t*, const bf16_t*, const void*, const void*, const int*, const int*, int, int, int, int, int, int, float, long, int, float*, float*, bool, hipStream_t);
 int paged_attn_nsplit(int, int, int);
 bool paged_attn_decode_supported(int, int);
 void launch_attn_prefill(bf16_t*, const bf16_t*, const bf16_t*, const bf16_t*, const int*, int, int, int, int, int, float, hipStream_t);
@@ -45,6 +45,17 @@ inline bf16_t* bf(torch::Tensor& t) {
 }
 inline const bf16_t* bfc(const torch::Tensor& t) {
   return reinterpret_cast<const bf16_t*>(t.data_ptr<at::BFloat16>());
+}
+
+// [ntok, h, d] tensors may be views of a fused qkv projection: inner two
+// dims contiguous, token-row stride arbitrary (saves the decode path three
+// elementwise copies per layer)
+void check_bf16_rows(const torch::Tensor& t, const char* name) {
+  TORCH_CHECK(t.is_cuda(), name, " must be on GPU");
+  TORCH_CHECK(t.scalar_type() == torch::kBFloat16, name, " must be bf16");
+  TORCH_CHECK(t.dim() == 3 && t.stride(2) == 1 &&
+                  t.stride(1) == t.size(2),
+              name, " must be [n, h, d] with contiguous rows");
 }
 
 void check_bf16(const torch::Tensor& t, const char* name) {
@@ -120,7 +131,7 @@ bool cache_kv8(const torch::Tensor& t, const char* name) {
 void rope_kv_append(torch::Tensor q, torch::Tensor k, torch::Tensor v,
                     torch::Tensor positions, torch::Tensor k_cache,
                     torch::Tensor v_cache, torch::Tensor slots, double theta) {
-  check_bf16(q, "q"); check_bf16(k, "k"); check_bf16(v, "v");
+  check_bf16_rows(q, "q"); check_bf16_rows(k, "k"); check_bf16_rows(v, "v");
   const bool kv8 = cache_kv8(k_cache, "k");
   TORCH_CHECK(cache_kv8(v_cache, "v") == kv8, "cache dtypes must match");
   TORCH_CHECK(k_cache.is_contiguous() && v_cache.is_contiguous());
@@ -134,7 +145,8 @@ void rope_kv_append(torch::Tensor q, torch::Tensor k, torch::Tensor v,
   launch_rope_kv_append(bf(q), bf(k), bfc(v), positions.data_ptr<int>(),
                         k_cache.data_ptr(), v_cache.data_ptr(),
                         slots.data_ptr<int>(), ntok,
-                        hq, hkv, d, block_size, (float)theta, kv8, stream());
+                        hq, hkv, d, block_size, (float)theta, kv8,
+                        q.stride(0), k.stride(0), v.stride(0), stream());
   post_launch_check("rope_kv_append");
 }
 
@@ -142,7 +154,7 @@ void paged_attn_decode(torch::Tensor out, torch::Tensor q,
                        torch::Tensor k_cache, torch::Tensor v_cache,
                        torch::Tensor block_table, torch::Tensor seq_lens,
                        double scale) {
-  check_bf16(out, "out"); check_bf16(q, "q");
+  check_bf16(out, "out"); check_bf16_rows(q, "q");
   const bool kv8 = cache_kv8(k_cache, "k");
   TORCH_CHECK(cache_kv8(v_cache, "v") == kv8, "cache dtypes must match");
   TORCH_CHECK(k_cache.is_contiguous() && v_cache.is_contiguous());
@@ -170,8 +182,8 @@ void paged_attn_decode(torch::Tensor out, torch::Tensor q,
                            v_cache.data_ptr(),
                            block_table.data_ptr<int>(),
                            seq_lens.data_ptr<int>(), bsz, hq, hkv, d,
-                           max_blocks, block_size, (float)scale, nsplit,
-                           po, pml, kv8, stream());
+                           max_blocks, block_size, (float)scale, q.stride(0),
+                           nsplit, po, pml, kv8, stream());
   post_launch_check("paged_attn_decode");
 }
 

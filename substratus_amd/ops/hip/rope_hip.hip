@@ -24,14 +24,17 @@
 template <bool KV8>
 __global__ __launch_bounds__(RW * WAVE)
 void rope_kv_append_kernel(
-    bf16_t* __restrict__ q,              // [ntok, hq, d]
+    bf16_t* __restrict__ q,              // [ntok, hq, d] rows qs apart
     bf16_t* __restrict__ k,              // [ntok, hkv, d] (rotated in place)
     const bf16_t* __restrict__ v,        // [ntok, hkv, d]
     const int* __restrict__ positions,   // [ntok]
     void* __restrict__ k_cache,          // [nb, hkv, bs, d]
     void* __restrict__ v_cache,
     const int* __restrict__ slots,       // [ntok]
-    int hq, int hkv, int d, int block_size, float ln_theta, bool do_rope) {
+    int hq, int hkv, int d, int block_size, float ln_theta, bool do_rope,
+    long qs, long ks, long vs) {         // token-row strides in elements —
+                                         // decode passes qkv-split VIEWS so
+                                         // the fused projection never copies
   const int tok = blockIdx.x;
   const int wave = threadIdx.x / WAVE;
   const int lane = threadIdx.x % WAVE;
@@ -54,7 +57,7 @@ void rope_kv_append_kernel(
     if (role >= hq + hkv) {  // V copy
       const int h = role - hq - hkv;
       const uint4* src = reinterpret_cast<const uint4*>(
-          v + ((long)tok * hkv + h) * d);
+          v + (long)tok * vs + (long)h * d);
       const long off = cache_row + (long)h * block_size * d;
       if (!KV8) {
         uint4* dst = reinterpret_cast<uint4*>(
@@ -79,9 +82,9 @@ void rope_kv_append_kernel(
     const bool is_q = role < hq;
     bf16_t* src;
     if (is_q) {
-      src = q + ((long)tok * hq + role) * d;
+      src = q + (long)tok * qs + (long)role * d;
     } else {
-      src = k + ((long)tok * hkv + (role - hq)) * d;
+      src = k + (long)tok * ks + (long)(role - hq) * d;
     }
     if (!do_rope) {
       if (!is_q) {
@@ -128,7 +131,8 @@ void launch_rope_kv_append(bf16_t* q, bf16_t* k, const bf16_t* v,
                            const int* positions, void* k_cache,
                            void* v_cache, const int* slots, int ntok,
                            int hq, int hkv, int d, int block_size,
-                           float theta, bool kv8, hipStream_t stream) {
+                           float theta, bool kv8, long qs, long ks, long vs,
+                           hipStream_t stream) {
   float ln_theta = theta > 0.f ? logf(theta) : 0.f;
   const int nroles = hq + 2 * hkv;
   int chunks = 1;
@@ -137,10 +141,10 @@ void launch_rope_kv_append(bf16_t* q, bf16_t* k, const bf16_t* v,
   if (kv8) {
    hipLaunchKernelGGL(( rope_kv_append_kernel<true>), dim3(grid), dim3(dim3(RW * WAVE)), 0, stream, 
         q, k, v, positions, k_cache, v_cache, slots, hq, hkv, d, block_size,
-        ln_theta, theta > 0.f);
+        ln_theta, theta > 0.f, qs, ks, vs);
   } else {
    hipLaunchKernelGGL(( rope_kv_append_kernel<false>), dim3(grid), dim3(dim3(RW * WAVE)), 0, stream, 
         q, k, v, positions, k_cache, v_cache, slots, hq, hkv, d, block_size,
-        ln_theta, theta > 0.f);
+        ln_theta, theta > 0.f, qs, ks, vs);
   }
 }

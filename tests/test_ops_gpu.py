@@ -574,3 +574,41 @@ def test_engine_churn_gpu():
     assert len(want) == 40
     for sid, mnew in want.items():
         assert got.get(sid, 0) == mnew, (sid, mnew, got.get(sid))
+
+
+def test_rope_attn_strided_qkv_views():
+    """rope_kv_append + paged_attn_decode accept qkv-split VIEWS (token-row
+    stride > h*d) and match the contiguous-copy path — the zero-copy decode
+    route (ops/hip/rope.hip, attention_decode.hip row-stride params)."""
+    from substratus_amd import ops
+    torch.manual_seed(9)
+    ntok, hq, hkv, d, bs = 4, 8, 4, 128, 16
+    nb = 8
+    qkv = torch.randn(ntok, (hq + 2 * hkv) * d, device=DEV).bfloat16()
+    qkv2 = qkv.clone()
+    kc = torch.zeros(nb, hkv, bs, d, device=DEV).bfloat16()
+    vc = torch.zeros_like(kc)
+    kc2, vc2 = kc.clone(), vc.clone()
+    pos = torch.arange(ntok, dtype=torch.int32, device=DEV)
+    slots = torch.tensor([0, 1, 2, 3], dtype=torch.int32, device=DEV)
+    bt = torch.zeros(ntok, 4, dtype=torch.int32, device=DEV)
+    sl = (pos + 1).int()
+
+    def split(t):
+        q, k, v = t.split([hq * d, hkv * d, hkv * d], dim=-1)
+        return (q.view(ntok, hq, d), k.view(ntok, hkv, d),
+                v.view(ntok, hkv, d))
+
+    # strided-view path
+    q, k, v = split(qkv)
+    assert not q.is_contiguous()
+    qr = ops.rope_kv_append(q, k, v, pos, kc, vc, slots, 10000.0)
+    a = ops.paged_attn_decode(qr, kc, vc, bt, sl, d ** -0.5)
+    # contiguous-copy path
+    q2, k2, v2 = (t.contiguous() for t in split(qkv2))
+    qr2 = ops.rope_kv_append(q2, k2, v2, pos, kc2, vc2, slots, 10000.0)
+    a2 = ops.paged_attn_decode(qr2, kc2, vc2, bt, sl, d ** -0.5)
+    torch.testing.assert_close(qr.contiguous(), qr2, atol=0, rtol=0)
+    torch.testing.assert_close(kc, kc2, atol=0, rtol=0)
+    torch.testing.assert_close(vc, vc2, atol=0, rtol=0)
+    torch.testing.assert_close(a, a2, atol=0, rtol=0)
