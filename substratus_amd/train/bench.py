@@ -40,10 +40,12 @@ def run_train_bench(args) -> None:
         torch.cuda.synchronize()
     barrier()
     t0 = time.time()
-    for _ in range(args.steps):
-        loss = trainer.train_step(next(it))
-    if on_gpu:
-        torch.cuda.synchronize()
+    from bench import profiled
+    with profiled(getattr(args, "profile", None)):
+        for _ in range(args.steps):
+            loss = trainer.train_step(next(it))
+        if on_gpu:
+            torch.cuda.synchronize()
     barrier()
     elapsed = time.time() - t0
     if st.initialized:
