@@ -186,6 +186,11 @@ class Block(nn.Module):
             q = _rope_train(q, cos, sin)
             k = _rope_train(k, cos, sin)
         if self.n_kv_heads != self.n_heads:
+            # Materialized expansion measured FASTER than SDPA's native
+            # enable_gqa on ROCm 7.2 (falcon mb4: 498 vs 602 ms/step — the
+            # aotriton flash backend declines grouped heads and SDPA falls
+            # back to a slower path), so the ~19 ms/step of expansion
+            # copies stays. Re-test when aotriton gains enable_gqa.
             rep = self.n_heads // self.n_kv_heads
             k = k.repeat_interleave(rep, dim=1)
             v = v.repeat_interleave(rep, dim=1)
