@@ -180,14 +180,15 @@ def cross_entropy_fwd_bwd(logits: torch.Tensor, targets: torch.Tensor,
 
 def adamw_step(param: torch.Tensor, grad: torch.Tensor, exp_avg: torch.Tensor,
                exp_avg_sq: torch.Tensor, lr: float, beta1: float, beta2: float,
-               eps: float, weight_decay: float, step: int) -> None:
+               eps: float, weight_decay: float, step: int,
+               grad_scale: float = 1.0) -> None:
     ext = _dispatch(param)
     if ext is None:
         ref.adamw_step(param, grad, exp_avg, exp_avg_sq, lr, beta1, beta2,
-                       eps, weight_decay, step)
+                       eps, weight_decay, step, grad_scale)
         return
     ext.adamw_step(param, grad, exp_avg, exp_avg_sq, lr, beta1, beta2,
-                   eps, weight_decay, step)
+                   eps, weight_decay, step, grad_scale)
 
 
 # ---- weight-streaming decode GEMM (ops/hip/wstream_gemm.hip) -------------

@@ -3,6 +3,10 @@
 // decay + bias-corrected param update in one grid-stride pass. bf16 params
 // and grads, fp32 optimizer state; pure HBM streaming (reads p,g,m,v; writes
 // p,m,v → 20 B in / 16 B out per element), vectorized ×4.
+//
+// gscale folds gradient clipping into the update (g := g * gscale), so the
+// trainer never runs a separate foreach_mul over every grad (~10 ms/step at
+// falcon-7b scale).
 #include "common.h"
 #include <algorithm>
 
@@ -10,7 +14,8 @@ __global__ void adamw_kernel(bf16_t* __restrict__ p,
                              const bf16_t* __restrict__ g,
                              float* __restrict__ m, float* __restrict__ v,
                              long n, float lr, float beta1, float beta2,
-                             float eps, float wd, float bc1, float bc2) {
+                             float eps, float wd, float bc1, float bc2,
+                             float gscale) {
   const long n4 = n / 4;
   uint2* p2 = reinterpret_cast<uint2*>(p);
   const uint2* g2 = reinterpret_cast<const uint2*>(g);
@@ -23,7 +28,8 @@ __global__ void adamw_kernel(bf16_t* __restrict__ p,
     float2 gf0 = unpack_bf16x2(gu.x), gf1 = unpack_bf16x2(gu.y);
     float4 mf = m4[i], vf = v4[i];
     float pv[4] = {pf0.x, pf0.y, pf1.x, pf1.y};
-    float gv[4] = {gf0.x, gf0.y, gf1.x, gf1.y};
+    float gv[4] = {gf0.x * gscale, gf0.y * gscale,
+                   gf1.x * gscale, gf1.y * gscale};
     float mv[4] = {mf.x, mf.y, mf.z, mf.w};
     float vv[4] = {vf.x, vf.y, vf.z, vf.w};
     #pragma unroll
@@ -41,7 +47,7 @@ __global__ void adamw_kernel(bf16_t* __restrict__ p,
   // tail
   for (long i = n4 * 4 + blockIdx.x * (long)blockDim.x + threadIdx.x; i < n;
        i += (long)gridDim.x * blockDim.x) {
-    float gj = bf2f(g[i]);
+    float gj = bf2f(g[i]) * gscale;
     float mj = beta1 * m[i] + (1.f - beta1) * gj;
     float vj = beta2 * v[i] + (1.f - beta2) * gj * gj;
     m[i] = mj; v[i] = vj;
@@ -56,10 +62,11 @@ __global__ void adamw_kernel_f32(float* __restrict__ p,
                                  const float* __restrict__ g,
                                  float* __restrict__ m, float* __restrict__ v,
                                  long n, float lr, float beta1, float beta2,
-                                 float eps, float wd, float bc1, float bc2) {
+                                 float eps, float wd, float bc1, float bc2,
+                                 float gscale) {
   for (long i = blockIdx.x * (long)blockDim.x + threadIdx.x; i < n;
        i += (long)gridDim.x * blockDim.x) {
-    float gj = g[i];
+    float gj = g[i] * gscale;
     float mj = beta1 * m[i] + (1.f - beta1) * gj;
     float vj = beta2 * v[i] + (1.f - beta2) * gj * gj;
     m[i] = mj; v[i] = vj;
@@ -69,20 +76,20 @@ __global__ void adamw_kernel_f32(float* __restrict__ p,
 
 void launch_adamw(bf16_t* p, const bf16_t* g, float* m, float* v, long n,
                   float lr, float beta1, float beta2, float eps, float wd,
-                  int step, hipStream_t stream) {
+                  int step, float gscale, hipStream_t stream) {
   float bc1 = 1.f - powf(beta1, (float)step);
   float bc2 = 1.f - powf(beta2, (float)step);
   int grid = (int)std::min((n / 4 + 255) / 256, (long)8192);
  hipLaunchKernelGGL(( adamw_kernel), dim3(dim3(std::max(grid, 1))), dim3(dim3(256)), 0, stream, 
-      p, g, m, v, n, lr, beta1, beta2, eps, wd, bc1, bc2);
+      p, g, m, v, n, lr, beta1, beta2, eps, wd, bc1, bc2, gscale);
 }
 
 void launch_adamw_f32(float* p, const float* g, float* m, float* v, long n,
                       float lr, float beta1, float beta2, float eps, float wd,
-                      int step, hipStream_t stream) {
+                      int step, float gscale, hipStream_t stream) {
   float bc1 = 1.f - powf(beta1, (float)step);
   float bc2 = 1.f - powf(beta2, (float)step);
   int grid = (int)std::min((n + 255) / 256, (long)8192);
  hipLaunchKernelGGL(( adamw_kernel_f32), dim3(dim3(std::max(grid, 1))), dim3(dim3(256)), 0, stream, 
-      p, g, m, v, n, lr, beta1, beta2, eps, wd, bc1, bc2);
+      p, g, m, v, n, lr, beta1, beta2, eps, wd, bc1, bc2, gscale);
 }

@@ -23,8 +23,8 @@ void launch_attn_prefill_probe(int, bf16_t*, const bf16_t*, const bf16_t*, const
 void launch_mfma_probe(float*, const bf16_t*, const bf16_t*, hipStream_t);
 void launch_softmax_sample(long*, const bf16_t*, const float*, const int*, const float*, const float*, int, int, hipStream_t);
 void launch_cross_entropy(float*, bf16_t*, const bf16_t*, const long*, long, int, long, float, hipStream_t);
-void launch_adamw(bf16_t*, const bf16_t*, float*, float*, long, float, float, float, float, float, int, hipStream_t);
-void launch_adamw_f32(float*, const float*, float*, float*, long, float, float, float, float, float, int, hipStream_t);
+void launch_adamw(bf16_t*, const bf16_t*, float*, float*, long, float, float, float, float, float, int, float, hipStream_t);
+void launch_adamw_f32(float*, const float*, float*, float*, long, float, float, float, float, float, int, float, hipStream_t);
 void launch_quant_gemv(bf16_t*, const bf16_t*, const void*, const unsigned char*, int, int, int, bool, hipStream_t);
 void launch_dequant(bf16_t*, const void*, const unsigned char*, long, int, bool, hipStream_t);
 int decode_gemm_splitk(int, int);
@@ -261,7 +261,7 @@ void cross_entropy_fwd_bwd(torch::Tensor loss, torch::Tensor dlogits,
 
 void adamw_step(torch::Tensor p, torch::Tensor g, torch::Tensor m,
                 torch::Tensor v, double lr, double beta1, double beta2,
-                double eps, double wd, long step) {
+                double eps, double wd, long step, double gscale) {
   TORCH_CHECK(p.is_cuda() && p.is_contiguous());
   TORCH_CHECK(m.scalar_type() == torch::kFloat && v.scalar_type() == torch::kFloat);
   long n = p.numel();
@@ -270,13 +270,13 @@ void adamw_step(torch::Tensor p, torch::Tensor g, torch::Tensor m,
     TORCH_CHECK(g.scalar_type() == torch::kBFloat16);
     launch_adamw(bf(p), bfc(g), m.data_ptr<float>(), v.data_ptr<float>(), n,
                  (float)lr, (float)beta1, (float)beta2, (float)eps, (float)wd,
-                 (int)step, stream());
+                 (int)step, (float)gscale, stream());
   } else {
     TORCH_CHECK(p.scalar_type() == torch::kFloat && g.scalar_type() == torch::kFloat);
     launch_adamw_f32(p.data_ptr<float>(), g.data_ptr<float>(),
                      m.data_ptr<float>(), v.data_ptr<float>(), n, (float)lr,
                      (float)beta1, (float)beta2, (float)eps, (float)wd,
-                     (int)step, stream());
+                     (int)step, (float)gscale, stream());
   }
   post_launch_check("adamw_step");
 }

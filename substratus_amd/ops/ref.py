@@ -217,10 +217,12 @@ def cross_entropy_fwd_bwd(logits: torch.Tensor, targets: torch.Tensor,
 
 def adamw_step(param: torch.Tensor, grad: torch.Tensor, exp_avg: torch.Tensor,
                exp_avg_sq: torch.Tensor, lr: float, beta1: float, beta2: float,
-               eps: float, weight_decay: float, step: int) -> None:
+               eps: float, weight_decay: float, step: int,
+               grad_scale: float = 1.0) -> None:
     """One fused AdamW update, in place. param may be bf16 (master-less);
-    exp_avg/exp_avg_sq are fp32. Decoupled weight decay."""
-    g = grad.float()
+    exp_avg/exp_avg_sq are fp32. Decoupled weight decay. grad_scale folds
+    gradient clipping into the read of g."""
+    g = grad.float() * grad_scale
     exp_avg.mul_(beta1).add_(g, alpha=1 - beta1)
     exp_avg_sq.mul_(beta2).addcmul_(g, g, value=1 - beta2)
     bc1 = 1 - beta1 ** step

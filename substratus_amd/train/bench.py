@@ -40,7 +40,12 @@ def run_train_bench(args) -> None:
         torch.cuda.synchronize()
     barrier()
     t0 = time.time()
-    from bench import profiled
+    try:
+        from bench import profiled  # repo-root bench.py (the CLI entry)
+    except ImportError:  # imported outside the repo root
+        import contextlib
+        def profiled(path):
+            return contextlib.nullcontext()
     with profiled(getattr(args, "profile", None)):
         for _ in range(args.steps):
             loss = trainer.train_step(next(it))

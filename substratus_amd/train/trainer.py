@@ -64,13 +64,14 @@ class AdamWState:
         self.v = [torch.zeros_like(m) for m in self.m]
         self.step_num = 0
 
-    def step(self, cfg: TrainConfig) -> None:
+    def step(self, cfg: TrainConfig, grad_scale: float = 1.0) -> None:
         self.step_num += 1
         for p, m, v in zip(self.params, self.m, self.v):
             if p.grad is None:
                 continue
             ops.adamw_step(p.data, p.grad, m, v, cfg.lr, cfg.beta1, cfg.beta2,
-                           cfg.eps, cfg.weight_decay, self.step_num)
+                           cfg.eps, cfg.weight_decay, self.step_num,
+                           grad_scale)
 
     def zero_grad(self) -> None:
         for p in self.params:
@@ -124,9 +125,16 @@ class Trainer:
         loss, dlogits = ops.cross_entropy_fwd_bwd(flat.detach(), targets)
         flat.backward(gradient=dlogits)
         self.buckets.finalize()
+        gscale = 1.0
         if self.cfg.grad_clip > 0:
-            torch.nn.utils.clip_grad_norm_(self.opt.params, self.cfg.grad_clip)
-        self.opt.step(self.cfg)
+            # norm only; the clip SCALE is folded into the fused AdamW
+            # kernel's gradient read (saves a foreach_mul pass over every
+            # grad — ~10 ms/step at falcon-7b scale, profiles/train)
+            grads = [p.grad for p in self.opt.params if p.grad is not None]
+            norms = torch._foreach_norm(grads, 2.0)
+            total = torch.linalg.vector_norm(torch.stack(norms))
+            gscale = min(1.0, self.cfg.grad_clip / (float(total) + 1e-6))
+        self.opt.step(self.cfg, grad_scale=gscale)
         self.opt.zero_grad()
         self.global_step += 1
         self.tokens_seen += inputs.numel() * max(self.st.dp_size, 1)
