@@ -576,13 +576,15 @@ def test_engine_churn_gpu():
         assert got.get(sid, 0) == mnew, (sid, mnew, got.get(sid))
 
 
-def test_rope_attn_strided_qkv_views():
+@pytest.mark.parametrize("hq,hkv", [(8, 4), (64, 8)])
+def test_rope_attn_strided_qkv_views(hq, hkv):
     """rope_kv_append + paged_attn_decode accept qkv-split VIEWS (token-row
     stride > h*d) and match the contiguous-copy path — the zero-copy decode
-    route (ops/hip/rope.hip, attention_decode.hip row-stride params)."""
+    route (ops/hip/rope.hip, attention_decode.hip row-stride params).
+    (64, 8) exercises the GC=8 GQA kernel variant (llama2-70b shape)."""
     from substratus_amd import ops
     torch.manual_seed(9)
-    ntok, hq, hkv, d, bs = 4, 8, 4, 128, 16
+    ntok, d, bs = 4, 128, 16
     nb = 8
     qkv = torch.randn(ntok, (hq + 2 * hkv) * d, device=DEV).bfloat16()
     qkv2 = qkv.clone()
