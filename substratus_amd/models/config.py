@@ -30,6 +30,9 @@ class ModelConfig:
     tie_word_embeddings: bool = False
     # falcon-style: attention and MLP in parallel off one layernorm
     parallel_attn: bool = False
+    # falcon-40b "new decoder architecture": parallel attn+MLP but with TWO
+    # norms of the residual (ln_attn feeds attention, ln_mlp feeds the MLP)
+    parallel_norms: bool = False
     # opt-style: learned positional embeddings + LayerNorm + gelu
     learned_pos_emb: bool = False
     activation: str = "silu"  # "silu" (gated) | "gelu"
@@ -102,6 +105,21 @@ register(ModelConfig(
     intermediate_size=1024, num_layers=2, num_heads=4, num_kv_heads=1,
     vocab_size=512, max_position_embeddings=2048, parallel_attn=True,
     activation="gelu", tie_word_embeddings=True))
+
+# Falcon-40b (reference: examples/falcon-40b/*.yaml). GQA (128 q / 8 kv
+# heads) with the "new decoder architecture": two parallel layernorms per
+# block (ln_attn -> attention, ln_mlp -> MLP).
+register(ModelConfig(
+    name="falcon-40b", family="falcon", hidden_size=8192,
+    intermediate_size=4 * 8192, num_layers=60, num_heads=128, num_kv_heads=8,
+    vocab_size=65024, max_position_embeddings=2048, layer_norm_eps=1e-5,
+    parallel_attn=True, parallel_norms=True, activation="gelu",
+    tie_word_embeddings=True))
+register(ModelConfig(
+    name="falcon40-tiny", family="falcon", hidden_size=256,
+    intermediate_size=1024, num_layers=2, num_heads=4, num_kv_heads=2,
+    vocab_size=512, max_position_embeddings=2048, parallel_attn=True,
+    parallel_norms=True, activation="gelu", tie_word_embeddings=True))
 
 # facebook/opt-125m (reference: examples/facebook-opt-125m/base-model.yaml —
 # the reference's CPU system-test model, test/system.sh).

@@ -96,7 +96,11 @@ class Block(nn.Module):
         self.wo = RowParallelLinear(cfg.num_heads * d, cfg.hidden_size,
                                     bias=bias, device=device, dtype=dtype)
         self.parallel_attn = cfg.parallel_attn
-        if not cfg.parallel_attn:
+        self.parallel_norms = cfg.parallel_norms
+        # norm2 is the post-attention norm (sequential blocks) OR falcon-40b's
+        # ln_mlp (parallel blocks with parallel_norms: norm1 = ln_attn feeds
+        # attention, norm2 = ln_mlp feeds the MLP, both of the same residual)
+        if not cfg.parallel_attn or cfg.parallel_norms:
             self.norm2 = NormParams(cfg.hidden_size, rms, device, dtype)
         if cfg.activation == "silu":
             self.w_gateup = ColumnParallelLinear(
@@ -142,7 +146,8 @@ class Block(nn.Module):
                                   self.scale)
         attn_out = self.wo(a.reshape(a.shape[0], -1))
         if self.parallel_attn:
-            return attn_out + self._mlp(y), h
+            mlp_in = self.norm2(h, self.eps) if self.parallel_norms else y
+            return attn_out + self._mlp(mlp_in), h
         y2, h = fused_norm_residual(attn_out, h, self.norm2, self.eps)
         return self._mlp(y2), h
 
@@ -160,7 +165,8 @@ class Block(nn.Module):
         a = ops.attn_prefill(q, k, v, seq_start, self.scale)
         attn_out = self.wo(a.reshape(a.shape[0], -1))
         if self.parallel_attn:
-            return h + attn_out + self._mlp(y)
+            mlp_in = self.norm2(h, self.eps) if self.parallel_norms else y
+            return h + attn_out + self._mlp(mlp_in)
         h = h + attn_out
         return h + self._mlp(self.norm2(h, self.eps))
 
@@ -198,7 +204,14 @@ class Block(nn.Module):
         a = a.transpose(1, 2).reshape(b, s, self.n_heads * d)
         attn_out = self.wo(a)
         if self.parallel_attn:
-            return h + attn_out + self._mlp_train(y)
+            if self.parallel_norms:
+                y_mlp = F.layer_norm(h.float(), (cfg.hidden_size,),
+                                     self.norm2.weight.float(),
+                                     self.norm2.bias.float(),
+                                     self.eps).to(h.dtype)
+            else:
+                y_mlp = y
+            return h + attn_out + self._mlp_train(y_mlp)
         h = h + attn_out
         if self.norm2.rms:
             y2 = F.rms_norm(h.float(), (cfg.hidden_size,),

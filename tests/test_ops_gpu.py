@@ -389,7 +389,7 @@ def test_quantized_engine_gpu_generates():
     assert len(toks) == 4
 
 
-@pytest.mark.parametrize("name", ["falcon-tiny", "llama-tiny"])
+@pytest.mark.parametrize("name", ["falcon-tiny", "llama-tiny", "falcon40-tiny"])
 def test_engine_gpu_families(name):
     """falcon (MQA + LayerNorm + parallel-attn) and llama families end-to-end
     on the HIP path."""
