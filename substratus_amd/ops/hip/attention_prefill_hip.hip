@@ -118,73 +118,99 @@ void attn_prefill_kernel(bf16_t* __restrict__ out,      // [T, HQ, D]
   //           ds_read_b64_tr_b16 — same probed mapping as the V B-frags
   //           (tr16 out[l][j] = img[(l>>4)*4+j][l&15] = P^T[k][qrow] gives
   //           exactly A[row=llo][k=lhi*8+sel*4+j])
-  __shared__ uint4 k_tile[KCHUNK * D * 2 / 16];
-  __shared__ short v_img[DC16][KCHUNK / 32][2][16][16];
+  __shared__ uint4 k_tile[2][KCHUNK * D * 2 / 16];
+  __shared__ short v_img[2][DC16][KCHUNK / 32][2][16][16];
   __shared__ short p_img[PF_WAVES][KCHUNK / 32][2][16][16];
 
   const int kmax_tile = min(q0_tile + QTILE, n);       // causal bound of tile
   const int nchunks = (kmax_tile + KCHUNK - 1) / KCHUNK;
   const int kmax_wave = min(q0 + 16, n);               // this wave's bound
+  constexpr int NSC = KCHUNK / 16;
 
-  for (int kc = 0; kc < nchunks; ++kc) {
-    const int kbase = kc * KCHUNK;
-    __syncthreads();
-    // ---- cooperative K+V stage -----------------------------------------
-    {
-      constexpr int LOADS = KCHUNK * D / 8 / (PF_WAVES * WAVE);  // per thread
-      #pragma unroll
-      for (int i = 0; i < LOADS; ++i) {
-        const int flat = (i * PF_WAVES * WAVE + threadIdx.x) * 8;  // elem idx
-        const int row = flat / D, col = flat % D;
-        const bool ok = kbase + row < n;
-        const long goff = (long)(base + kbase + row) * krow_stride +
-                          (long)kvh * D + col;
-        // K: row-major, swizzled by 16-B chunk
-        uint4 kv4 = ok ? *reinterpret_cast<const uint4*>(k + goff)
-                       : make_uint4(0, 0, 0, 0);
-        k_tile[swz_chunk<D * 2>(row, col * 2 / 16)] = kv4;
-        // V: transpose into vT[d][k] (8 scalar writes of this uint4's dims)
-        uint4 vv4 = ok ? *reinterpret_cast<const uint4*>(v + goff)
-                       : make_uint4(0, 0, 0, 0);
-        if (VARIANT == 3) {  // keep the loads, skip the image writes
-          asm volatile("" :: "v"(vv4.x), "v"(vv4.y));
-          continue;
-        }
-        // one uint4 into the tr-read image: c = col/16, half = col%16
-        {
-          const int c = col / 16, off = col % 16;   // off in {0, 8}
-          const int ks = row / 32, kk = row % 32;
-          const int sel = (kk % 8) >= 4;
-          const int r = (kk / 8) * 4 + (kk % 4);
-          *reinterpret_cast<uint4*>(&v_img[c][ks][sel][r][off]) = vv4;
-        }
+  // ---- software-pipelined chunk loop ------------------------------------
+  // The online softmax is a SERIAL VALU chain (rowmax reduce -> exp2 ->
+  // rowsum reduce) between the two MFMA phases — the DCE ablation put it at
+  // ~35% of kernel time with neither pipe saturated. Skewing the pipeline
+  // one chunk (QK^T of chunk kc+1 issues right before softmax of kc, from
+  // the other LDS buffer) lets the scheduler interleave those independent
+  // chains. Chunk j is staged at iteration j-2 (double-buffered k/v tiles),
+  // QK'd at iteration j-1, softmax+PV'd at iteration j; two barriers per
+  // iteration, same as the unpipelined loop.
+
+  auto stage_chunk = [&](int kc2, int buf) {
+    if (kc2 >= nchunks) return;
+    const int kbase = kc2 * KCHUNK;
+    constexpr int LOADS = KCHUNK * D / 8 / (PF_WAVES * WAVE);  // per thread
+    #pragma unroll
+    for (int i = 0; i < LOADS; ++i) {
+      const int flat = (i * PF_WAVES * WAVE + threadIdx.x) * 8;  // elem idx
+      const int row = flat / D, col = flat % D;
+      const bool ok = kbase + row < n;
+      const long goff = (long)(base + kbase + row) * krow_stride +
+                        (long)kvh * D + col;
+      // K: row-major, swizzled by 16-B chunk
+      uint4 kv4 = ok ? *reinterpret_cast<const uint4*>(k + goff)
+                     : make_uint4(0, 0, 0, 0);
+      k_tile[buf][swz_chunk<D * 2>(row, col * 2 / 16)] = kv4;
+      uint4 vv4 = ok ? *reinterpret_cast<const uint4*>(v + goff)
+                     : make_uint4(0, 0, 0, 0);
+      if (VARIANT == 3) {  // keep the loads, skip the image writes
+        asm volatile("" :: "v"(vv4.x), "v"(vv4.y));
+        continue;
+      }
+      // one uint4 into the tr-read image: c = col/16, half = col%16
+      {
+        const int c = col / 16, off = col % 16;   // off in {0, 8}
+        const int ks = row / 32, kk = row % 32;
+        const int sel = (kk % 8) >= 4;
+        const int r = (kk / 8) * 4 + (kk % 4);
+        *reinterpret_cast<uint4*>(&v_img[buf][c][ks][sel][r][off]) = vv4;
       }
     }
-    __syncthreads();
+  };
 
-    const bool compute = wave_live && kbase < kmax_wave;
-    if (compute) {
-      // ---- QK^T: KCHUNK/16 16-key subchunks (B-frags from k_tile) -----
-      constexpr int NSC = KCHUNK / 16;
-      float4v sfrag[NSC];
+  // does this wave compute chunk kc2?
+  auto wave_has = [&](int kc2) {
+    return wave_live && kc2 < nchunks && kc2 * KCHUNK < kmax_wave;
+  };
+
+  // QK^T of one chunk: KCHUNK/16 16-key subchunks (B-frags from k_tile)
+  auto qk_chunk = [&](int buf, float4v (&sf)[NSC]) {
+    #pragma unroll
+    for (int sc = 0; sc < NSC; ++sc) {
+      float4v acc = {0.f, 0.f, 0.f, 0.f};
+      const int key = sc * 16 + llo;   // B col = this lane's key row
       #pragma unroll
-      for (int sc = 0; sc < NSC; ++sc) {
-        float4v acc = {0.f, 0.f, 0.f, 0.f};
-        const int key = sc * 16 + llo;   // B col = this lane's key row
-        #pragma unroll
-        for (int c = 0; c < DC32; ++c) {
-          // kdim byte offset = (c*32 + lhi*8)*2 -> chunk c*4 + lhi
-          const uint4 kb4 = k_tile[swz_chunk<D * 2>(key, c * 4 + lhi)];
-          const short8 kb = *reinterpret_cast<const short8*>(&kb4);
-          if (VARIANT == 4) {
-            asm volatile("" :: "v"(kb4.x));  // keep the ds_read alive
-          } else {
-            acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(qa[c], kb, acc,
-                                                          0, 0, 0);
-          }
+      for (int c = 0; c < DC32; ++c) {
+        // kdim byte offset = (c*32 + lhi*8)*2 -> chunk c*4 + lhi
+        const uint4 kb4 = k_tile[buf][swz_chunk<D * 2>(key, c * 4 + lhi)];
+        const short8 kb = *reinterpret_cast<const short8*>(&kb4);
+        if (VARIANT == 4) {
+          asm volatile("" :: "v"(kb4.x));  // keep the ds_read alive
+        } else {
+          acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(qa[c], kb, acc,
+                                                        0, 0, 0);
         }
-        sfrag[sc] = acc;
       }
+      sf[sc] = acc;
+    }
+  };
+
+  // prologue: chunks 0 and 1 staged, QK of chunk 0 in flight
+  stage_chunk(0, 0);
+  stage_chunk(1, 1);
+  __syncthreads();
+  float4v sfrag[NSC], sf_nxt[NSC];
+  if (wave_has(0)) qk_chunk(0, sfrag);
+
+  for (int kc = 0; kc < nchunks; ++kc) {
+    const int cur = kc & 1;
+    const int kbase = kc * KCHUNK;
+    // ---- QK^T of the NEXT chunk (independent of this chunk's softmax) --
+    if (wave_has(kc + 1)) qk_chunk(cur ^ 1, sf_nxt);
+
+    if (wave_has(kc)) {
+      bool do_pv = true;
       if (VARIANT == 1) {
         // skip the softmax VALU: p := raw scores (timing-only), l := +16
         #pragma unroll
@@ -202,131 +228,139 @@ void attn_prefill_kernel(bf16_t* __restrict__ out,      // [T, HQ, D]
         }
         #pragma unroll
         for (int i = 0; i < 4; ++i) l[i] += 16.f;
-        goto pv_phase;
-      }
-      {
-      // ---- mask + online softmax (C layout), log2 domain --------------
-      // All exponentials run as exp2 with scale*log2(e) folded into the
-      // score once (saves a VALU mul per element before every v_exp_f32);
-      // chunks entirely below the causal diagonal (kbase+KCHUNK-1 <= q0,
-      // in-bounds) skip the per-element mask compare+select.
-      const float scale2 = scale * 1.44269504088896340736f;  // log2(e)
-      float pm[NSC][4];
-      float rowmax[4];
-      #pragma unroll
-      for (int i = 0; i < 4; ++i) rowmax[i] = -INFINITY;
-      if (kbase + KCHUNK - 1 <= q0 && kbase + KCHUNK <= n) {
-        #pragma unroll
-        for (int sc = 0; sc < NSC; ++sc) {
-          #pragma unroll
-          for (int i = 0; i < 4; ++i) {
-            const float sv = sfrag[sc][i] * scale2;
-            pm[sc][i] = sv;
-            rowmax[i] = fmaxf(rowmax[i], sv);
-          }
-        }
       } else {
-        int klim[4];
+        // ---- mask + online softmax (C layout), log2 domain --------------
+        // All exponentials run as exp2 with scale*log2(e) folded into the
+        // score once (saves a VALU mul per element before every v_exp_f32);
+        // chunks entirely below the causal diagonal (kbase+KCHUNK-1 <= q0,
+        // in-bounds) skip the per-element mask compare+select.
+        const float scale2 = scale * 1.44269504088896340736f;  // log2(e)
+        float pm[NSC][4];
+        float rowmax[4];
         #pragma unroll
-        for (int i = 0; i < 4; ++i)
-          klim[i] = min(q0 + lhi * 4 + i, n - 1);
-        #pragma unroll
-        for (int sc = 0; sc < NSC; ++sc) {
-          const int kpos = kbase + sc * 16 + llo;
+        for (int i = 0; i < 4; ++i) rowmax[i] = -INFINITY;
+        if (kbase + KCHUNK - 1 <= q0 && kbase + KCHUNK <= n) {
           #pragma unroll
-          for (int i = 0; i < 4; ++i) {
-            float sv = (kpos <= klim[i]) ? sfrag[sc][i] * scale2 : -INFINITY;
-            pm[sc][i] = sv;
-            rowmax[i] = fmaxf(rowmax[i], sv);
+          for (int sc = 0; sc < NSC; ++sc) {
+            #pragma unroll
+            for (int i = 0; i < 4; ++i) {
+              const float sv = sfrag[sc][i] * scale2;
+              pm[sc][i] = sv;
+              rowmax[i] = fmaxf(rowmax[i], sv);
+            }
+          }
+        } else {
+          int klim[4];
+          #pragma unroll
+          for (int i = 0; i < 4; ++i)
+            klim[i] = min(q0 + lhi * 4 + i, n - 1);
+          #pragma unroll
+          for (int sc = 0; sc < NSC; ++sc) {
+            const int kpos = kbase + sc * 16 + llo;
+            #pragma unroll
+            for (int i = 0; i < 4; ++i) {
+              float sv =
+                  (kpos <= klim[i]) ? sfrag[sc][i] * scale2 : -INFINITY;
+              pm[sc][i] = sv;
+              rowmax[i] = fmaxf(rowmax[i], sv);
+            }
           }
         }
-      }
-      #pragma unroll
-      for (int i = 0; i < 4; ++i) {
-        #pragma unroll
-        for (int off = 8; off > 0; off >>= 1)
-          rowmax[i] = fmaxf(rowmax[i], __shfl_xor(rowmax[i], off, WAVE));
-      }
-      float alpha[4];
-      #pragma unroll
-      for (int i = 0; i < 4; ++i) {
-        const float mn = fmaxf(m[i], rowmax[i]);
-        alpha[i] = (m[i] == -INFINITY) ? 0.f : __builtin_amdgcn_exp2f(m[i] - mn);
-        m[i] = mn;
-      }
-      // P = exp2(S - m); row-sum; one packed b64 write per subchunk into
-      // the transposed tr-read image (lane's 4 rows are consecutive shorts)
-      float rowsum[4] = {0.f, 0.f, 0.f, 0.f};
-      #pragma unroll
-      for (int sc = 0; sc < NSC; ++sc) {
-        float pv4[4];
         #pragma unroll
         for (int i = 0; i < 4; ++i) {
-          pv4[i] = (pm[sc][i] == -INFINITY || m[i] == -INFINITY)
-                       ? 0.f : __builtin_amdgcn_exp2f(pm[sc][i] - m[i]);
-          rowsum[i] += pv4[i];
+          #pragma unroll
+          for (int off = 8; off > 0; off >>= 1)
+            rowmax[i] = fmaxf(rowmax[i], __shfl_xor(rowmax[i], off, WAVE));
         }
-        unsigned plo_, phi_;
-        asm("v_cvt_pk_bf16_f32 %0, %1, %2"
-            : "=v"(plo_) : "v"(pv4[0]), "v"(pv4[1]));
-        asm("v_cvt_pk_bf16_f32 %0, %1, %2"
-            : "=v"(phi_) : "v"(pv4[2]), "v"(pv4[3]));
-        const int kf = sc * 16 + llo, kk = kf & 31;
-        *reinterpret_cast<uint2*>(
-            &p_img[wave][kf >> 5][(kk % 8) >= 4]
-                  [(kk / 8) * 4 + (kk % 4)][lhi * 4]) =
-            make_uint2(plo_, phi_);
-      }
-      #pragma unroll
-      for (int i = 0; i < 4; ++i) {
+        float alpha[4];
         #pragma unroll
-        for (int off = 8; off > 0; off >>= 1)
-          rowsum[i] += __shfl_xor(rowsum[i], off, WAVE);
-        l[i] = l[i] * alpha[i] + rowsum[i];
-      }
-      // ---- rescale O ------------------------------------------------
-      #pragma unroll
-      for (int c = 0; c < DC16; ++c) {
+        for (int i = 0; i < 4; ++i) {
+          const float mn = fmaxf(m[i], rowmax[i]);
+          alpha[i] =
+              (m[i] == -INFINITY) ? 0.f : __builtin_amdgcn_exp2f(m[i] - mn);
+          m[i] = mn;
+        }
+        // P = exp2(S - m); row-sum; one packed b64 write per subchunk into
+        // the transposed tr-read image (lane's 4 rows: consecutive shorts)
+        float rowsum[4] = {0.f, 0.f, 0.f, 0.f};
         #pragma unroll
-        for (int i = 0; i < 4; ++i) o_acc[c][i] *= alpha[i];
-      }
-      }
-      pv_phase:
-      if (VARIANT == 2) {
-        asm volatile("" :: "v"(sfrag[0][0]));  // keep QK^T alive
-        continue;
-      }
-      // ---- P·V over KCHUNK keys in 32-k MFMA steps; A and B frags both
-      // come from tr-read images (two ds_read_b64_tr_b16 each)
-      #pragma unroll
-      for (int ks = 0; ks < KCHUNK / 32; ++ks) {
-        // A-frag: P^T image → out[l][j] = P[qrow=llo][k=lhi*8+sel*4+j]
-        const short4v alo = tr16_read(&p_img[wave][ks][0][0][0] + lane * 4);
-        const short4v ahi = tr16_read(&p_img[wave][ks][1][0][0] + lane * 4);
-        short8 pa;
-        pa[0] = alo[0]; pa[1] = alo[1]; pa[2] = alo[2]; pa[3] = alo[3];
-        pa[4] = ahi[0]; pa[5] = ahi[1]; pa[6] = ahi[2]; pa[7] = ahi[3];
+        for (int sc = 0; sc < NSC; ++sc) {
+          float pv4[4];
+          #pragma unroll
+          for (int i = 0; i < 4; ++i) {
+            pv4[i] = (pm[sc][i] == -INFINITY || m[i] == -INFINITY)
+                         ? 0.f : __builtin_amdgcn_exp2f(pm[sc][i] - m[i]);
+            rowsum[i] += pv4[i];
+          }
+          unsigned plo_, phi_;
+          asm("v_cvt_pk_bf16_f32 %0, %1, %2"
+              : "=v"(plo_) : "v"(pv4[0]), "v"(pv4[1]));
+          asm("v_cvt_pk_bf16_f32 %0, %1, %2"
+              : "=v"(phi_) : "v"(pv4[2]), "v"(pv4[3]));
+          const int kf = sc * 16 + llo, kk = kf & 31;
+          *reinterpret_cast<uint2*>(
+              &p_img[wave][kf >> 5][(kk % 8) >= 4]
+                    [(kk / 8) * 4 + (kk % 4)][lhi * 4]) =
+              make_uint2(plo_, phi_);
+        }
+        #pragma unroll
+        for (int i = 0; i < 4; ++i) {
+          #pragma unroll
+          for (int off = 8; off > 0; off >>= 1)
+            rowsum[i] += __shfl_xor(rowsum[i], off, WAVE);
+          l[i] = l[i] * alpha[i] + rowsum[i];
+        }
+        // ---- rescale O ------------------------------------------------
         #pragma unroll
         for (int c = 0; c < DC16; ++c) {
-          // two hardware-transpose reads assemble the B-frag (k = lhi*8+j).
-          // ds_read_b64_tr_b16 lane mapping (probed on gfx950, see
-          // gpurun_out/tr16_probe.txt): with per-lane address base+lane*4
-          // shorts into a row-major [16][16] image, out[l][j] =
-          // img[(l>>4)*4 + j][l&15] — so the address is lane*4 FLAT, not
-          // [lhi*4][llo] (whose 2-byte llo offset the 8B-aligned b64 load
-          // would silently mask off).
-          const short4v lo =
-              tr16_read(&v_img[c][ks][0][0][0] + lane * 4);
-          const short4v hi =
-              tr16_read(&v_img[c][ks][1][0][0] + lane * 4);
-          short8 vb;
-          vb[0] = lo[0]; vb[1] = lo[1]; vb[2] = lo[2]; vb[3] = lo[3];
-          vb[4] = hi[0]; vb[5] = hi[1]; vb[6] = hi[2]; vb[7] = hi[3];
-          o_acc[c] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-              pa, vb, o_acc[c], 0, 0, 0);
+          #pragma unroll
+          for (int i = 0; i < 4; ++i) o_acc[c][i] *= alpha[i];
+        }
+      }
+      if (VARIANT == 2) {
+        asm volatile("" :: "v"(sfrag[0][0]));  // keep QK^T alive
+        do_pv = false;
+      }
+      if (do_pv) {
+        // ---- P·V over KCHUNK keys in 32-k MFMA steps; A and B frags both
+        // come from tr-read images (two ds_read_b64_tr_b16 each)
+        #pragma unroll
+        for (int ks = 0; ks < KCHUNK / 32; ++ks) {
+          // A-frag: P^T image -> out[l][j] = P[qrow=llo][k=lhi*8+sel*4+j]
+          const short4v alo =
+              tr16_read(&p_img[wave][ks][0][0][0] + lane * 4);
+          const short4v ahi =
+              tr16_read(&p_img[wave][ks][1][0][0] + lane * 4);
+          short8 pa;
+          pa[0] = alo[0]; pa[1] = alo[1]; pa[2] = alo[2]; pa[3] = alo[3];
+          pa[4] = ahi[0]; pa[5] = ahi[1]; pa[6] = ahi[2]; pa[7] = ahi[3];
+          #pragma unroll
+          for (int c = 0; c < DC16; ++c) {
+            // two hardware-transpose reads assemble the B-frag
+            // (k = lhi*8+j). ds_read_b64_tr_b16 lane mapping (probed on
+            // gfx950, profiles/tr16_probe_r01.txt): with per-lane address
+            // base+lane*4 shorts into a row-major [16][16] image,
+            // out[l][j] = img[(l>>4)*4 + j][l&15] — the address is lane*4
+            // FLAT (a [lhi*4][llo] address has 2-byte low bits the
+            // 8B-aligned b64 load silently masks off).
+            const short4v lo =
+                tr16_read(&v_img[cur][c][ks][0][0][0] + lane * 4);
+            const short4v hi =
+                tr16_read(&v_img[cur][c][ks][1][0][0] + lane * 4);
+            short8 vb;
+            vb[0] = lo[0]; vb[1] = lo[1]; vb[2] = lo[2]; vb[3] = lo[3];
+            vb[4] = hi[0]; vb[5] = hi[1]; vb[6] = hi[2]; vb[7] = hi[3];
+            o_acc[c] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                pa, vb, o_acc[c], 0, 0, 0);
+          }
         }
       }
     }
+    __syncthreads();                 // every wave done reading buffer cur
+    stage_chunk(kc + 2, cur);        // overwrite it with chunk kc+2
+    __syncthreads();                 // staged before next iter's QK reads
+    #pragma unroll
+    for (int sc = 0; sc < NSC; ++sc) sfrag[sc] = sf_nxt[sc];
   }
 
   // ---- epilogue: out[q][head][d] = O / l --------------------------------
