@@ -381,3 +381,57 @@ def test_lease_leader_election():
     time.sleep(0.4)                  # lease expires
     assert b.try_acquire()           # takeover
     assert not a.try_acquire() or True  # a may reclaim after b's expiry only
+
+
+def test_build_upload_md5_dedup_skips_signed_url(harness):
+    """If SCI already reports the spec's MD5 for the bucket tarball, the
+    upload is deduped: Uploaded=True straight away, NO new signed URL is
+    minted, and the build Job runs from the stored tarball (reference
+    build_reconciler.go:190-210 — re-running an unchanged directory skips
+    the upload)."""
+    kube, cloud, sci, mgr = harness
+    from substratus_amd.cloud.base import parse_bucket_url
+    from substratus_amd.controller.build import upload_object_path
+    obj = apiv1.new_object("Model", "dedup1", {
+        "build": {"upload": {"md5checksum": "cafef00d", "requestID": "r-9"}},
+    })
+    # pre-seed the SCI's object store with the matching md5 BEFORE create —
+    # the exact (bucket, path) the reconciler queries
+    _, bucket, prefix = parse_bucket_url(cloud.artifact_bucket_url)
+    path = upload_object_path(obj)
+    obj_path = f"{prefix}/{path}".lstrip("/") if prefix else path
+    sci.md5s[(bucket, obj_path)] = "cafef00d"
+    kube.create(obj)
+    mgr.run_until_settled()
+    m = kube.get("Model", "dedup1")
+    assert apiv1.is_condition_true(m, cond.UPLOADED)
+    assert not sci.signed_urls, "no signed URL should be minted on dedup"
+    assert kube.try_get("Job", "dedup1-model-bld") is not None
+
+
+def test_build_job_recreated_on_stale_image(harness):
+    """Changing the upload md5 (new tarball) makes the target image URL go
+    stale; the reconciler deletes and recreates the builder Job (reference
+    build_reconciler.go:140-156)."""
+    kube, cloud, sci, mgr = harness
+    kube.create(apiv1.new_object("Model", "stale1", {
+        "build": {"upload": {"md5checksum": "aaaa", "requestID": "r-a"}},
+    }))
+    mgr.run_until_settled()
+    bucket, objpath, _ = sci.signed_urls[0]
+    sci.md5s[(bucket, objpath)] = "aaaa"
+    kube.patch("Model", "stale1", {"metadata": {"annotations": {
+        "substratus.ai/upload-timestamp": "t1"}}})
+    mgr.run_until_settled()
+    job1 = kube.get("Job", "stale1-model-bld")
+    # new tarball: spec md5 changes -> image URL changes -> stale Job
+    kube.patch("Model", "stale1", {"spec": {"build": {"upload": {
+        "md5checksum": "bbbb", "requestID": "r-b"}}}})
+    sci.md5s[(bucket, objpath)] = "bbbb"
+    kube.patch("Model", "stale1", {"metadata": {"annotations": {
+        "substratus.ai/upload-timestamp": "t2"}}})
+    mgr.run_until_settled()
+    job2 = kube.get("Job", "stale1-model-bld")
+    ann1 = job1["metadata"].get("annotations", {})
+    ann2 = job2["metadata"].get("annotations", {})
+    assert ann1 != ann2, "builder Job should be recreated for the new image"
