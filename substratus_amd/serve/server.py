@@ -207,19 +207,42 @@ def create_app(engine: Optional[Engine] = None,
                 hbm_gauge.set(torch.cuda.memory_allocated())
             return Response(generate_latest(), media_type=CONTENT_TYPE_LATEST)
 
+    def _bad_request(msg: str) -> JSONResponse:
+        # OpenAI-style error body, 400 instead of a 500 from deep asserts
+        return JSONResponse({"error": {"message": msg,
+                                       "type": "invalid_request_error"}},
+                            status_code=400)
+
     @app.post("/v1/completions")
     async def completions(req: Request):
-        body = await req.json()
+        try:
+            body = await req.json()
+        except Exception:
+            return _bad_request("request body is not valid JSON")
+        if not isinstance(body, dict):
+            return _bad_request("request body must be a JSON object")
         prompt = body.get("prompt", "")
-        if isinstance(prompt, list):
-            prompt_tokens = [int(t) for t in prompt]
-        else:
-            prompt_tokens = tok.encode(str(prompt), vocab)
-        max_tokens = int(body.get("max_tokens", 16))
-        temperature = float(body.get("temperature", 1.0))
-        top_p = float(body.get("top_p", 1.0))
-        top_k = int(body.get("top_k", 0))
-        stream = bool(body.get("stream", False))
+        try:
+            if isinstance(prompt, list):
+                prompt_tokens = [int(t) for t in prompt]
+            else:
+                prompt_tokens = tok.encode(str(prompt), vocab)
+            max_tokens = int(body.get("max_tokens", 16))
+            temperature = float(body.get("temperature", 1.0))
+            top_p = float(body.get("top_p", 1.0))
+            top_k = int(body.get("top_k", 0))
+            stream = bool(body.get("stream", False))
+        except (TypeError, ValueError) as e:
+            return _bad_request(f"invalid parameter: {e}")
+        if not prompt_tokens:
+            return _bad_request("prompt must not be empty")
+        if max_tokens < 1:
+            return _bad_request("max_tokens must be >= 1")
+        limit = engine.cfg.max_seq_len
+        if len(prompt_tokens) + max_tokens > limit:
+            return _bad_request(
+                f"prompt ({len(prompt_tokens)} tokens) + max_tokens "
+                f"({max_tokens}) exceeds the server max_length ({limit})")
         t_start = time.time()
         sid, out_q = worker.submit(prompt_tokens, max_tokens, temperature,
                                    top_k, top_p, eos_token=tok.EOS)

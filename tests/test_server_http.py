@@ -111,3 +111,28 @@ def test_dataset_loader(tmp_path):
     files = os.listdir(out)
     assert "src.jsonl" in files and "synthetic.jsonl" in files
     assert "completed.json" in files
+
+
+def test_completion_error_paths(client):
+    """Invalid requests get OpenAI-style 400 errors, not 500s from deep
+    engine asserts: over-limit max_tokens, empty prompt, bad types, and a
+    non-JSON body."""
+    r = client.post("/v1/completions",
+                    json={"prompt": "hi", "max_tokens": 10 ** 6})
+    assert r.status_code == 400
+    assert r.json()["error"]["type"] == "invalid_request_error"
+    assert client.post("/v1/completions",
+                       json={"prompt": [], "max_tokens": 2}).status_code == 400
+    assert client.post("/v1/completions",
+                       json={"prompt": "x", "max_tokens": 0}).status_code == 400
+    assert client.post("/v1/completions",
+                       json={"prompt": "x",
+                             "temperature": "warm"}).status_code == 400
+    r = client.post("/v1/completions", content=b"not json{",
+                    headers={"content-type": "application/json"})
+    assert r.status_code == 400
+    # the server still works after the bad requests
+    ok = client.post("/v1/completions", json={"prompt": "hello",
+                                              "max_tokens": 2})
+    assert ok.status_code == 200
+    assert ok.json()["usage"]["completion_tokens"] == 2
