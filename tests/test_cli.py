@@ -169,3 +169,27 @@ def test_logs_command(daemon):
     r = runner.invoke(cli.app, ["logs", "cli-srv-server"])
     # uvicorn logs or a clean 404 depending on test ordering; accept both
     assert r.exit_code in (0, 1)
+
+
+def test_run_replace_flag(daemon, tmp_path):
+    """-r replaces an existing object instead of erroring/incrementing
+    (reference internal/cli/run.go -r)."""
+    platform, url = daemon
+    ctx = tmp_path / "rctx"
+    ctx.mkdir()
+    (ctx / "Dockerfile").write_text("FROM scratch\n")
+    (ctx / "dataset.yaml").write_text(yaml.dump({
+        "apiVersion": "substratus.ai/v1", "kind": "Dataset",
+        "metadata": {"name": "cli-replace"},
+        "spec": {"params": {"synthetic_rows": 3}},
+    }))
+    r = runner.invoke(cli.app, ["run", str(ctx)])
+    assert r.exit_code == 0, r.output
+    gen1 = platform.kube.get("Dataset", "cli-replace")["metadata"].get(
+        "generation") or platform.kube.get(
+            "Dataset", "cli-replace")["metadata"]["resourceVersion"]
+    r = runner.invoke(cli.app, ["run", str(ctx), "-r"])
+    assert r.exit_code == 0, r.output
+    obj = platform.kube.get("Dataset", "cli-replace")
+    assert obj is not None  # same name, recreated/replaced
+    assert platform.kube.try_get("Dataset", "cli-replace-2") is None

@@ -112,6 +112,9 @@ def test_notebook_flow(tmp_path):
                   content=b"print(1)\n", timeout=5).raise_for_status()
         assert "hello.py" in httpx.get(f"{ep}/files?token=default",
                                        timeout=5).json()["files"]
+        # wrong token -> 403 (Jupyter ?token= behavior)
+        assert httpx.get(f"{ep}/files?token=WRONG",
+                         timeout=5).status_code == 403
         # suspend deletes the pod
         kube.patch("Notebook", "nb1", {"spec": {"suspend": True}})
         mgr.run_until_settled()
