@@ -133,8 +133,18 @@ class Engine:
         while self.waiting and len(self.running) + len(admitted) < self.cfg.max_batch:
             seq = self.waiting[0]
             nblk = seq.needed_blocks(self.cfg.block_size, extra_tokens=1)
-            if seq.length > budget:
+            if admitted and seq.length > budget:
+                # budget gates only the 2nd+ admission: a prompt longer than
+                # prefill_chunk must still go through alone or it would
+                # never be admitted (livelock)
                 break
+            if nblk > self.kv.num_blocks:
+                # can never fit even an empty cache — fail the request
+                # loudly instead of spinning forever
+                self.waiting.pop(0)
+                raise RuntimeError(
+                    f"request {seq.seq_id} needs {nblk} KV blocks but the "
+                    f"cache has only {self.kv.num_blocks}")
             if self.kv.allocator.free_count < nblk + 2 * len(self.running):
                 break  # keep headroom for running seqs to grow
             seq.blocks = self.kv.allocator.alloc(nblk)

@@ -174,3 +174,35 @@ def test_wstream_repack_layout():
                     for j in range(8):
                         assert p[base + j] == w[n16 * 16 + llo][
                             t * 32 + lhi * 8 + j]
+
+
+def test_engine_long_prompt_exceeding_prefill_chunk():
+    """A single prompt longer than prefill_chunk is still admitted (alone) —
+    the budget only gates the second+ admission, otherwise the request
+    would livelock in waiting forever."""
+    import torch
+    from substratus_amd.serve.engine import Engine, EngineConfig
+    eng = Engine(EngineConfig(model="llama-tiny", dtype=torch.float32,
+                              max_batch=2, max_seq_len=96, kv_blocks=64,
+                              prefill_chunk=16))
+    sid = eng.add_request(list(range(40)), max_new_tokens=3)
+    toks = []
+    for _ in range(20):
+        if not eng.has_work():
+            break
+        toks += [o.token for o in eng.step()]
+    assert len(toks) == 3, toks
+
+
+def test_engine_impossible_request_raises():
+    """A request needing more KV blocks than the whole cache fails loudly
+    instead of spinning in the waiting queue."""
+    import pytest as _pytest
+    import torch
+    from substratus_amd.serve.engine import Engine, EngineConfig
+    eng = Engine(EngineConfig(model="llama-tiny", dtype=torch.float32,
+                              max_batch=2, max_seq_len=96, kv_blocks=2))
+    eng.add_request(list(range(90)), max_new_tokens=4)
+    with _pytest.raises(RuntimeError, match="KV blocks"):
+        for _ in range(5):
+            eng.step()
