@@ -89,6 +89,10 @@ def _write_random_init(name: str, out: str, params: dict) -> None:
     tensors = {k: v.detach().contiguous()
                for k, v in model.state_dict().items()}
     save_file(tensors, os.path.join(out, "model.safetensors"))
+    # real tokenizer artifact (serve/tokenizer.py loads it; byte-level BPE
+    # since vocabularies aren't downloadable offline)
+    from ..serve.tokenizer import write_byte_level_tokenizer
+    write_byte_level_tokenizer(os.path.join(out, "tokenizer.json"))
     with open(os.path.join(out, "config.json"), "w") as f:
         json.dump({"model": cfg.name, "dtype": str(dtype).split(".")[-1],
                    "source": "random-init"}, f)

@@ -28,7 +28,7 @@ import torch
 from fastapi import FastAPI, Request
 from fastapi.responses import JSONResponse, StreamingResponse
 
-from . import tokenizer as tok
+from . import tokenizer as tokenizer_mod
 from .engine import Engine, EngineConfig, StepOutput
 
 
@@ -163,13 +163,24 @@ def create_app(engine: Optional[Engine] = None,
     app.state.engine = engine
     vocab = engine.mcfg.vocab_size
     model_id = engine.cfg.model
+    # real tokenizer.json from the artifacts when present (loader exports
+    # one), byte-level fallback otherwise
+    tok_obj = tokenizer_mod.load(os.environ.get("MODEL_DIR",
+                                                "/content/model"), vocab)
 
     try:
-        from prometheus_client import (CONTENT_TYPE_LATEST, Gauge, Counter,
-                                       generate_latest)
-        tokens_counter = Counter("served_tokens_total",
-                                 "tokens generated")
-        hbm_gauge = Gauge("hbm_allocated_bytes", "GPU memory allocated")
+        from prometheus_client import (CONTENT_TYPE_LATEST, REGISTRY, Gauge,
+                                       Counter, generate_latest)
+        try:
+            tokens_counter = Counter("served_tokens_total",
+                                     "tokens generated")
+            hbm_gauge = Gauge("hbm_allocated_bytes", "GPU memory allocated")
+        except ValueError:
+            # app re-created in the same process (tests, embedded platform
+            # restart): reuse the registered collectors
+            tokens_counter = REGISTRY._names_to_collectors[
+                "served_tokens_total"]
+            hbm_gauge = REGISTRY._names_to_collectors["hbm_allocated_bytes"]
         have_prom = True
     except ImportError:  # pragma: no cover
         have_prom = False
@@ -226,7 +237,7 @@ def create_app(engine: Optional[Engine] = None,
             if isinstance(prompt, list):
                 prompt_tokens = [int(t) for t in prompt]
             else:
-                prompt_tokens = tok.encode(str(prompt), vocab)
+                prompt_tokens = tok_obj.encode(str(prompt))
             max_tokens = int(body.get("max_tokens", 16))
             temperature = float(body.get("temperature", 1.0))
             top_p = float(body.get("top_p", 1.0))
@@ -245,7 +256,7 @@ def create_app(engine: Optional[Engine] = None,
                 f"({max_tokens}) exceeds the server max_length ({limit})")
         t_start = time.time()
         sid, out_q = worker.submit(prompt_tokens, max_tokens, temperature,
-                                   top_k, top_p, eos_token=tok.EOS)
+                                   top_k, top_p, eos_token=tok_obj.eos)
 
         def collect() -> list[int]:
             toks = []
@@ -264,7 +275,7 @@ def create_app(engine: Optional[Engine] = None,
                         "id": f"cmpl-{sid}", "object": "text_completion",
                         "model": model_id,
                         "choices": [{"index": 0,
-                                     "text": tok.decode([o.token]),
+                                     "text": tok_obj.decode([o.token]),
                                      "finish_reason": "stop" if o.finished
                                      else None}],
                     }
@@ -283,7 +294,7 @@ def create_app(engine: Optional[Engine] = None,
             "model": model_id,
             "choices": [{
                 "index": 0,
-                "text": tok.decode(toks),
+                "text": tok_obj.decode(toks),
                 "finish_reason": "length" if len(toks) >= max_tokens
                                  else "stop",
             }],

@@ -136,3 +136,28 @@ def test_completion_error_paths(client):
                                               "max_tokens": 2})
     assert ok.status_code == 200
     assert ok.json()["usage"]["completion_tokens"] == 2
+
+
+def test_server_uses_artifacts_tokenizer(tmp_path, monkeypatch):
+    """With a tokenizer.json in MODEL_DIR the server tokenizes through the
+    real `tokenizers` path: completion text round-trips cleanly (the
+    byte-fallback would also work, but this pins the HF-tokenizer wiring,
+    loader export included)."""
+    import torch
+    from starlette.testclient import TestClient
+    from substratus_amd.serve import tokenizer as T
+    from substratus_amd.serve.engine import Engine, EngineConfig
+    from substratus_amd.serve.server import create_app
+    mdir = tmp_path / "model"
+    mdir.mkdir()
+    T.write_byte_level_tokenizer(str(mdir / "tokenizer.json"))
+    monkeypatch.setenv("MODEL_DIR", str(mdir))
+    eng = Engine(EngineConfig(model="llama-tiny", dtype=torch.float32,
+                              max_batch=2, max_seq_len=128, kv_blocks=64))
+    app = create_app(engine=eng)
+    with TestClient(app) as c:
+        r = c.post("/v1/completions", json={"prompt": "Hi!", "max_tokens": 3})
+        assert r.status_code == 200
+        body = r.json()
+        assert body["usage"]["prompt_tokens"] >= 3  # bos + byte-level ids
+        assert body["usage"]["completion_tokens"] == 3
