@@ -53,7 +53,21 @@ class LocalRuntime:
 
     # ------------------------------------------------------------- driving
     def sync(self) -> None:
-        """One pass: start pending Jobs/Deployments/Pods, probe readiness."""
+        """One pass: start pending Jobs/Deployments/Pods, probe readiness,
+        and reap processes whose backing object is gone (a kubelet would
+        kill the container when its Pod is deleted — e.g. Notebook
+        suspend)."""
+        for key in list(self.procs):
+            kind, ns, name = key
+            if self.kube.try_get(kind, name, ns) is None:
+                proc = self.procs.pop(key)
+                self.server_ports.pop(f"{ns}/{name}", None)
+                if proc.poll() is None:
+                    proc.terminate()
+                    try:
+                        proc.wait(timeout=5)
+                    except subprocess.TimeoutExpired:
+                        proc.kill()
         for job in self.kube.list("Job"):
             self._sync_job(job)
         for dep in self.kube.list("Deployment"):

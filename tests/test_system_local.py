@@ -115,10 +115,18 @@ def test_notebook_flow(tmp_path):
         # wrong token -> 403 (Jupyter ?token= behavior)
         assert httpx.get(f"{ep}/files?token=WRONG",
                          timeout=5).status_code == 403
-        # suspend deletes the pod
+        # suspend deletes the pod AND the runtime reaps the shim process
         kube.patch("Notebook", "nb1", {"spec": {"suspend": True}})
         mgr.run_until_settled()
         assert kube.try_get("Pod", "nb1-notebook") is None
+        rt.sync()
+        assert ("Pod", "default", "nb1-notebook") not in rt.procs
+        try:
+            httpx.get(f"{ep}/api", timeout=2)
+            alive = True
+        except Exception:
+            alive = False
+        assert not alive, "notebook shim should be terminated on suspend"
     finally:
         rt.stop()
 
