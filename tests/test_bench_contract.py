@@ -65,3 +65,20 @@ def test_bench_torchrun_2rank_gloo():
     assert r["n_gpus"] == 2
     assert r["config"]["parallelism"] == "dp2"
     assert r["config"]["global_batch"] == 4  # 2 per replica × 2 replicas
+
+
+@pytest.mark.timeout(600)
+def test_bench_torchrun_tp2_gloo():
+    """TP serving bench: 2 ranks serve ONE shared batch (tokens counted
+    once, parallelism tp2) — the llama2-70b TP=8 accounting at CPU scale."""
+    out = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+         "--master-port", "29913", "bench.py", "--gpus", "2", "--tp", "2",
+         "--model", "llama-tiny", "--batch", "2", "--prompt-len", "16",
+         "--steps", "4", "--warmup", "1"],
+        cwd=ROOT, capture_output=True, text=True, timeout=580)
+    assert out.returncode == 0, out.stderr[-2000:]
+    r = _last_json(out.stdout)
+    assert r["config"]["parallelism"] == "tp2"
+    assert r["config"]["global_batch"] == 2  # ONE shared batch, not 2x
