@@ -168,6 +168,13 @@ class Trainer:
         ckpts = []
         for d in os.listdir(self.cfg.artifacts_dir):
             if d.startswith("checkpoint-"):
+                # trainer_state.json is written LAST by save_checkpoint, so
+                # its presence is the completion marker: a crash mid-save
+                # leaves a partial dir that resume must skip (falling back
+                # to the previous complete checkpoint)
+                if not os.path.exists(os.path.join(
+                        self.cfg.artifacts_dir, d, "trainer_state.json")):
+                    continue
                 try:
                     ckpts.append((int(d.split("-")[1]), d))
                 except ValueError:

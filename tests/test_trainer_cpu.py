@@ -168,3 +168,29 @@ def test_gradient_checkpointing_matches(tmp_path):
     t2.train_step(batch.clone())
     for p1, p2 in zip(t1.model.parameters(), t2.model.parameters()):
         torch.testing.assert_close(p1.data, p2.data, atol=1e-6, rtol=1e-5)
+
+
+def test_resume_skips_partial_checkpoint(tmp_path):
+    """A crash mid-save leaves a checkpoint dir without trainer_state.json
+    (the last-written file); resume falls back to the newest COMPLETE
+    checkpoint instead of crashing on the partial one."""
+    import os
+    import torch
+    from substratus_amd.train.data import SyntheticDataset
+    from substratus_amd.train.trainer import TrainConfig, Trainer
+    cfg = TrainConfig(model="llama-tiny", micro_batch=2, seq_len=16,
+                      dtype=torch.float32, artifacts_dir=str(tmp_path),
+                      resume=True, save_steps=0)
+    t = Trainer(cfg)
+    data = SyntheticDataset(t.mcfg.vocab_size, 16)
+    it = data.batches(2, 0, 1)
+    t.train_step(next(it))
+    t.train_step(next(it))
+    t.save_checkpoint()  # complete checkpoint-2
+    # fake a crashed save: newer dir with weights but no trainer_state.json
+    partial = tmp_path / "checkpoint-5"
+    partial.mkdir()
+    (partial / "model.safetensors").write_bytes(b"garbage")
+    t2 = Trainer(cfg)
+    assert t2.maybe_resume() == 2  # resumed from checkpoint-2, not -5
+    assert t2.global_step == 2
