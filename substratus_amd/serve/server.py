@@ -108,7 +108,26 @@ class InferenceWorker:
                 moved = True
             if self.engine.has_work():
                 t0 = time.time()
-                outs = self.engine.step()
+                try:
+                    outs = self.engine.step()
+                except Exception as e:  # noqa: BLE001
+                    # engine-level failure (e.g. a request that can never
+                    # fit the KV cache): fail the waiting requests instead
+                    # of killing the worker thread — serving must survive
+                    import sys as _sys
+                    print(f"[server] engine.step failed: {e}",
+                          file=_sys.stderr, flush=True)
+                    with self.lock:
+                        for sid, q_ in list(self.outputs.items()):
+                            q_.put(StepOutput(sid, 0, True))
+                            self.outputs.pop(sid, None)
+                    self.engine.waiting.clear()
+                    for s in list(self.engine.running):
+                        if s.blocks:
+                            self.engine.kv.allocator.free(s.blocks)
+                            s.blocks = []
+                    self.engine.running.clear()
+                    continue
                 dt = time.time() - t0
                 with self.lock:
                     self.latencies.append(dt)

@@ -161,3 +161,27 @@ def test_server_uses_artifacts_tokenizer(tmp_path, monkeypatch):
         body = r.json()
         assert body["usage"]["prompt_tokens"] >= 3  # bos + byte-level ids
         assert body["usage"]["completion_tokens"] == 3
+
+
+def test_worker_survives_engine_failure(tmp_path):
+    """An engine-level failure (request that can never fit the KV cache at
+    step time) fails the in-flight requests but does NOT kill the worker —
+    subsequent requests are served normally."""
+    import torch
+    from starlette.testclient import TestClient
+    from substratus_amd.serve.engine import Engine, EngineConfig
+    from substratus_amd.serve.server import create_app
+    # max_seq_len allows the request through HTTP validation, but the cache
+    # (2 blocks = 32 tokens) cannot ever hold it -> engine.step raises
+    eng = Engine(EngineConfig(model="llama-tiny", dtype=torch.float32,
+                              max_batch=2, max_seq_len=128, kv_blocks=2))
+    app = create_app(engine=eng)
+    with TestClient(app) as c:
+        r = c.post("/v1/completions",
+                   json={"prompt": list(range(100)), "max_tokens": 2})
+        assert r.status_code == 200  # failed request completes (empty-ish)
+        # the worker is still alive and serves a normal request
+        ok = c.post("/v1/completions", json={"prompt": [1, 2, 3],
+                                             "max_tokens": 2})
+        assert ok.status_code == 200
+        assert ok.json()["usage"]["completion_tokens"] == 2
