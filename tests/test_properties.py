@@ -1,0 +1,86 @@
+"""Property-based tests (hypothesis) for the pure-math/pure-data layers:
+quantization codec error bounds, strategic-merge semantics, tokenizer
+roundtrips. These run on random inputs every time — the fixed-seed unit
+tests elsewhere pin exact values."""
+
+import math
+
+import torch
+from hypothesis import given, settings, strategies as st
+
+from substratus_amd.kube.memapi import deep_merge
+from substratus_amd.models import gguf
+from substratus_amd.serve import tokenizer as tok
+
+
+@settings(max_examples=30, deadline=None)
+@given(st.integers(1, 8), st.integers(1, 8), st.random_module())
+def test_q8_0_roundtrip_bound(rows, blocks, _rng):
+    w = torch.randn(rows, blocks * 32)
+    packed = gguf.quantize_q8_0(w)
+    back = gguf.dequantize_q8_0(packed, w.shape[-1])
+    # int8 symmetric quant: |err| <= scale/2 = amax/254 per block, plus the
+    # fp16 rounding of the stored scale (~2^-11 relative)
+    amax = w.reshape(-1, 32).abs().amax(dim=1, keepdim=True)
+    bound = (amax / 254.0 + amax * 2e-3 + 1e-6).expand(-1, 32).reshape(w.shape)
+    assert ((back - w).abs() <= bound).all()
+
+
+@settings(max_examples=30, deadline=None)
+@given(st.integers(1, 8), st.integers(1, 8), st.random_module())
+def test_q4_0_roundtrip_bound(rows, blocks, _rng):
+    w = torch.randn(rows, blocks * 32)
+    packed = gguf.quantize_q4_0(w)
+    back = gguf.dequantize_q4_0(packed, w.shape[-1])
+    # llama.cpp q4_0 uses the SIGNED max: d = max/-8, values clamp to
+    # [-8, 7], so the +amax side can err by up to amax/8 (plus fp16 scale
+    # rounding)
+    amax = w.reshape(-1, 32).abs().amax(dim=1, keepdim=True)
+    bound = (amax / 8.0 + amax * 3e-3 + 1e-6).expand(-1, 32).reshape(w.shape)
+    assert ((back - w).abs() <= bound).all()
+
+
+json_scalars = st.one_of(st.integers(-10, 10), st.text(max_size=5),
+                         st.booleans())
+json_vals = st.recursive(
+    json_scalars,
+    lambda c: st.dictionaries(st.text(max_size=4), c, max_size=3),
+    max_leaves=10)
+
+
+@settings(max_examples=50, deadline=None)
+@given(st.dictionaries(st.text(max_size=4), json_vals, max_size=4))
+def test_deep_merge_identity_and_idempotent(d):
+    # merging a dict into itself is the dict; merging {} changes nothing
+    assert deep_merge(d, {}) == d
+    assert deep_merge(d, d) == d
+    assert deep_merge({}, d) == d
+
+
+@settings(max_examples=50, deadline=None)
+@given(st.text(max_size=64))
+def test_byte_tokenizer_roundtrip(text):
+    t = tok.ByteTokenizer(32000)
+    out = t.decode(t.encode(text))
+    assert out == text
+
+
+def _hf_tok():
+    global _HF_TOK_CACHE
+    try:
+        return _HF_TOK_CACHE
+    except NameError:
+        import os
+        import tempfile
+        d = tempfile.mkdtemp()
+        p = os.path.join(d, "tokenizer.json")
+        tok.write_byte_level_tokenizer(p)
+        _HF_TOK_CACHE = tok.HFTokenizer(p)
+        return _HF_TOK_CACHE
+
+
+@settings(max_examples=20, deadline=None)
+@given(st.text(st.characters(blacklist_categories=("Cs",)), max_size=48))
+def test_hf_byte_level_tokenizer_roundtrip(text):
+    t = _hf_tok()
+    assert t.decode(t.encode(text)) == text
