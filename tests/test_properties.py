@@ -3,7 +3,6 @@ quantization codec error bounds, strategic-merge semantics, tokenizer
 roundtrips. These run on random inputs every time — the fixed-seed unit
 tests elsewhere pin exact values."""
 
-import math
 
 import torch
 from hypothesis import given, settings, strategies as st
