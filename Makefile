@@ -41,3 +41,6 @@ crds:  ## regenerate config/crd/bases
 .PHONY: up
 up:  ## start the local platform daemon
 	$(PY) -m substratus_amd.kube.apiserver
+
+install:  ## console scripts (sub, nbwatch) into the current environment
+	pip install -e . --no-deps --no-build-isolation
