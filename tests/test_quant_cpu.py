@@ -109,3 +109,22 @@ def test_loader_gguf_format(tmp_path):
                                      "quant": "q8_0"})
     files = os.listdir(out)
     assert any(f.endswith(".gguf") for f in files)
+
+
+def test_gguf_reader_rejects_corrupt_files(tmp_path):
+    """Truncated or non-GGUF files raise cleanly (ValueError/struct.error),
+    never hang or return garbage tensors."""
+    import struct
+    import pytest
+    from substratus_amd.models import gguf
+
+    bad_magic = tmp_path / "bad.gguf"
+    bad_magic.write_bytes(b"NOTG" + b"\x00" * 40)
+    with pytest.raises(ValueError, match="not a GGUF"):
+        gguf.read_gguf(str(bad_magic))
+
+    truncated = tmp_path / "trunc.gguf"
+    truncated.write_bytes(struct.pack("<IIQQ", gguf.GGUF_MAGIC,
+                                      gguf.GGUF_VERSION, 5, 5))
+    with pytest.raises((ValueError, struct.error, EOFError)):
+        gguf.read_gguf(str(truncated))
