@@ -123,6 +123,86 @@ def quantize_model(model: nn.Module, quant: str) -> nn.Module:
     return model
 
 
+def _llamacpp_to_native(tensors: dict, cfg) -> dict:
+    """Map llama.cpp GGUF tensor names (TheBloke-style files — the
+    reference's examples/llama2-13b-chat-gguf served exactly these through
+    its llama.cpp image) onto the native CausalLM state-dict names.
+
+    Fusions are row-granular so they work on PACKED q4_0/q8_0 data:
+      blk.i.attn_q|attn_k|attn_v -> blocks.i.wqkv   (row concat)
+      blk.i.ffn_gate|ffn_up      -> blocks.i.w_gateup (row concat)
+    llama.cpp stores wq/wk with its rope row-permutation applied
+    (convert.py permute: within each head, row j comes from row
+    (j%2)*(d/2)+(j//2)); our rope is HF/neox layout, so the inverse
+    permutation — also pure row reordering — is applied here."""
+    from .gguf import GGUFTensor
+
+    def rows_perm_inverse(t: GGUFTensor, n_head: int) -> GGUFTensor:
+        n = t.shape[0]
+        dh = n // n_head
+        half = dh // 2
+        idx = []
+        for h in range(n_head):
+            for j in range(dh):
+                idx.append(h * dh + (j % half) * 2 + (j // half))
+        data = t.data[torch.tensor(idx)]
+        return GGUFTensor(t.name, t.shape, t.ggml_type, data)
+
+    def cat_rows(parts: list[GGUFTensor], name: str) -> GGUFTensor:
+        gt0 = parts[0]
+        for p in parts:
+            if p.ggml_type != gt0.ggml_type:
+                raise ValueError(
+                    f"mixed ggml types in fused tensor {name}: "
+                    f"{[p.ggml_type for p in parts]} (mixed-precision "
+                    "K-quant files are unsupported; re-quantize to pure "
+                    "q4_0 or q8_0)")
+        shape = [sum(p.shape[0] for p in parts)] + list(gt0.shape[1:])
+        return GGUFTensor(name, shape, gt0.ggml_type,
+                          torch.cat([p.data for p in parts], dim=0))
+
+    out = {}
+    hq, hkv = cfg.num_heads, cfg.num_kv_heads
+    flat = {"token_embd.weight": "embed_tokens.weight",
+            "output_norm.weight": "final_norm.weight",
+            "output.weight": "lm_head.weight",
+            "output_norm.bias": "final_norm.bias"}
+    for name, gt in tensors.items():
+        if name in flat:
+            out[flat[name]] = gt
+            continue
+        if not name.startswith("blk."):
+            continue
+        _, i, rest = name.split(".", 2)
+        pre = f"blocks.{i}."
+        if rest == "attn_norm.weight":
+            out[pre + "norm1.weight"] = gt
+        elif rest == "attn_norm.bias":
+            out[pre + "norm1.bias"] = gt
+        elif rest == "ffn_norm.weight":
+            out[pre + "norm2.weight"] = gt
+        elif rest == "ffn_norm.bias":
+            out[pre + "norm2.bias"] = gt
+        elif rest == "attn_output.weight":
+            out[pre + "wo.weight"] = gt
+        elif rest == "ffn_down.weight":
+            out[pre + "w_down.weight"] = gt
+        # fused parts collected below
+    for i in range(cfg.num_layers):
+        qn, kn, vn = (f"blk.{i}.attn_q.weight", f"blk.{i}.attn_k.weight",
+                      f"blk.{i}.attn_v.weight")
+        if qn in tensors:
+            q = rows_perm_inverse(tensors[qn], hq)
+            k = rows_perm_inverse(tensors[kn], hkv)
+            out[f"blocks.{i}.wqkv.weight"] = cat_rows(
+                [q, k, tensors[vn]], f"blocks.{i}.wqkv.weight")
+        gn, un = f"blk.{i}.ffn_gate.weight", f"blk.{i}.ffn_up.weight"
+        if gn in tensors:
+            out[f"blocks.{i}.w_gateup.weight"] = cat_rows(
+                [tensors[gn], tensors[un]], f"blocks.{i}.w_gateup.weight")
+    return out
+
+
 def load_gguf_into_model(model: nn.Module, path: str, device,
                          dtype=torch.bfloat16) -> str:
     """Load a GGUF checkpoint: quantized 2-D tensors become QuantLinear
@@ -131,6 +211,17 @@ def load_gguf_into_model(model: nn.Module, path: str, device,
     from .gguf import GGML_Q4_0, GGML_Q8_0, read_gguf
     meta, tensors = read_gguf(path)
     quant = meta.get("substratus.quant", "q4_0")
+    if any(n.startswith("blk.") or n == "token_embd.weight"
+           for n in tensors):
+        # llama.cpp-named file (ecosystem GGUFs)
+        tensors = _llamacpp_to_native(tensors, model.cfg)
+        for gt in tensors.values():
+            if gt.ggml_type == GGML_Q4_0:
+                quant = "q4_0"
+                break
+            if gt.ggml_type == GGML_Q8_0:
+                quant = "q8_0"
+                break
     sd = {}
     qmap = {}
     for name, gt in tensors.items():

@@ -128,3 +128,85 @@ def test_gguf_reader_rejects_corrupt_files(tmp_path):
                                       gguf.GGUF_VERSION, 5, 5))
     with pytest.raises((ValueError, struct.error, EOFError)):
         gguf.read_gguf(str(truncated))
+
+
+def test_llamacpp_named_gguf_loads(tmp_path):
+    """An ecosystem GGUF with llama.cpp tensor names (blk.N.attn_q etc. +
+    llama.cpp's q/k rope row-permutation, as TheBloke-style files carry —
+    the reference's llama2-13b-chat-gguf example) loads into the native
+    model and generates the SAME tokens as loading the native-named export
+    of the same weights."""
+    import torch
+    from substratus_amd.models import gguf
+    from substratus_amd.models.config import get_config
+    from substratus_amd.models.model import CausalLM
+    from substratus_amd.models.quant import load_gguf_into_model
+    from substratus_amd.serve.engine import Engine, EngineConfig
+
+    cfg = get_config("llama-tiny")
+    torch.manual_seed(4)
+    ref = CausalLM(cfg, device="meta", dtype=torch.float32)
+    ref = ref.to_empty(device="cpu")
+    ref.init_random(4)
+    sd = {k: v.float() for k, v in ref.state_dict().items()}
+
+    hq, hkv, d = cfg.num_heads, cfg.num_kv_heads, cfg.head_dim
+
+    def perm_fwd(w, n_head):
+        # llama.cpp convert.py: within each head rows reorder via
+        # reshape(n_head, 2, d/2, k).swapaxes(1, 2)
+        n, k = w.shape
+        return (w.reshape(n_head, 2, d // 2, k).swapaxes(1, 2)
+                 .reshape(n, k))
+
+    tensors = {}
+    def q40(name, w, logical_shape=None):
+        tensors[name] = (gguf.quantize_q4_0(w), gguf.GGML_Q4_0,
+                         logical_shape or list(w.shape))
+    def f32(name, w):
+        tensors[name] = (w.contiguous(), gguf.GGML_F32, list(w.shape))
+
+    f32("token_embd.weight", sd["embed_tokens.weight"])
+    f32("output_norm.weight", sd["final_norm.weight"])
+    q40("output.weight", sd["lm_head.weight"])
+    for i in range(cfg.num_layers):
+        p = f"blocks.{i}."
+        f32(f"blk.{i}.attn_norm.weight", sd[p + "norm1.weight"])
+        f32(f"blk.{i}.ffn_norm.weight", sd[p + "norm2.weight"])
+        wqkv = sd[p + "wqkv.weight"]
+        wq, wk, wv = wqkv.split([hq * d, hkv * d, hkv * d], dim=0)
+        q40(f"blk.{i}.attn_q.weight", perm_fwd(wq, hq))
+        q40(f"blk.{i}.attn_k.weight", perm_fwd(wk, hkv))
+        q40(f"blk.{i}.attn_v.weight", wv)
+        q40(f"blk.{i}.attn_output.weight", sd[p + "wo.weight"])
+        gu = sd[p + "w_gateup.weight"]
+        g, u = gu.chunk(2, dim=0)
+        q40(f"blk.{i}.ffn_gate.weight", g)
+        q40(f"blk.{i}.ffn_up.weight", u)
+        q40(f"blk.{i}.ffn_down.weight", sd[p + "w_down.weight"])
+    path = str(tmp_path / "llamacpp.gguf")
+    gguf.write_gguf(path, tensors, {"general.architecture": "llama"})
+
+    def gen(gguf_path):
+        eng = Engine(EngineConfig(model="llama-tiny", dtype=torch.float32,
+                                  max_batch=1, max_seq_len=64, kv_blocks=16,
+                                  seed=5))
+        load_gguf_into_model(eng.model, gguf_path, eng.device, torch.float32)
+        eng.add_request([5, 9, 2, 7], max_new_tokens=6)
+        toks = []
+        while eng.has_work():
+            toks += [o.token for o in eng.step()]
+        return toks
+
+    # native-named export of the SAME weights for comparison
+    native = {}
+    for k, v in sd.items():
+        if v.dim() == 2 and "embed" not in k:
+            native[k] = (gguf.quantize_q4_0(v), gguf.GGML_Q4_0,
+                         list(v.shape))
+        else:
+            native[k] = (v.contiguous(), gguf.GGML_F32, list(v.shape))
+    npath = str(tmp_path / "native.gguf")
+    gguf.write_gguf(npath, native, {"substratus.quant": "q4_0"})
+
+    assert gen(path) == gen(npath)
