@@ -185,3 +185,57 @@ def test_worker_survives_engine_failure(tmp_path):
                                              "max_tokens": 2})
         assert ok.status_code == 200
         assert ok.json()["usage"]["completion_tokens"] == 2
+
+
+def test_hf_named_checkpoint_serves(tmp_path, monkeypatch):
+    """An HF-layout llama safetensors checkpoint (model.layers.N.self_attn.
+    q_proj...) loads via the name mapping and produces the SAME generation
+    as the native-named export of the same weights."""
+    import torch
+    from safetensors.torch import save_file
+    from starlette.testclient import TestClient
+    from substratus_amd.models.config import get_config
+    from substratus_amd.models.model import CausalLM
+    from substratus_amd.serve.engine import Engine, EngineConfig
+    from substratus_amd.serve.server import _maybe_load_weights
+
+    cfg = get_config("llama-tiny")
+    ref = CausalLM(cfg, device="meta", dtype=torch.float32)
+    ref = ref.to_empty(device="cpu")
+    ref.init_random(6)
+    sd = {k: v.clone().contiguous() for k, v in ref.state_dict().items()}
+
+    hq, hkv, d = cfg.num_heads, cfg.num_kv_heads, cfg.head_dim
+    hf = {"model.embed_tokens.weight": sd["embed_tokens.weight"],
+          "model.norm.weight": sd["final_norm.weight"],
+          "lm_head.weight": sd["lm_head.weight"]}
+    for i in range(cfg.num_layers):
+        p, h = f"blocks.{i}.", f"model.layers.{i}."
+        hf[h + "input_layernorm.weight"] = sd[p + "norm1.weight"]
+        hf[h + "post_attention_layernorm.weight"] = sd[p + "norm2.weight"]
+        wq, wk, wv = sd[p + "wqkv.weight"].split(
+            [hq * d, hkv * d, hkv * d], dim=0)
+        hf[h + "self_attn.q_proj.weight"] = wq.contiguous()
+        hf[h + "self_attn.k_proj.weight"] = wk.contiguous()
+        hf[h + "self_attn.v_proj.weight"] = wv.contiguous()
+        hf[h + "self_attn.o_proj.weight"] = sd[p + "wo.weight"]
+        g, u = sd[p + "w_gateup.weight"].chunk(2, dim=0)
+        hf[h + "mlp.gate_proj.weight"] = g.contiguous()
+        hf[h + "mlp.up_proj.weight"] = u.contiguous()
+        hf[h + "mlp.down_proj.weight"] = sd[p + "w_down.weight"]
+
+    def gen(state):
+        mdir = tmp_path / f"m{len(state)}"
+        mdir.mkdir()
+        save_file(state, str(mdir / "model.safetensors"))
+        eng = Engine(EngineConfig(model="llama-tiny", dtype=torch.float32,
+                                  max_batch=1, max_seq_len=64,
+                                  kv_blocks=16, seed=5))
+        _maybe_load_weights(eng, model_dir=str(mdir))
+        eng.add_request([5, 9, 2, 7], max_new_tokens=6)
+        toks = []
+        while eng.has_work():
+            toks += [o.token for o in eng.step()]
+        return toks
+
+    assert gen(hf) == gen(sd)
