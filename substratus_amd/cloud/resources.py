@@ -105,6 +105,15 @@ def apply_resources(pod_spec: dict, resources: dict, cloud_name: str) -> None:
                 "value": "true", "effect": "NoSchedule"})
     for c in pod_spec.get("containers", []):
         c["resources"] = {"requests": requests, "limits": limits}
+        if gpu.get("count"):
+            # container contract "Multi-GPU": the image entrypoints run
+            # `torchrun --nproc-per-node ${NUM_GPUS:-1}` (one rank per GPU
+            # over RCCL) — without this env a TP=8 Server would start a
+            # single rank (images/Dockerfile.model-server, -trainer)
+            env = c.setdefault("env", [])
+            if not any(e.get("name") == "NUM_GPUS" for e in env):
+                env.append({"name": "NUM_GPUS",
+                            "value": str(int(gpu["count"]))})
     if node_selector:
         pod_spec.setdefault("nodeSelector", {}).update(node_selector)
 

@@ -435,3 +435,27 @@ def test_build_job_recreated_on_stale_image(harness):
     ann1 = job1["metadata"].get("annotations", {})
     ann2 = job2["metadata"].get("annotations", {})
     assert ann1 != ann2, "builder Job should be recreated for the new image"
+
+
+def test_gpu_pods_get_num_gpus_env(harness):
+    """Multi-GPU Pods carry NUM_GPUS so the image entrypoints torchrun one
+    rank per amd.com/gpu (container contract; a TP=8 Server without it
+    would start a single rank)."""
+    kube, cloud, sci, mgr = harness
+    kube.create(apiv1.new_object("Model", "gm", {"image": "x/model-loader",
+                                                 "params": {"name": "llama2-70b"}}))
+    mgr.run_until_settled()
+    fake_job_complete(kube, "gm-modeller")
+    mgr.run_until_settled()
+    kube.create(apiv1.new_object("Server", "gs", {
+        "image": "x/model-server",
+        "model": {"name": "gm"},
+        "resources": {"memory": 160,
+                      "gpu": {"type": "amd-mi355x", "count": 8}},
+    }))
+    mgr.run_until_settled()
+    dep = kube.get("Deployment", "gs-server")
+    c = dep["spec"]["template"]["spec"]["containers"][0]
+    envs = {e["name"]: e.get("value") for e in c.get("env", [])}
+    assert envs.get("NUM_GPUS") == "8"
+    assert c["resources"]["limits"]["amd.com/gpu"] == "8"
