@@ -1,4 +1,4 @@
-"""HF-format checkpoint interop for the llama family.
+"""HF-format checkpoint interop for the llama and falcon families.
 
 The reference's loader image was `substratusai/model-loader-huggingface` —
 users bring checkpoints in HF layout. The native loader exports OUR
@@ -8,8 +8,10 @@ native CausalLM names so such checkpoints serve directly. HF's q/k layout
 is the neox/half-split one our rope kernel uses, so the mapping is pure
 renaming + row-concat fusion (qkv, gate+up) — no permutations.
 
-Falcon/OPT HF layouts (fused interleaved qkv, decoder.* prefixes) are not
-mapped yet; the native loader covers those families.
+Falcon HF layouts are mapped too (hf_falcon_to_native): 7b-style MQA
+fused qkv directly, 40b-style per-kv-group interleaved qkv via a row
+degroup. OPT (decoder.* prefixes) is not mapped; the native loader covers
+it.
 """
 
 from __future__ import annotations
@@ -48,4 +50,62 @@ def hf_llama_to_native(sd: dict, cfg: ModelConfig) -> dict:
             [sd[h + "mlp.gate_proj.weight"],
              sd[h + "mlp.up_proj.weight"]], dim=0)
         out[p + "w_down.weight"] = sd[h + "mlp.down_proj.weight"]
+    return out
+
+
+def is_hf_falcon_state_dict(sd: dict) -> bool:
+    return any(k.startswith("transformer.h.") and
+               "self_attention.query_key_value" in k for k in sd)
+
+
+def hf_falcon_to_native(sd: dict, cfg: ModelConfig) -> dict:
+    """tiiuae/falcon-* HF layout -> native names.
+
+    falcon-7b (multi_query): query_key_value rows are [q0..q(hq-1), k, v] —
+    exactly the native wqkv order, pure rename. falcon-40b
+    (new_decoder_architecture, GQA): rows are PER-KV-GROUP interleaved
+    [q_g0..q_g(gq-1), k_g, v_g | g=0..hkv-1]; the native layout wants all q
+    (same g-major head order), then all k, then all v — a row reorder.
+    ln_attn/ln_mlp map to norm1/norm2 (two parallel norms)."""
+    d = cfg.head_dim
+    hq, hkv = cfg.num_heads, cfg.num_kv_heads
+    gq = hq // hkv
+
+    def degroup_qkv(w: torch.Tensor) -> torch.Tensor:
+        if hkv == 1:
+            return w  # multi_query layout == native layout
+        per = (gq + 2) * d
+        qs, ks, vs = [], [], []
+        for g in range(hkv):
+            blockk = w[g * per:(g + 1) * per]
+            qs.append(blockk[:gq * d])
+            ks.append(blockk[gq * d:(gq + 1) * d])
+            vs.append(blockk[(gq + 1) * d:])
+        return torch.cat(qs + ks + vs, dim=0)
+
+    out = {}
+    flat = {"transformer.word_embeddings.weight": "embed_tokens.weight",
+            "transformer.ln_f.weight": "final_norm.weight",
+            "transformer.ln_f.bias": "final_norm.bias",
+            "lm_head.weight": "lm_head.weight"}
+    for src, dst in flat.items():
+        if src in sd:
+            out[dst] = sd[src]
+    for i in range(cfg.num_layers):
+        h = f"transformer.h.{i}."
+        p = f"blocks.{i}."
+        qkv = h + "self_attention.query_key_value.weight"
+        if qkv not in sd:
+            continue
+        out[p + "wqkv.weight"] = degroup_qkv(sd[qkv])
+        out[p + "wo.weight"] = sd[h + "self_attention.dense.weight"]
+        out[p + "w_up.weight"] = sd[h + "mlp.dense_h_to_4h.weight"]
+        out[p + "w_down.weight"] = sd[h + "mlp.dense_4h_to_h.weight"]
+        for hf_n, our_n in (("input_layernorm", "norm1"),
+                            ("ln_attn", "norm1"), ("ln_mlp", "norm2"),
+                            ("post_attention_layernorm", "norm2")):
+            for suf in ("weight", "bias"):
+                k = f"{h}{hf_n}.{suf}"
+                if k in sd:
+                    out[f"{p}{our_n}.{suf}"] = sd[k]
     return out

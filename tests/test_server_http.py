@@ -239,3 +239,70 @@ def test_hf_named_checkpoint_serves(tmp_path, monkeypatch):
         return toks
 
     assert gen(hf) == gen(sd)
+
+
+@pytest.mark.parametrize("model", ["falcon-tiny", "falcon40-tiny"])
+def test_hf_falcon_checkpoint_serves(tmp_path, model):
+    """HF falcon layouts load via hf_compat: falcon-7b-style MQA fused qkv
+    (direct) and falcon-40b-style per-kv-group interleaved qkv (degrouped),
+    generating identically to the native state dict."""
+    import torch
+    from safetensors.torch import save_file
+    from substratus_amd.models.config import get_config
+    from substratus_amd.models.model import CausalLM
+    from substratus_amd.serve.engine import Engine, EngineConfig
+    from substratus_amd.serve.server import _maybe_load_weights
+
+    cfg = get_config(model)
+    ref = CausalLM(cfg, device="meta", dtype=torch.float32)
+    ref = ref.to_empty(device="cpu")
+    ref.init_random(8)
+    sd = {k: v.clone().contiguous() for k, v in ref.state_dict().items()}
+
+    d, hq, hkv = cfg.head_dim, cfg.num_heads, cfg.num_kv_heads
+    gq = hq // hkv
+
+    def regroup(w):  # native [allq, allk, allv] -> HF per-group interleave
+        if hkv == 1:
+            return w
+        q, k, v = w.split([hq * d, hkv * d, hkv * d], dim=0)
+        parts = []
+        for g in range(hkv):
+            parts += [q[g * gq * d:(g + 1) * gq * d],
+                      k[g * d:(g + 1) * d], v[g * d:(g + 1) * d]]
+        return torch.cat(parts, dim=0).contiguous()
+
+    hf = {"transformer.word_embeddings.weight": sd["embed_tokens.weight"],
+          "transformer.ln_f.weight": sd["final_norm.weight"],
+          "transformer.ln_f.bias": sd["final_norm.bias"]}
+    for i in range(cfg.num_layers):
+        p, h = f"blocks.{i}.", f"transformer.h.{i}."
+        hf[h + "self_attention.query_key_value.weight"] = regroup(
+            sd[p + "wqkv.weight"])
+        hf[h + "self_attention.dense.weight"] = sd[p + "wo.weight"]
+        hf[h + "mlp.dense_h_to_4h.weight"] = sd[p + "w_up.weight"]
+        hf[h + "mlp.dense_4h_to_h.weight"] = sd[p + "w_down.weight"]
+        if cfg.parallel_norms:
+            hf[h + "ln_attn.weight"] = sd[p + "norm1.weight"]
+            hf[h + "ln_attn.bias"] = sd[p + "norm1.bias"]
+            hf[h + "ln_mlp.weight"] = sd[p + "norm2.weight"]
+            hf[h + "ln_mlp.bias"] = sd[p + "norm2.bias"]
+        else:
+            hf[h + "input_layernorm.weight"] = sd[p + "norm1.weight"]
+            hf[h + "input_layernorm.bias"] = sd[p + "norm1.bias"]
+
+    def gen(state, tag):
+        mdir = tmp_path / tag
+        mdir.mkdir()
+        save_file(state, str(mdir / "model.safetensors"))
+        eng = Engine(EngineConfig(model=model, dtype=torch.float32,
+                                  max_batch=1, max_seq_len=64,
+                                  kv_blocks=16, seed=5))
+        _maybe_load_weights(eng, model_dir=str(mdir))
+        eng.add_request([5, 9, 2, 7], max_new_tokens=6)
+        toks = []
+        while eng.has_work():
+            toks += [o.token for o in eng.step()]
+        return toks
+
+    assert gen(hf, "hf") == gen(sd, "native")
