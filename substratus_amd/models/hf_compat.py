@@ -1,4 +1,4 @@
-"""HF-format checkpoint interop for the llama and falcon families.
+"""HF-format checkpoint interop for the llama, falcon and opt families.
 
 The reference's loader image was `substratusai/model-loader-huggingface` —
 users bring checkpoints in HF layout. The native loader exports OUR
@@ -8,10 +8,11 @@ native CausalLM names so such checkpoints serve directly. HF's q/k layout
 is the neox/half-split one our rope kernel uses, so the mapping is pure
 renaming + row-concat fusion (qkv, gate+up) — no permutations.
 
-Falcon HF layouts are mapped too (hf_falcon_to_native): 7b-style MQA
-fused qkv directly, 40b-style per-kv-group interleaved qkv via a row
-degroup. OPT (decoder.* prefixes) is not mapped; the native loader covers
-it.
+Falcon HF layouts map too (hf_falcon_to_native): 7b-style MQA fused qkv
+directly, 40b-style per-kv-group interleaved qkv via a row degroup. OPT
+(hf_opt_to_native) maps the decoder.* projections with their biases and
+the +2-offset embed_positions. All served model families accept their HF
+checkpoints.
 """
 
 from __future__ import annotations
@@ -108,4 +109,48 @@ def hf_falcon_to_native(sd: dict, cfg: ModelConfig) -> dict:
                 k = f"{h}{hf_n}.{suf}"
                 if k in sd:
                     out[f"{p}{our_n}.{suf}"] = sd[k]
+    return out
+
+
+def is_hf_opt_state_dict(sd: dict) -> bool:
+    return any(k.startswith("model.decoder.layers.") and
+               "self_attn.q_proj" in k for k in sd)
+
+
+def hf_opt_to_native(sd: dict, cfg: ModelConfig) -> dict:
+    """facebook/opt-* HF layout -> native names: renames + qkv fusion
+    (weights AND biases — OPT projections carry biases). OPT's
+    embed_positions has a +2 offset quirk (first two rows reserved); the
+    native embed_positions is sized max_position_embeddings, so the HF
+    tensor's tail rows are taken."""
+    out = {}
+    if "model.decoder.embed_tokens.weight" in sd:
+        out["embed_tokens.weight"] = sd["model.decoder.embed_tokens.weight"]
+    if "model.decoder.embed_positions.weight" in sd:
+        pos = sd["model.decoder.embed_positions.weight"]
+        want = cfg.max_position_embeddings
+        out["embed_positions.weight"] = pos[-want:] if pos.shape[0] > want \
+            else pos
+    for hf_n, our_n in (("model.decoder.final_layer_norm", "final_norm"),):
+        for suf in ("weight", "bias"):
+            k = f"{hf_n}.{suf}"
+            if k in sd:
+                out[f"{our_n}.{suf}"] = sd[k]
+    if "lm_head.weight" in sd:
+        out["lm_head.weight"] = sd["lm_head.weight"]
+    for i in range(cfg.num_layers):
+        h = f"model.decoder.layers.{i}."
+        p = f"blocks.{i}."
+        if h + "self_attn.q_proj.weight" not in sd:
+            continue
+        for suf in ("weight", "bias"):
+            out[p + f"wqkv.{suf}"] = torch.cat(
+                [sd[h + f"self_attn.q_proj.{suf}"],
+                 sd[h + f"self_attn.k_proj.{suf}"],
+                 sd[h + f"self_attn.v_proj.{suf}"]], dim=0)
+            out[p + f"wo.{suf}"] = sd[h + f"self_attn.out_proj.{suf}"]
+            out[p + f"w_up.{suf}"] = sd[h + f"fc1.{suf}"]
+            out[p + f"w_down.{suf}"] = sd[h + f"fc2.{suf}"]
+            out[p + f"norm1.{suf}"] = sd[h + f"self_attn_layer_norm.{suf}"]
+            out[p + f"norm2.{suf}"] = sd[h + f"final_layer_norm.{suf}"]
     return out

@@ -353,8 +353,10 @@ def _maybe_load_weights(engine: Engine,
         from safetensors.torch import load_file
         from ..models.hf_compat import (hf_falcon_to_native,
                                         hf_llama_to_native,
+                                        hf_opt_to_native,
                                         is_hf_falcon_state_dict,
-                                        is_hf_llama_state_dict)
+                                        is_hf_llama_state_dict,
+                                        is_hf_opt_state_dict)
         from ..parallel.comm import state as pstate
         sd = load_file(path)
         if is_hf_llama_state_dict(sd):
@@ -362,6 +364,8 @@ def _maybe_load_weights(engine: Engine,
             sd = hf_llama_to_native(sd, engine.mcfg)
         elif is_hf_falcon_state_dict(sd):
             sd = hf_falcon_to_native(sd, engine.mcfg)
+        elif is_hf_opt_state_dict(sd):
+            sd = hf_opt_to_native(sd, engine.mcfg)
         st = pstate()
         if st.tp_size > 1:
             from ..models.sharding import shard_state_dict

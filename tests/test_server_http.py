@@ -306,3 +306,59 @@ def test_hf_falcon_checkpoint_serves(tmp_path, model):
         return toks
 
     assert gen(hf, "hf") == gen(sd, "native")
+
+
+def test_hf_opt_checkpoint_serves(tmp_path):
+    """facebook/opt-* HF layout (q/k/v/out projections with biases,
+    decoder.* prefixes, +2-offset embed_positions) loads via hf_compat and
+    generates identically to the native state dict."""
+    import torch
+    from safetensors.torch import save_file
+    from substratus_amd.models.config import get_config
+    from substratus_amd.models.model import CausalLM
+    from substratus_amd.serve.engine import Engine, EngineConfig
+    from substratus_amd.serve.server import _maybe_load_weights
+
+    cfg = get_config("opt-125m")
+    ref = CausalLM(cfg, device="meta", dtype=torch.float32)
+    ref = ref.to_empty(device="cpu")
+    ref.init_random(9)
+    sd = {k: v.clone().contiguous() for k, v in ref.state_dict().items()}
+
+    d, hq, hkv = cfg.head_dim, cfg.num_heads, cfg.num_kv_heads
+    hf = {"model.decoder.embed_tokens.weight": sd["embed_tokens.weight"],
+          "model.decoder.final_layer_norm.weight": sd["final_norm.weight"],
+          "model.decoder.final_layer_norm.bias": sd["final_norm.bias"]}
+    # HF opt positions carry the +2 reserved rows
+    pos = sd["embed_positions.weight"]
+    hf["model.decoder.embed_positions.weight"] = torch.cat(
+        [torch.zeros(2, pos.shape[1]), pos], dim=0).contiguous()
+    for i in range(cfg.num_layers):
+        p, h = f"blocks.{i}.", f"model.decoder.layers.{i}."
+        for suf in ("weight", "bias"):
+            q, k, v = sd[p + f"wqkv.{suf}"].split(
+                [hq * d, hkv * d, hkv * d], dim=0)
+            hf[h + f"self_attn.q_proj.{suf}"] = q.contiguous()
+            hf[h + f"self_attn.k_proj.{suf}"] = k.contiguous()
+            hf[h + f"self_attn.v_proj.{suf}"] = v.contiguous()
+            hf[h + f"self_attn.out_proj.{suf}"] = sd[p + f"wo.{suf}"]
+            hf[h + f"fc1.{suf}"] = sd[p + f"w_up.{suf}"]
+            hf[h + f"fc2.{suf}"] = sd[p + f"w_down.{suf}"]
+            hf[h + f"self_attn_layer_norm.{suf}"] = sd[p + f"norm1.{suf}"]
+            hf[h + f"final_layer_norm.{suf}"] = sd[p + f"norm2.{suf}"]
+
+    def gen(state, tag):
+        mdir = tmp_path / tag
+        mdir.mkdir()
+        save_file(state, str(mdir / "model.safetensors"))
+        eng = Engine(EngineConfig(model="opt-125m", dtype=torch.float32,
+                                  max_batch=1, max_seq_len=64,
+                                  kv_blocks=16, seed=5))
+        _maybe_load_weights(eng, model_dir=str(mdir))
+        eng.add_request([5, 9, 2, 7], max_new_tokens=4)
+        toks = []
+        while eng.has_work():
+            toks += [o.token for o in eng.step()]
+        return toks
+
+    assert gen(hf, "hf") == gen(sd, "native")
