@@ -17,8 +17,11 @@ from torch.utils.cpp_extension import BuildExtension, CUDAExtension  # noqa: E40
 ROOT = os.path.dirname(os.path.abspath(__file__))
 HIP_DIR = os.path.join(ROOT, "substratus_amd", "ops", "hip")
 
+# Hand-written kernel sources only. torch's CUDAExtension hipify pass emits
+# *_hip.hip/*_hip.cpp siblings into the build dir; exclude any that leak
+# in-tree so the generated copies can never shadow the originals.
 sources = [os.path.join(HIP_DIR, f) for f in sorted(os.listdir(HIP_DIR))
-           if f.endswith(".hip")]
+           if f.endswith(".hip") and not f.endswith("_hip.hip")]
 
 setup(
     name="substratus_amd",
