@@ -119,10 +119,13 @@ def _w_kv(f: BinaryIO, key: str, vtype: int, packfmt: str, value) -> None:
 
 
 def write_gguf(path: str, tensors: dict[str, tuple[torch.Tensor, int, list[int]]],
-               metadata: Optional[dict] = None) -> None:
+               metadata: Optional[dict] = None, alignment: int = 32) -> None:
     """tensors: name -> (raw bytes tensor uint8 or float tensor, ggml_type,
-    logical_shape). metadata: str -> (str|int|float)."""
-    metadata = metadata or {}
+    logical_shape). metadata: str -> (str|int|float). alignment: data-section
+    alignment; recorded as general.alignment when non-default."""
+    metadata = dict(metadata or {})
+    if alignment != 32:
+        metadata["general.alignment"] = alignment
     with open(path, "wb") as f:
         f.write(struct.pack("<IIQQ", GGUF_MAGIC, GGUF_VERSION,
                             len(tensors), len(metadata)))
@@ -143,7 +146,7 @@ def write_gguf(path: str, tensors: dict[str, tuple[torch.Tensor, int, list[int]]
         for name, (data, gtype, shape) in tensors.items():
             nbytes = data.numel() * data.element_size()
             infos.append((name, gtype, shape, offset, data))
-            offset += (nbytes + 31) // 32 * 32  # 32-byte alignment
+            offset += -(-nbytes // alignment) * alignment
         for name, gtype, shape, off, _ in infos:
             _w_str(f, name)
             f.write(struct.pack("<I", len(shape)))
@@ -153,7 +156,7 @@ def write_gguf(path: str, tensors: dict[str, tuple[torch.Tensor, int, list[int]]
             f.write(struct.pack("<IQ", gtype, off))
         # align data section
         pos = f.tell()
-        pad = (32 - pos % 32) % 32
+        pad = (alignment - pos % alignment) % alignment
         f.write(b"\0" * pad)
         base = f.tell()
         for name, gtype, shape, off, data in infos:
@@ -223,7 +226,11 @@ def read_gguf(path: str) -> tuple[dict, dict[str, GGUFTensor]]:
             gtype, off = struct.unpack("<IQ", f.read(12))
             infos.append((name, shape, gtype, off))
         pos = f.tell()
-        base = pos + (32 - pos % 32) % 32
+        # the data section and per-tensor offsets are aligned to
+        # general.alignment (llama.cpp default 32) — honor non-default
+        # alignments from ecosystem files
+        align = int(metadata.get("general.alignment", 32))
+        base = pos + (align - pos % align) % align
         tensors = {}
         for name, shape, gtype, off in infos:
             f.seek(base + off)

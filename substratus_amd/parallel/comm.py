@@ -35,6 +35,11 @@ class ParallelState:
     dp_size: int = 1
     tp_group: Optional[object] = None
     dp_group: Optional[object] = None
+    # gloo group over the same TP ranks for host-side control messages
+    # (per-iteration admission broadcasts): broadcast_object_list over the
+    # RCCL group would stage through the GPU and synchronize the device
+    # every step — scheduling traffic belongs on the CPU.
+    tp_ctrl_group: Optional[object] = None
 
     @property
     def is_rank0(self) -> bool:
@@ -87,12 +92,17 @@ def init_distributed(tp_size: Optional[int] = None,
     tp = tp_size if tp_size is not None else world
     assert world % tp == 0, f"world {world} not divisible by tp {tp}"
     dp = world // tp
-    tp_group = dp_group = None
+    tp_group = dp_group = tp_ctrl_group = None
     if tp > 1 and tp < world:
         for i in range(dp):
-            g = dist.new_group(list(range(i * tp, (i + 1) * tp)))
+            ranks = list(range(i * tp, (i + 1) * tp))
+            g = dist.new_group(ranks)
+            gc = dist.new_group(ranks, backend="gloo") if backend == "nccl" \
+                else g
             if rank // tp == i:
-                tp_group = g
+                tp_group, tp_ctrl_group = g, gc
+    elif tp == world and tp > 1 and backend == "nccl":
+        tp_ctrl_group = dist.new_group(backend="gloo")
     if dp > 1 and dp < world:
         for j in range(tp):
             g = dist.new_group(list(range(j, world, tp)))
@@ -101,7 +111,7 @@ def init_distributed(tp_size: Optional[int] = None,
     _STATE = ParallelState(
         rank=rank, world_size=world, local_rank=local, initialized=True,
         tp_rank=rank % tp, tp_size=tp, dp_rank=rank // tp, dp_size=dp,
-        tp_group=tp_group, dp_group=dp_group)
+        tp_group=tp_group, dp_group=dp_group, tp_ctrl_group=tp_ctrl_group)
     return _STATE
 
 

@@ -45,6 +45,20 @@ class EngineConfig:
     kv_dtype: Optional[str] = None         # None=model dtype | "fp8" (e4m3)
     seed: int = 0
 
+    def to_wire(self) -> dict:
+        """Plain-dict form for broadcast_object_list: the TP leader ships its
+        EXACT config to follower ranks at startup so every rank builds an
+        identical engine (quant/kv_dtype/graph settings included)."""
+        d = self.__dict__.copy()
+        d["dtype"] = str(self.dtype).removeprefix("torch.")
+        return d
+
+    @classmethod
+    def from_wire(cls, d: dict) -> "EngineConfig":
+        d = dict(d)
+        d["dtype"] = getattr(torch, d["dtype"])
+        return cls(**d)
+
 
 @dataclass
 class StepOutput:
@@ -52,6 +66,7 @@ class StepOutput:
     token: int
     finished: bool
     text_done: bool = False
+    error: Optional[str] = None     # set => request failed, token is invalid
 
 
 class Engine:
@@ -127,7 +142,7 @@ class Engine:
         return bool(self.waiting or self.running)
 
     # ------------------------------------------------------------ scheduling
-    def _try_admit(self) -> list[SequenceState]:
+    def _try_admit(self, out: list[StepOutput]) -> list[SequenceState]:
         admitted = []
         budget = self.cfg.prefill_chunk
         while self.waiting and len(self.running) + len(admitted) < self.cfg.max_batch:
@@ -139,12 +154,15 @@ class Engine:
                 # never be admitted (livelock)
                 break
             if nblk > self.kv.num_blocks:
-                # can never fit even an empty cache — fail the request
-                # loudly instead of spinning forever
+                # can never fit even an empty cache — fail ONLY this request
+                # (error StepOutput) and keep scheduling the rest; one
+                # poisoned request must not take down the batch
                 self.waiting.pop(0)
-                raise RuntimeError(
-                    f"request {seq.seq_id} needs {nblk} KV blocks but the "
-                    f"cache has only {self.kv.num_blocks}")
+                out.append(StepOutput(
+                    seq.seq_id, 0, True,
+                    error=(f"request needs {nblk} KV blocks but the cache "
+                           f"has only {self.kv.num_blocks}")))
+                continue
             if self.kv.allocator.free_count < nblk + 2 * len(self.running):
                 break  # keep headroom for running seqs to grow
             seq.blocks = self.kv.allocator.alloc(nblk)
@@ -157,8 +175,9 @@ class Engine:
         seq = self.running.pop()  # youngest = most recently admitted
         self.kv.allocator.free(seq.blocks)
         seq.blocks = []
-        # drop generated tokens? keep them: re-prefill the whole prefix.
-        seq.num_prompt = seq.length
+        # keep generated tokens: readmission re-prefills the whole token list
+        # (prompt + generated so far). num_prompt stays at the true prompt
+        # length so the max_new_tokens budget is not reset by preemption.
         self.waiting.insert(0, seq)
 
     # --------------------------------------------------------------- prefill
@@ -183,7 +202,11 @@ class Engine:
     def _decode(self) -> list[StepOutput]:
         if not self.running:
             return []
-        if self.cfg.use_graphs and self.device.type == "cuda":
+        # hipGraph capture of a forward containing RCCL collectives
+        # (dist.all_reduce inside the row-parallel layers) is unvalidated on
+        # ROCm — TP decodes eagerly until graph+RCCL capture is proven.
+        if (self.cfg.use_graphs and self.device.type == "cuda"
+                and state().tp_size <= 1):
             return self._decode_graph()
         return self._decode_eager()
 
@@ -280,7 +303,7 @@ class Engine:
     def step(self) -> list[StepOutput]:
         """One engine iteration: admit+prefill, then decode everyone running."""
         out: list[StepOutput] = []
-        admitted = self._try_admit()
+        admitted = self._try_admit(out)
         if admitted:
             out.extend(self._prefill(admitted))
             self.running.extend(admitted)
@@ -298,6 +321,19 @@ class Engine:
         self._retire(self.running, res)
         self.steps += 1
         return res
+
+    def abort_all(self) -> None:
+        """Drop every waiting/running request and free their KV blocks.
+        Whole-engine failure recovery; deterministic, so TP leader and
+        followers can apply it independently and stay lockstep."""
+        self.waiting.clear()
+        for s in self.running:
+            if s.blocks:
+                self.kv.allocator.free(s.blocks)
+                s.blocks = []
+        self.running.clear()
+        if self.graph_runner is not None:
+            self.graph_runner.invalidate()
 
     def _retire(self, pool: list[SequenceState], outs: list[StepOutput]) -> None:
         finished = {o.seq_id for o in outs if o.finished}

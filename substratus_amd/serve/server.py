@@ -111,22 +111,20 @@ class InferenceWorker:
                 try:
                     outs = self.engine.step()
                 except Exception as e:  # noqa: BLE001
-                    # engine-level failure (e.g. a request that can never
-                    # fit the KV cache): fail the waiting requests instead
-                    # of killing the worker thread — serving must survive
+                    # whole-engine failure (per-request problems like an
+                    # unfittable prompt are returned as error StepOutputs by
+                    # engine.step, not raised): fail all in-flight requests
+                    # WITH an error marker — handlers answer 5xx, never a
+                    # fake completion — and keep the worker thread serving
                     import sys as _sys
                     print(f"[server] engine.step failed: {e}",
                           file=_sys.stderr, flush=True)
                     with self.lock:
                         for sid, q_ in list(self.outputs.items()):
-                            q_.put(StepOutput(sid, 0, True))
+                            q_.put(StepOutput(sid, 0, True,
+                                              error=f"engine failure: {e}"))
                             self.outputs.pop(sid, None)
-                    self.engine.waiting.clear()
-                    for s in list(self.engine.running):
-                        if s.blocks:
-                            self.engine.kv.allocator.free(s.blocks)
-                            s.blocks = []
-                    self.engine.running.clear()
+                    self.engine.abort_all()
                     continue
                 dt = time.time() - t0
                 with self.lock:
@@ -148,34 +146,56 @@ class InferenceWorker:
         self._stop = True
 
 
+def build_engine_config(params: dict,
+                        model_name: Optional[str] = None
+                        ) -> tuple[EngineConfig, Optional[str]]:
+    """Resolve the serving configuration (model, dtype, quant, GGUF path)
+    from params/env. Under TP only the leader runs this; followers receive
+    the result over the wire so every rank's engine is identical."""
+    name = model_name or resolve_model_name(params)
+    on_gpu = torch.cuda.is_available()
+    quant = None
+    # reference knob: MODEL_LOAD_IN_8BIT / _4BIT env
+    # (examples/llama2-70b/server.yaml)
+    if params.get("load_in_4bit") or os.environ.get("MODEL_LOAD_IN_4BIT"):
+        quant = "q4_0"
+    if params.get("load_in_8bit") or os.environ.get("MODEL_LOAD_IN_8BIT"):
+        quant = "q8_0"
+    gguf_path = _find_gguf()
+    ecfg = EngineConfig(
+        model=name,
+        dtype=torch.bfloat16 if on_gpu else torch.float32,
+        max_batch=int(params.get("max_batch", 64)),
+        max_seq_len=int(params.get("max_length", 2048)),
+        quant=None if gguf_path else quant,
+    )
+    return ecfg, gguf_path
+
+
+def build_engine(ecfg: EngineConfig,
+                 gguf_path: Optional[str] = None) -> Engine:
+    """Construct the engine + load weights. Identical on every TP rank."""
+    engine = Engine(ecfg)
+    if gguf_path:
+        from ..models.quant import load_gguf_into_model
+        from ..parallel.comm import state as pstate
+        if pstate().tp_size > 1:
+            raise RuntimeError(
+                "GGUF-quantized serving is single-rank for now: remove the "
+                "*.gguf from the model dir or set gpu.count to 1")
+        load_gguf_into_model(engine.model, gguf_path, engine.device,
+                             engine.cfg.dtype)
+    else:
+        _maybe_load_weights(engine)
+    return engine
+
+
 def create_app(engine: Optional[Engine] = None,
                model_name: Optional[str] = None) -> FastAPI:
     params = load_params()
     if engine is None:
-        name = model_name or resolve_model_name(params)
-        on_gpu = torch.cuda.is_available()
-        quant = None
-        # reference knob: MODEL_LOAD_IN_8BIT / _4BIT env
-        # (examples/llama2-70b/server.yaml)
-        if params.get("load_in_4bit") or os.environ.get("MODEL_LOAD_IN_4BIT"):
-            quant = "q4_0"
-        if params.get("load_in_8bit") or os.environ.get("MODEL_LOAD_IN_8BIT"):
-            quant = "q8_0"
-        gguf_path = _find_gguf()
-        ecfg = EngineConfig(
-            model=name,
-            dtype=torch.bfloat16 if on_gpu else torch.float32,
-            max_batch=int(params.get("max_batch", 64)),
-            max_seq_len=int(params.get("max_length", 2048)),
-            quant=None if gguf_path else quant,
-        )
-        engine = Engine(ecfg)
-        if gguf_path:
-            from ..models.quant import load_gguf_into_model
-            load_gguf_into_model(engine.model, gguf_path, engine.device,
-                                 engine.cfg.dtype)
-        else:
-            _maybe_load_weights(engine)
+        ecfg, gguf_path = build_engine_config(params, model_name)
+        engine = build_engine(ecfg, gguf_path)
     worker = InferenceWorker(engine)
     app = FastAPI(title="substratus-amd model server")
     app.state.worker = worker
@@ -188,18 +208,29 @@ def create_app(engine: Optional[Engine] = None,
                                                 "/content/model"), vocab)
 
     try:
-        from prometheus_client import (CONTENT_TYPE_LATEST, REGISTRY, Gauge,
-                                       Counter, generate_latest)
-        try:
-            tokens_counter = Counter("served_tokens_total",
-                                     "tokens generated")
-            hbm_gauge = Gauge("hbm_allocated_bytes", "GPU memory allocated")
-        except ValueError:
-            # app re-created in the same process (tests, embedded platform
-            # restart): reuse the registered collectors
-            tokens_counter = REGISTRY._names_to_collectors[
-                "served_tokens_total"]
-            hbm_gauge = REGISTRY._names_to_collectors["hbm_allocated_bytes"]
+        from prometheus_client import CONTENT_TYPE_LATEST, generate_latest
+        from prometheus_client.core import (CollectorRegistry,
+                                            CounterMetricFamily,
+                                            GaugeMetricFamily)
+
+        class _ServerCollector:
+            """Reads live server state at scrape time — public collector
+            API only, and a per-app registry so app re-creation in one
+            process (tests, embedded restarts) never collides."""
+
+            def collect(self):
+                c = CounterMetricFamily("served_tokens", "tokens generated")
+                c.add_metric([], worker.total_tokens)
+                yield c
+                g = GaugeMetricFamily("hbm_allocated_bytes",
+                                      "GPU memory allocated")
+                g.add_metric([], torch.cuda.memory_allocated()
+                             if torch.cuda.is_available() else 0)
+                yield g
+
+        prom_registry = CollectorRegistry()
+        prom_registry.register(_ServerCollector())
+        app.state.prom_registry = prom_registry
         have_prom = True
     except ImportError:  # pragma: no cover
         have_prom = False
@@ -232,10 +263,8 @@ def create_app(engine: Optional[Engine] = None,
         @app.get("/metrics")
         def metrics():
             from fastapi import Response
-            tokens_counter._value.set(worker.total_tokens)  # type: ignore
-            if torch.cuda.is_available():
-                hbm_gauge.set(torch.cuda.memory_allocated())
-            return Response(generate_latest(), media_type=CONTENT_TYPE_LATEST)
+            return Response(generate_latest(prom_registry),
+                            media_type=CONTENT_TYPE_LATEST)
 
     def _bad_request(msg: str) -> JSONResponse:
         # OpenAI-style error body, 400 instead of a 500 from deep asserts
@@ -277,19 +306,26 @@ def create_app(engine: Optional[Engine] = None,
         sid, out_q = worker.submit(prompt_tokens, max_tokens, temperature,
                                    top_k, top_p, eos_token=tok_obj.eos)
 
-        def collect() -> list[int]:
+        def collect() -> tuple[list[int], Optional[str]]:
             toks = []
             while True:
                 o: StepOutput = out_q.get(timeout=600)
+                if o.error:
+                    return toks, o.error
                 toks.append(o.token)
                 if o.finished:
-                    return toks
+                    return toks, None
 
         if stream:
             def gen():
-                idx = 0
                 while True:
                     o: StepOutput = out_q.get(timeout=600)
+                    if o.error:
+                        err = {"error": {"message": o.error,
+                                         "type": "server_error"}}
+                        yield f"data: {json.dumps(err)}\n\n"
+                        yield "data: [DONE]\n\n"
+                        return
                     chunk = {
                         "id": f"cmpl-{sid}", "object": "text_completion",
                         "model": model_id,
@@ -299,13 +335,16 @@ def create_app(engine: Optional[Engine] = None,
                                      else None}],
                     }
                     yield f"data: {json.dumps(chunk)}\n\n"
-                    idx += 1
                     if o.finished:
                         yield "data: [DONE]\n\n"
                         return
             return StreamingResponse(gen(), media_type="text/event-stream")
 
-        toks = collect()
+        toks, err = collect()
+        if err is not None:
+            return JSONResponse({"error": {"message": err,
+                                           "type": "server_error"}},
+                                status_code=500)
         return JSONResponse({
             "id": f"cmpl-{sid}",
             "object": "text_completion",
@@ -376,24 +415,28 @@ def _maybe_load_weights(engine: Engine,
 
 def main() -> None:
     import uvicorn
-    from ..parallel.comm import init_distributed, state as pstate
+    from ..parallel.comm import init_distributed
     st = init_distributed()  # tp over all ranks of the Pod (amd.com/gpu: N)
-    if st.initialized and st.tp_rank != 0:
-        # follower rank: build the engine, mirror the leader's schedule
-        params = load_params()
-        name = resolve_model_name(params)
-        ecfg = EngineConfig(
-            model=name,
-            dtype=torch.bfloat16 if torch.cuda.is_available() else torch.float32,
-            max_batch=int(params.get("max_batch", 64)),
-            max_seq_len=int(params.get("max_length", 2048)),
-        )
-        engine = Engine(ecfg)
-        _maybe_load_weights(engine)
-        from .tp import follower_loop
-        follower_loop(engine)
-        return
-    app = create_app()
+    if st.initialized and st.tp_size > 1:
+        # All ranks build their engine from the LEADER's exact resolved
+        # configuration (one startup broadcast): a follower that guessed
+        # its own quant/GGUF/kv_dtype settings would desync the first
+        # collective of the first forward.
+        from .tp import TPCoordinator, follower_loop
+        coord = TPCoordinator()
+        if st.tp_rank == 0:
+            ecfg, gguf_path = build_engine_config(load_params())
+            coord.broadcast_config([ecfg.to_wire(), gguf_path])
+        else:
+            wire, gguf_path = coord.broadcast_config()
+            ecfg = EngineConfig.from_wire(wire)
+        engine = build_engine(ecfg, gguf_path)
+        if st.tp_rank != 0:
+            follower_loop(engine)
+            return
+        app = create_app(engine=engine)
+    else:
+        app = create_app()
     uvicorn.run(app, host="0.0.0.0", port=int(os.environ.get("PORT", 8080)))
 
 

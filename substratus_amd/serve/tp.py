@@ -32,24 +32,43 @@ class TPCoordinator:
     def active(self) -> bool:
         return self.st.initialized and self.st.tp_size > 1
 
+    @property
+    def _ctrl(self):
+        # host-side control messages ride the gloo group (no GPU sync);
+        # tp_group only when running pure-gloo (CPU tests)
+        return self.st.tp_ctrl_group or self.st.tp_group
+
     def publish(self, adds: list[dict], stop: bool = False) -> None:
         """Leader: broadcast this iteration's admissions to followers."""
         if not self.active:
             return
         obj = [{"adds": adds, "stop": stop}]
         dist.broadcast_object_list(obj, src=self.st.rank - self.st.tp_rank,
-                                   group=self.st.tp_group)
+                                   group=self._ctrl)
 
     def receive(self) -> dict:
         """Follower: receive the leader's iteration command."""
         obj = [None]
         dist.broadcast_object_list(obj, src=self.st.rank - self.st.tp_rank,
-                                   group=self.st.tp_group)
+                                   group=self._ctrl)
         return obj[0]
+
+    def broadcast_config(self, payload: Optional[list] = None) -> list:
+        """One-shot startup broadcast of the leader's engine configuration
+        (EngineConfig wire dict + gguf path) so follower ranks build an
+        IDENTICAL engine — quant/GGUF/kv_dtype desync between ranks hangs
+        the first collective otherwise."""
+        obj = list(payload) if payload is not None else [None, None]
+        dist.broadcast_object_list(obj, src=self.st.rank - self.st.tp_rank,
+                                   group=self._ctrl)
+        return obj
 
 
 def follower_loop(engine) -> None:
-    """Ranks 1..tp-1: mirror the leader's admissions and step in lockstep."""
+    """Ranks 1..tp-1: mirror the leader's admissions and step in lockstep.
+    A step failure is handled exactly like the leader's worker handles it
+    (engine.abort_all) — both sides recover to the same empty schedule
+    instead of the follower dying and hanging the next collective."""
     coord = TPCoordinator()
     while True:
         cmd = coord.receive()
@@ -58,4 +77,7 @@ def follower_loop(engine) -> None:
         for add in cmd["adds"]:
             engine.add_request(**add)
         if engine.has_work():
-            engine.step()
+            try:
+                engine.step()
+            except Exception:  # noqa: BLE001  (mirror leader recovery)
+                engine.abort_all()

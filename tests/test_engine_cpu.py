@@ -194,15 +194,24 @@ def test_engine_long_prompt_exceeding_prefill_chunk():
     assert len(toks) == 3, toks
 
 
-def test_engine_impossible_request_raises():
-    """A request needing more KV blocks than the whole cache fails loudly
-    instead of spinning in the waiting queue."""
-    import pytest as _pytest
+def test_engine_impossible_request_fails_only_itself():
+    """A request needing more KV blocks than the whole cache fails as an
+    error StepOutput (not an engine-wide exception) while a concurrent
+    fitting request completes normally."""
     import torch
     from substratus_amd.serve.engine import Engine, EngineConfig
     eng = Engine(EngineConfig(model="llama-tiny", dtype=torch.float32,
-                              max_batch=2, max_seq_len=96, kv_blocks=2))
-    eng.add_request(list(range(90)), max_new_tokens=4)
-    with _pytest.raises(RuntimeError, match="KV blocks"):
-        for _ in range(5):
-            eng.step()
+                              max_batch=2, max_seq_len=96, kv_blocks=3))
+    bad = eng.add_request(list(range(90)), max_new_tokens=4)
+    good = eng.add_request([1, 2, 3], max_new_tokens=2)
+    outs = []
+    for _ in range(10):
+        if not eng.has_work():
+            break
+        outs.extend(eng.step())
+    bad_outs = [o for o in outs if o.seq_id == bad]
+    good_outs = [o for o in outs if o.seq_id == good]
+    assert len(bad_outs) == 1 and bad_outs[0].error is not None \
+        and "KV blocks" in bad_outs[0].error and bad_outs[0].finished
+    assert len(good_outs) == 2 and all(o.error is None for o in good_outs) \
+        and good_outs[-1].finished

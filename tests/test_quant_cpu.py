@@ -210,3 +210,20 @@ def test_llamacpp_named_gguf_loads(tmp_path):
     gguf.write_gguf(npath, native, {"substratus.quant": "q4_0"})
 
     assert gen(path) == gen(npath)
+
+
+def test_gguf_nondefault_alignment_roundtrip(tmp_path):
+    """Ecosystem GGUF files may set general.alignment != 32; the reader must
+    honor it for both the data-section base and per-tensor offsets."""
+    import torch
+    from substratus_amd.models.gguf import (GGML_F32, read_gguf, write_gguf)
+    a = torch.randn(7, 32)
+    b = torch.randn(3, 64)
+    path = str(tmp_path / "aligned.gguf")
+    write_gguf(path, {"a": (a, GGML_F32, list(a.shape)),
+                      "b": (b, GGML_F32, list(b.shape))},
+               metadata={"x": "y"}, alignment=64)
+    meta, tensors = read_gguf(path)
+    assert meta["general.alignment"] == 64
+    assert torch.equal(tensors["a"].dequantize(), a)
+    assert torch.equal(tensors["b"].dequantize(), b)

@@ -179,7 +179,8 @@ def test_worker_survives_engine_failure(tmp_path):
     with TestClient(app) as c:
         r = c.post("/v1/completions",
                    json={"prompt": list(range(100)), "max_tokens": 2})
-        assert r.status_code == 200  # failed request completes (empty-ish)
+        assert r.status_code == 500  # failed request is an explicit error
+        assert "error" in r.json()
         # the worker is still alive and serves a normal request
         ok = c.post("/v1/completions", json={"prompt": [1, 2, 3],
                                              "max_tokens": 2})
@@ -362,3 +363,32 @@ def test_hf_opt_checkpoint_serves(tmp_path):
         return toks
 
     assert gen(hf, "hf") == gen(sd, "native")
+
+
+def test_engine_failure_blast_radius():
+    """One unfittable request gets a 5xx while a concurrent request
+    completes normally (a poisoned request must not
+    take down the batch)."""
+    import concurrent.futures as cf
+    cfg = EngineConfig(model="llama-tiny", dtype=torch.float32, max_batch=4,
+                       max_seq_len=256, kv_blocks=4)
+    eng = Engine(cfg, device="cpu")
+    app = create_app(engine=eng)
+    with TestClient(app) as c:
+        def bad():
+            # 90 tokens passes the HTTP max_length check (256) but needs 6
+            # KV blocks out of a 4-block cache -> per-request engine error
+            return c.post("/v1/completions", json={
+                "prompt": list(range(90)), "max_tokens": 4})
+
+        def good():
+            return c.post("/v1/completions", json={
+                "prompt": "hello", "max_tokens": 4})
+
+        with cf.ThreadPoolExecutor(2) as ex:
+            fb, fg = ex.submit(bad), ex.submit(good)
+            rb, rg = fb.result(), fg.result()
+    assert rb.status_code == 500
+    assert "KV blocks" in rb.json()["error"]["message"]
+    assert rg.status_code == 200
+    assert rg.json()["usage"]["completion_tokens"] >= 1
