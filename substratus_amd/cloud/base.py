@@ -77,6 +77,14 @@ class Cloud(ABC):
                      mount: "BucketMount") -> None:
         """Wire the object's artifact bucket into a Pod spec."""
 
+    def read_artifact_json(self, obj: dict, relpath: str) -> Optional[dict]:
+        """Read a small JSON artifact (e.g. completed.json) from the
+        object's bucket, or None when the cloud can't reach it from the
+        controller (GCS/S3 without mounted credentials). The Model
+        controller uses this to surface loader-reported metadata
+        (sizeBytes) into status."""
+        return None
+
 
 class BucketMount:
     def __init__(self, name: str, mount_path: str, read_only: bool = False):

@@ -38,6 +38,18 @@ class KindCloud(Cloud):
     def get_principal(self, sa: dict) -> tuple[str, bool]:
         return "", True
 
+    def read_artifact_json(self, obj: dict, relpath: str):
+        """hostPath bucket is directly readable by the controller."""
+        import json
+        url = self.object_artifact_url(obj)
+        digest = url.rsplit("/", 1)[-1]
+        path = os.path.join(self.bucket_dir, digest, "artifacts", relpath)
+        try:
+            with open(path) as f:
+                return json.load(f)
+        except (OSError, ValueError):
+            return None
+
     def mount_bucket(self, pod_metadata: dict, pod_spec: dict, obj: dict,
                     mount: BucketMount) -> None:
         """hostPath volume at {bucket_dir}/{hash}/artifacts

@@ -88,8 +88,13 @@ class ModelReconciler:
             return Result.wait(3)
         apiv1.set_condition(obj, cond.MODELLED, True, cond.REASON_JOB_COMPLETE)
         apiv1.set_status_ready(obj, True)
-        obj.setdefault("status", {})["artifacts"] = {
-            "url": self.cloud.object_artifact_url(obj)}
+        artifacts = {"url": self.cloud.object_artifact_url(obj)}
+        # loader writes sizeBytes into completed.json; surface it so the
+        # Server controller's HBM admission has a real number to check
+        completed = self.cloud.read_artifact_json(obj, "completed.json")
+        if completed and completed.get("sizeBytes") is not None:
+            artifacts["sizeBytes"] = int(completed["sizeBytes"])
+        obj.setdefault("status", {})["artifacts"] = artifacts
         return Result()
 
     def _modeller_job(self, obj: dict, base_model, dataset) -> dict:

@@ -54,10 +54,25 @@ def run(artifacts_dir: str = "/content/artifacts",
     else:
         _write_random_init(name, artifacts_dir, params)
 
+    # total artifact byte size: the controller propagates this into
+    # Model.status.artifacts.sizeBytes and the Server controller's HBM
+    # admission (gpu.count × 288 GB) reads it — without it admission can
+    # never trigger
+    size = _artifacts_size(artifacts_dir)
     with open(sentinel, "w") as f:
-        json.dump({"completed": True, "name": name}, f)
-    print(f"loaded {name} into {artifacts_dir}")
+        json.dump({"completed": True, "name": name, "sizeBytes": size}, f)
+    print(f"loaded {name} into {artifacts_dir} ({size} bytes)")
     return artifacts_dir
+
+
+def _artifacts_size(artifacts_dir: str) -> int:
+    total = 0
+    for root, _dirs, files in os.walk(artifacts_dir):
+        for fn in files:
+            if fn == "completed.json":
+                continue
+            total += os.path.getsize(os.path.join(root, fn))
+    return total
 
 
 def _try_hub_download(name: str, files_filter: str, out: str) -> bool:
