@@ -185,23 +185,33 @@ class Block(nn.Module):
         d = cfg.head_dim
         q, k, v = self.wqkv(y).split(
             [self.n_heads * d, self.n_kv_heads * d, self.n_kv_heads * d], dim=-1)
-        q = q.view(b, s, self.n_heads, d).transpose(1, 2)
-        k = k.view(b, s, self.n_kv_heads, d).transpose(1, 2)
-        v = v.view(b, s, self.n_kv_heads, d).transpose(1, 2)
-        if self.rope_theta > 0:
-            q = _rope_train(q, cos, sin)
-            k = _rope_train(k, cos, sin)
-        if self.n_kv_heads != self.n_heads:
-            # Materialized expansion measured FASTER than SDPA's native
-            # enable_gqa on ROCm 7.2 (falcon mb4: 498 vs 602 ms/step — the
-            # aotriton flash backend declines grouped heads and SDPA falls
-            # back to a slower path), so the ~19 ms/step of expansion
-            # copies stays. Re-test when aotriton gains enable_gqa.
-            rep = self.n_heads // self.n_kv_heads
-            k = k.repeat_interleave(rep, dim=1)
-            v = v.repeat_interleave(rep, dim=1)
-        a = F.scaled_dot_product_attention(q, k, v, is_causal=True)
-        a = a.transpose(1, 2).reshape(b, s, self.n_heads * d)
+        if (h.is_cuda and h.dtype == torch.bfloat16
+                and ops.flash_attn_train_available(d)):
+            # first-party MFMA flash fwd+bwd (attention_prefill.hip /
+            # attention_bwd.hip): BSHD layout end-to-end — no transposes,
+            # no GQA head materialization, no torch-SDPA/aotriton on the
+            # trainer hot path
+            q = q.view(b, s, self.n_heads, d)
+            k = k.view(b, s, self.n_kv_heads, d)
+            v = v.view(b, s, self.n_kv_heads, d)
+            if self.rope_theta > 0:
+                q = _rope_train(q, cos, sin, bshd=True)
+                k = _rope_train(k, cos, sin, bshd=True)
+            a = ops.flash_attn_train(q, k, v, self.scale)
+            a = a.reshape(b, s, self.n_heads * d)
+        else:
+            q = q.view(b, s, self.n_heads, d).transpose(1, 2)
+            k = k.view(b, s, self.n_kv_heads, d).transpose(1, 2)
+            v = v.view(b, s, self.n_kv_heads, d).transpose(1, 2)
+            if self.rope_theta > 0:
+                q = _rope_train(q, cos, sin)
+                k = _rope_train(k, cos, sin)
+            if self.n_kv_heads != self.n_heads:
+                rep = self.n_heads // self.n_kv_heads
+                k = k.repeat_interleave(rep, dim=1)
+                v = v.repeat_interleave(rep, dim=1)
+            a = F.scaled_dot_product_attention(q, k, v, is_causal=True)
+            a = a.transpose(1, 2).reshape(b, s, self.n_heads * d)
         attn_out = self.wo(a)
         if self.parallel_attn:
             if self.parallel_norms:
@@ -230,12 +240,17 @@ class Block(nn.Module):
         return self.w_down(F.gelu(self.w_up(y), approximate="tanh"))
 
 
-def _rope_train(x: torch.Tensor, cos: torch.Tensor, sin: torch.Tensor) -> torch.Tensor:
-    # x: [B, H, S, D]; cos/sin: [S, D/2]
+def _rope_train(x: torch.Tensor, cos: torch.Tensor, sin: torch.Tensor,
+                bshd: bool = False) -> torch.Tensor:
+    # x: [B, H, S, D] (or [B, S, H, D] when bshd); cos/sin: [S, D/2]
     d = x.shape[-1]
     x1, x2 = x[..., : d // 2], x[..., d // 2:]
-    c = cos[None, None, :, :]
-    s = sin[None, None, :, :]
+    if bshd:
+        c = cos[None, :, None, :]
+        s = sin[None, :, None, :]
+    else:
+        c = cos[None, None, :, :]
+        s = sin[None, None, :, :]
     return torch.cat([x1 * c - x2 * s, x2 * c + x1 * s], dim=-1).to(x.dtype)
 
 

