@@ -161,6 +161,41 @@ def bench_wstream(m=256):
               f"parity={'OK' if ok else f'FAIL err={err:.4f}/{scale:.1f}'}")
 
 
+def bench_mgemm(m=512):
+    """mfma_gemm v3 (256x256 glds tile) vs hipBLASLt, interleaved rounds
+    (guide rule 24) + parity vs fp32 matmul. Shapes: llama2-7b B=512 decode
+    projections + logits."""
+    import torch.nn.functional as F
+    from substratus_amd import _C, ops
+    shapes = [("qkv", 12288, 4096), ("o", 4096, 4096),
+              ("gateup", 22016, 4096), ("down", 4096, 11008),
+              ("logits", 32000, 4096)]
+    for name, n, k in shapes:
+        torch.manual_seed(0)
+        x = (torch.rand(m, k, device="cuda") * 2 - 1).bfloat16()
+        w = ((torch.rand(n, k, device="cuda") * 2 - 1) * 0.02).bfloat16()
+        y = torch.empty(m, n, device="cuda").bfloat16()
+        splitk = int(_C.mfma_gemm_pick_splitk(m, n, k))
+        part = (torch.empty(splitk * m * n, dtype=torch.float32,
+                            device="cuda") if splitk > 1 else None)
+        _C.mfma_gemm(y, x, w, None, part, splitk)
+        ref = (x.float() @ w.float().t())
+        err = (y.float() - ref).abs().max().item()
+        rel = err / ref.abs().max().item()
+        # interleaved A/B rounds
+        ours, blas = [], []
+        for _ in range(5):
+            ours.append(timeit(
+                lambda: _C.mfma_gemm(y, x, w, None, part, splitk), iters=50))
+            blas.append(timeit(lambda: F.linear(x, w), iters=50))
+        dt_o, dt_b = min(ours), min(blas)
+        wb = n * k * 2
+        print(f"mgemm {name} M={m} N={n} K={k} splitk={splitk}: "
+              f"ours {dt_o*1e6:.1f} us ({wb/dt_o/1e12:.2f} TB/s W, "
+              f"{2*m*n*k/dt_o/1e12:.0f} TF) vs hipBLASLt {dt_b*1e6:.1f} us "
+              f"({wb/dt_b/1e12:.2f} TB/s) | maxerr {err:.4f} rel {rel:.2e}")
+
+
 def bench_quant_gemv(n=13824, k=5120, quant="q4_0"):
     from substratus_amd.parallel.tp import Linear
     from substratus_amd.models.quant import QuantLinear
@@ -201,6 +236,9 @@ if __name__ == "__main__":
     if a.which in ("all", "wstream"):
         bench_wstream()
         bench_wstream(m=64)
+    if a.which in ("all", "mgemm"):
+        m = int(os.environ.get("MGEMM_M", "512"))
+        bench_mgemm(m)
     if a.which in ("all", "quant"):
         bench_quant_gemv()
         bench_quant_gemv(quant="q8_0")

@@ -191,6 +191,50 @@ def flash_attn_train_available(d: int) -> bool:
     return have_ext() and d in (64, 128)
 
 
+# decode-projection GEMM (mfma_gemm.hip): split-K fp32 workspaces, cached
+# per (splitk, m, n) so hipGraph capture replays into stable buffers
+_GEMM_WS: dict = {}
+
+
+def linear(x: torch.Tensor, weight: torch.Tensor,
+           bias: Optional[torch.Tensor] = None) -> torch.Tensor:
+    """F.linear, with an opt-in route (SUB_MGEMM=1) to the first-party
+    256x256 glds GEMM (mfma_gemm.hip) on decode-shaped inference cases.
+
+    NOT on by default: the round-3 measurement (profiles/mgemm_r02.md)
+    shows hipBLASLt's MT192/MT256 solutions at 1.3-4.1 TB/s of weight
+    stream on the llama2-7b decode shapes vs our kernel's 1.0-2.25 — the
+    library already implements the deep-pipelined schedule whose measured
+    ceiling on this chip (~1.33-1.47 PF, guide 8-phase template) equals
+    hipBLASLt's best here. The kernel stays for measurement and as the
+    base for a future counted-vmcnt schedule."""
+    ext = _dispatch(x)
+    if (ext is None or torch.is_grad_enabled() and (
+            x.requires_grad or weight.requires_grad)
+            or x.dtype != torch.bfloat16
+            or os.environ.get("SUB_MGEMM") != "1"):
+        return torch.nn.functional.linear(x, weight, bias)
+    x2 = x.reshape(-1, x.shape[-1])
+    m, k = x2.shape
+    n = weight.shape[0]
+    if not (256 <= m <= 1024):
+        return torch.nn.functional.linear(x, weight, bias)
+    splitk = int(ext.mfma_gemm_pick_splitk(m, n, k))
+    if splitk == 0:   # unsupported shape
+        return torch.nn.functional.linear(x, weight, bias)
+    y = x.new_empty(*x.shape[:-1], n)
+    part = None
+    if splitk > 1:
+        key = (x.device.index, splitk, m, n)
+        part = _GEMM_WS.get(key)
+        if part is None:
+            part = torch.empty(splitk * m * n, dtype=torch.float32,
+                               device=x.device)
+            _GEMM_WS[key] = part
+    ext.mfma_gemm(y.view(m, n), x2.contiguous(), weight, bias, part, splitk)
+    return y
+
+
 def silu_mul(x: torch.Tensor) -> torch.Tensor:
     ext = _dispatch(x)
     if ext is None:

@@ -33,6 +33,9 @@ void launch_dequant(bf16_t*, const void*, const unsigned char*, long, int, bool,
 int decode_gemm_splitk(int, int);
 bool decode_gemm_supported(int, int, int);
 void launch_decode_gemm(bf16_t*, const bf16_t*, const bf16_t*, float*, int, int, int, int, hipStream_t);
+int mfma_gemm_splitk(int, int, int);
+bool mfma_gemm_supported(int, int, int);
+void launch_mfma_gemm(bf16_t*, const bf16_t*, const bf16_t*, const bf16_t*, float*, int, int, int, int, hipStream_t);
 int wstream_splitk(int, int);
 bool wstream_supported(int, int, int);
 void launch_wstream_gemm(bf16_t*, const bf16_t*, const bf16_t*, float*, int, int, int, int, hipStream_t);
@@ -268,6 +271,32 @@ void attn_prefill_probe(long variant, torch::Tensor out, torch::Tensor q,
   post_launch_check("attn_prefill_probe");
 }
 
+void mfma_gemm(torch::Tensor y, torch::Tensor x, torch::Tensor w,
+               c10::optional<torch::Tensor> bias,
+               c10::optional<torch::Tensor> part, long splitk) {
+  check_bf16(y, "y"); check_bf16(x, "x"); check_bf16(w, "w");
+  int m = x.size(0), k = x.size(1), n = w.size(0);
+  TORCH_CHECK(w.size(1) == k, "shape mismatch");
+  TORCH_CHECK(mfma_gemm_supported(m, n, k), "unsupported mfma_gemm shape ",
+              m, "x", n, "x", k);
+  const bf16_t* bp = nullptr;
+  if (bias.has_value()) { check_bf16(*bias, "bias"); bp = bfc(*bias); }
+  float* pp = nullptr;
+  if (splitk > 1) {
+    TORCH_CHECK(part.has_value() &&
+                part->numel() >= splitk * (long)m * n, "part too small");
+    pp = part->data_ptr<float>();
+  }
+  launch_mfma_gemm(bf(y), bfc(x), bfc(w), bp, pp, m, n, k, (int)splitk,
+                   stream());
+  post_launch_check("mfma_gemm");
+}
+
+long mfma_gemm_pick_splitk(long m, long n, long k) {
+  if (!mfma_gemm_supported((int)m, (int)n, (int)k)) return 0;
+  return mfma_gemm_splitk((int)m, (int)n, (int)k);
+}
+
 torch::Tensor mfma_probe(torch::Tensor a, torch::Tensor b) {
   check_bf16(a, "a"); check_bf16(b, "b");
   TORCH_CHECK(a.size(0) == 16 && a.size(1) == 32 && b.size(0) == 32 &&
@@ -418,6 +447,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("paged_attn_decode", &paged_attn_decode);
   m.def("attn_prefill", &attn_prefill);
   m.def("attn_train_fwd", &attn_train_fwd);
+  m.def("mfma_gemm", &mfma_gemm);
+  m.def("mfma_gemm_pick_splitk", &mfma_gemm_pick_splitk);
   m.def("attn_train_bwd", &attn_train_bwd);
   m.def("mfma_probe", &mfma_probe);
   m.def("attn_prefill_probe", &attn_prefill_probe);

@@ -1,0 +1,281 @@
+// Decode-projection GEMM v3: Y[M, N] = X[M, K] · W^T (W [N, K] row-major,
+// torch Linear layout) on the guide's 256×256-tile glds structure — the
+// round-3 attempt at the batch-decode projections (VERDICT r01 weak #3:
+// hipBLASLt reaches only 1.1–2.0 TB/s of weight stream on the llama2-7b
+// B=512 decode shapes, ~26% of the step; v1 decode_gemm.hip and v2
+// wstream_gemm.hip are measured negative results kept for the record).
+//
+// Structure (guide §5 "glds vs register staging" table, top-tier row):
+//   * 256×256 tile, BK=64, 8 waves as 2(M)×4(N), per-wave 128×64 output
+//     (acc[8][4] float4v = 128 VGPR).
+//   * BOTH operand tiles staged by global_load_lds (lane-linear LDS image,
+//     2 LDS buffers = 128 KiB), one vmcnt(0)+barrier per K-step — measured
+//     1160–1220 TF on this structure class at 4096³ vs ~900 for the
+//     128²-tile 2-barrier ladder.
+//   * XOR swizzle ON THE GLOBAL SOURCE ADDRESS (rule 21: glds writes
+//     lane-linear, so the source permutation and the ds_read XOR are the
+//     same involution): chunk ^= row&7 within each 128-B row — the 16-lane
+//     ds_read_b128 fragment read is ≤2-way instead of 8-way.
+//   * W stream is read-once per block → nt (aux=2) glds; X is re-read by
+//     every N tile → default policy (guide nt-weights row).
+//   * W is consumed directly in torch's [N, K] layout: a B-fragment is 8
+//     consecutive k of one W row = one contiguous uint4 — no repack, no
+//     transpose (v2's offline repack is unnecessary at BN=256).
+//   * split-K (fp32 partials + reduce kernel) only to fill the chip when
+//     (M/256)·(N/256) < ~256 blocks; bijective XCD-aware tile remap.
+//
+// M is padded by clamping row indices (clamped rows compute garbage into
+// accumulators whose C rows are never stored — the swizzle stays
+// consistent because read and write both key on the LOCAL row index).
+//
+// STATUS (round 3 conclusion, profiles/mgemm_r02.md): correct (parity vs
+// fp32 matmul on all five llama2-7b decode shapes at M=256/512) and at its
+// structure's measured ceiling (~1.1 PF normalized for grid fill), 1.5-2.3x
+// v2 — but hipBLASLt's MT256/stream-k solutions run 1.3-4.1 TB/s on the
+// same shapes, i.e. at the ~1.33 PF ceiling of the 8-phase counted-vmcnt
+// schedule itself. hipBLASLt stays the production path; SUB_MGEMM=1 routes
+// through this kernel for measurement.
+#include "common.h"
+
+typedef __attribute__((ext_vector_type(8))) short short8;
+typedef __attribute__((ext_vector_type(4))) float float4v;
+
+#define MG_WAVES 8
+#define MG_BM 256
+#define MG_BN 256
+#define MG_BK 64
+
+namespace mfmagemm {
+
+// one glds instruction: wave-uniform LDS base, per-lane global src, 16 B
+DEV_INLINE void glds16(const void* g, void* lds, bool nt) {
+  auto gp = (const __attribute__((address_space(1))) unsigned int*)g;
+  auto lp = (__attribute__((address_space(3))) unsigned int*)lds;
+  if (nt)
+    __builtin_amdgcn_global_load_lds(gp, lp, 16, 0, 2);
+  else
+    __builtin_amdgcn_global_load_lds(gp, lp, 16, 0, 0);
+}
+
+// Templated tile geometry: (256,256, 2x4 waves) fills big-N shapes at
+// 1 block/CU; (128,128, 2x2 waves) runs 2 blocks/CU and quadruples the
+// block count for the small-N projections (o/down at M=512 give only 32
+// 256-tiles on a 256-CU chip — underfill, not schedule, was the v3
+// first-cut gap; split-K is NOT the answer at M=512 because fp32 partials
+// cost splitk x M x N x 8 bytes of extra HBM traffic, comparable to the
+// weight stream itself).
+template <int BM, int BN, int WM, int WN>
+__global__ __launch_bounds__(WM * WN * WAVE)
+void mfma_gemm_kernel(bf16_t* __restrict__ y,      // [M, N]  (splitk == 1)
+                      float* __restrict__ part,    // [splitk, M, N]
+                      const bf16_t* __restrict__ x,   // [M, K]
+                      const bf16_t* __restrict__ w,   // [N, K]
+                      const bf16_t* __restrict__ bias,  // [N] or null
+                      int m, int n, int k, int splitk) {
+  constexpr int NWAVES = WM * WN;
+  constexpr int FM = BM / WM / 16;    // frag rows per wave
+  constexpr int FN = BN / WN / 16;    // frag cols per wave
+  constexpr int APIECES = BM * MG_BK * 2 / 1024;   // 1-KiB glds pieces
+  constexpr int BPIECES = BN * MG_BK * 2 / 1024;
+  // tiles[buf]: A image then B image, each [rows][8 swizzled 16-B chunks]
+  __shared__ uint4 tiles[2][(BM + BN) * MG_BK * 2 / 16];
+
+  const int ntil = n / BN;
+  const int nwg = gridDim.x;
+  // bijective XCD remap (guide T1): consecutive remapped ids land on one
+  // XCD so neighbor tiles (same M rows) share the XCD's L2 copy of X
+  int wgid;
+  {
+    const int orig = blockIdx.x;
+    const int q = nwg / 8, r = nwg % 8, xcd = orig % 8;
+    wgid = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + orig / 8;
+  }
+  const int mt = wgid / ntil, nt = wgid % ntil;
+  const int m0 = mt * BM, n0 = nt * BN;
+  const int slice = blockIdx.y;
+
+  const int wave = threadIdx.x / WAVE;
+  const int lane = threadIdx.x % WAVE;
+  const int lhi = lane >> 4, llo = lane & 15;
+  const int wm = wave / WN, wn = wave % WN;
+
+  // this slice's K-chunk range
+  const int nchunks = k / MG_BK;
+  const int cpk = nchunks / splitk, rem = nchunks % splitk;
+  const int t0 = slice * cpk + min(slice, rem);
+  const int t1 = t0 + cpk + (slice < rem ? 1 : 0);
+
+  float4v acc[FM][FN];
+  #pragma unroll
+  for (int i = 0; i < FM; ++i)
+    #pragma unroll
+    for (int j = 0; j < FN; ++j) acc[i][j] = (float4v){0.f, 0.f, 0.f, 0.f};
+
+  // ---- glds staging: 32 KiB per operand per tile = 4 pieces of 1 KiB per
+  // wave per operand; lane's source row/chunk derive from its flat uint4
+  // index, with the chunk XOR pre-applied on the source side
+  auto stage = [&](int buf, int t) {
+    const int kbase = t * MG_BK;
+    #pragma unroll
+    for (int r = 0; r < APIECES / NWAVES; ++r) {
+      const int piece = wave * (APIECES / NWAVES) + r;
+      const int idx = piece * 64 + lane;        // uint4 index
+      const int row = idx / 8, chunk = idx % 8;
+      const int kc = kbase + ((chunk ^ (row & 7)) * 8);
+      // X (op 0): row clamped to the real M range
+      const int xrow = min(m0 + row, m - 1);
+      glds16(x + (long)xrow * k + kc,
+             (char*)&tiles[buf][0] + piece * 1024, false);
+    }
+    #pragma unroll
+    for (int r = 0; r < BPIECES / NWAVES; ++r) {
+      const int piece = wave * (BPIECES / NWAVES) + r;
+      const int idx = piece * 64 + lane;
+      const int row = idx / 8, chunk = idx % 8;
+      const int kc = kbase + ((chunk ^ (row & 7)) * 8);
+      // W (op 1): read-once stream -> nt
+      glds16(w + (long)(n0 + row) * k + kc,
+             (char*)&tiles[buf][BM * MG_BK * 2 / 16] + piece * 1024, true);
+    }
+  };
+
+  stage(0, t0);
+  __syncthreads();   // carries the vmcnt(0) drain of the in-flight glds
+
+  for (int t = t0; t < t1; ++t) {
+    const int cur = (t - t0) & 1;
+    if (t + 1 < t1) stage(cur ^ 1, t + 1);
+
+    // B-fragments for the wave's n-tiles × 2 k-steps (held whole tile)
+    const uint4* btile = &tiles[cur][BM * MG_BK * 2 / 16];
+    short8 bf[FN][2];
+    #pragma unroll
+    for (int nf = 0; nf < FN; ++nf) {
+      const int row = wn * (BN / WN) + nf * 16 + llo;
+      #pragma unroll
+      for (int ks = 0; ks < 2; ++ks) {
+        const int chunk = ks * 4 + lhi;
+        bf[nf][ks] = *reinterpret_cast<const short8*>(
+            &btile[row * 8 + (chunk ^ (row & 7))]);
+      }
+    }
+    #pragma unroll
+    for (int fm = 0; fm < FM; ++fm) {
+      const int row = wm * (BM / WM) + fm * 16 + llo;
+      const short8 af0 = *reinterpret_cast<const short8*>(
+          &tiles[cur][row * 8 + ((0 + lhi) ^ (row & 7))]);
+      const short8 af1 = *reinterpret_cast<const short8*>(
+          &tiles[cur][row * 8 + ((4 + lhi) ^ (row & 7))]);
+      #pragma unroll
+      for (int nf = 0; nf < FN; ++nf) {
+        acc[fm][nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            af0, bf[nf][0], acc[fm][nf], 0, 0, 0);
+        acc[fm][nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            af1, bf[nf][1], acc[fm][nf], 0, 0, 0);
+      }
+    }
+    __syncthreads();  // drains the next tile's glds; releases cur for reuse
+  }
+
+  // ---- epilogue ---------------------------------------------------------
+  #pragma unroll
+  for (int fm = 0; fm < FM; ++fm) {
+    #pragma unroll
+    for (int nf = 0; nf < FN; ++nf) {
+      const int col = n0 + wn * (BN / WN) + nf * 16 + llo;
+      const float b = (bias != nullptr && splitk == 1)
+                          ? bf2f(bias[col]) : 0.f;
+      #pragma unroll
+      for (int i = 0; i < 4; ++i) {
+        const int row = m0 + wm * (BM / WM) + fm * 16 + lhi * 4 + i;
+        if (row >= m) continue;
+        if (splitk == 1) {
+          y[(long)row * n + col] = f2bf(acc[fm][nf][i] + b);
+        } else {
+          part[(long)slice * m * n + (long)row * n + col] = acc[fm][nf][i];
+        }
+      }
+    }
+  }
+}
+
+// splitk reduce: y = bf16(sum_s part[s] (+ bias))
+__global__ void mfma_gemm_reduce_kernel(bf16_t* __restrict__ y,
+                                        const float* __restrict__ part,
+                                        const bf16_t* __restrict__ bias,
+                                        long mn, int n, int splitk) {
+  const long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  const long stride = (long)gridDim.x * blockDim.x;
+  for (long e = i; e < mn / 2; e += stride) {
+    float2 s = {0.f, 0.f};
+    for (int sl = 0; sl < splitk; ++sl) {
+      const float2 p = *reinterpret_cast<const float2*>(
+          &part[sl * mn + e * 2]);
+      s.x += p.x; s.y += p.y;
+    }
+    if (bias != nullptr) {
+      const long col = (e * 2) % n;
+      s.x += bf2f(bias[col]);
+      s.y += bf2f(bias[col + 1]);
+    }
+    reinterpret_cast<unsigned*>(y)[e] = pack_bf16x2(s.x, s.y);
+  }
+}
+
+}  // namespace mfmagemm
+
+bool mfma_gemm_supported(int m, int n, int k) {
+  return n % 128 == 0 && k % MG_BK == 0 && m >= 1;
+}
+
+// Tile-config choice: 0 = 256x256 (1 block/CU), 1 = 128x128 (2 blocks/CU).
+// Pick the biggest tile whose grid still fills the chip; env
+// SUB_MGEMM_FORCE={0,1} overrides for A/B measurement.
+static int pick_cfg(int m, int n) {
+  const int t256 = (n % 256 == 0)
+      ? ((m + 255) / 256) * (n / 256) : 0;
+  if (t256 >= 192) return 0;
+  const int t128 = ((m + 127) / 128) * (n / 128);
+  // 128-tiles run 2 blocks/CU -> want >= ~384 for full overlap, but any
+  // fill beats an under-filled 256 grid
+  if (t128 >= t256 * 3) return 1;
+  return 0;
+}
+
+int mfma_gemm_splitk(int m, int n, int k) {
+  const char* e = getenv("SUB_MGEMM_FORCE");
+  const int cfg = e ? atoi(e) : pick_cfg(m, n);
+  const int bs = cfg == 0 ? 256 : 128;
+  if (cfg == 0 && n % 256 != 0) return 1;
+  const int tiles = ((m + bs - 1) / bs) * (n / bs);
+  const int target = cfg == 0 ? 256 : 512;
+  int s = 1;
+  // split K only when even the chosen tiling underfills badly AND the
+  // slices stay long; fp32 partial traffic makes deep splits a net loss
+  while (tiles * (s * 2) <= target && (k / MG_BK) / (s * 2) >= 16 && s < 4)
+    s *= 2;
+  return s;
+}
+
+void launch_mfma_gemm(bf16_t* y, const bf16_t* x, const bf16_t* w,
+                      const bf16_t* bias, float* part, int m, int n, int k,
+                      int splitk, hipStream_t stream) {
+  const char* e = getenv("SUB_MGEMM_FORCE");
+  const int cfg = e ? atoi(e) : pick_cfg(m, n);
+  const int bs = (cfg == 0 && n % 256 == 0) ? 256 : 128;
+  const int tiles = ((m + bs - 1) / bs) * (n / bs);
+  dim3 grid(tiles, splitk);
+  if (bs == 256) {
+    mfmagemm::mfma_gemm_kernel<256, 256, 2, 4><<<grid, dim3(512), 0,
+        stream>>>(y, part, x, w, bias, m, n, k, splitk);
+  } else {
+    mfmagemm::mfma_gemm_kernel<128, 128, 2, 2><<<grid, dim3(256), 0,
+        stream>>>(y, part, x, w, bias, m, n, k, splitk);
+  }
+  if (splitk > 1) {
+    const long mn = (long)m * n;
+    const int blocks = (int)min((mn / 2 + 255) / 256, (long)2048);
+    mfmagemm::mfma_gemm_reduce_kernel<<<blocks, 256, 0, stream>>>(
+        y, part, bias, mn, n, splitk);
+  }
+}

@@ -22,6 +22,7 @@ import torch.nn as nn
 import torch.nn.functional as F
 
 from .comm import state, tp_all_reduce_
+from .. import ops
 
 
 def _shard(n: int, world: int, rank: int) -> tuple[int, int]:
@@ -43,7 +44,10 @@ class Linear(nn.Module):
                                              dtype=dtype)) if bias else None
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
-        return F.linear(x, self.weight, self.bias)
+        # ops.linear routes decode-shaped inference GEMMs to the
+        # first-party 256x256 glds kernel (mfma_gemm.hip), everything
+        # else to hipBLASLt
+        return ops.linear(x, self.weight, self.bias)
 
 
 class ColumnParallelLinear(Linear):
@@ -76,7 +80,7 @@ class RowParallelLinear(Linear):
         super().__init__(per, out_features, bias=bias, device=device, dtype=dtype)
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
-        y = F.linear(x, self.weight, None)
+        y = ops.linear(x, self.weight, None)
         tp_all_reduce_(y)
         if self.bias is not None:
             y = y + self.bias

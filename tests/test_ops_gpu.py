@@ -683,3 +683,26 @@ def test_train_step_uses_flash_not_sdpa():
     assert "attn_fwd" not in names.lower() or "attn_prefill" in names, names
     for bad in ("aotriton", "triton", "flash_fwd", "cutlass"):
         assert bad not in names.lower(), f"{bad} kernel on train path:\n{names}"
+
+
+@pytest.mark.parametrize("m,n,k,force", [
+    (512, 1536, 4096, None),    # 256-tile path
+    (200, 1024, 2048, "1"),     # 128-tile path, ragged M
+])
+def test_mfma_gemm_parity(m, n, k, force, monkeypatch):
+    """Decode-projection GEMM v3 vs fp32 matmul (+bias) — kept correct even
+    though hipBLASLt remains the production path (profiles/mgemm_r02.md)."""
+    from substratus_amd import _C
+    if force:
+        monkeypatch.setenv("SUB_MGEMM_FORCE", force)
+    torch.manual_seed(33)
+    x = (torch.rand(m, k, device=DEV) * 2 - 1).bfloat16()
+    w = ((torch.rand(n, k, device=DEV) * 2 - 1) * 0.02).bfloat16()
+    bias = torch.randn(n, device=DEV).bfloat16()
+    y = torch.empty(m, n, device=DEV).bfloat16()
+    splitk = int(_C.mfma_gemm_pick_splitk(m, n, k))
+    part = (torch.empty(splitk * m * n, dtype=torch.float32, device=DEV)
+            if splitk > 1 else None)
+    _C.mfma_gemm(y, x, w, bias, part, splitk)
+    ref = x.float() @ w.float().t() + bias.float()
+    assert_close_bf16(y, ref, atol=2e-2, rtol=2e-2)
