@@ -120,12 +120,26 @@ def bench_serve(args) -> None:
         eng.add_request(prompt, max_new_tokens=args.warmup + args.steps + 32,
                         temperature=0.0)
 
-    # prefill all requests (untimed), then warmup decode steps
+    # prefill all requests (untimed region, but measured: prefill
+    # throughput + TTFT — time to the first sampled token of the first
+    # admitted chunk — go into config for the multi-context table)
+    if on_gpu:
+        torch.cuda.synchronize()
     t_pref = time.time()
+    t_first = None
     while eng.waiting:
-        eng.step()
+        outs = eng.step()
+        if outs and t_first is None:
+            if on_gpu:
+                torch.cuda.synchronize()
+            t_first = time.time() - t_pref
+    if on_gpu:
+        torch.cuda.synchronize()
+    prefill_s = time.time() - t_pref
+    prefill_tok_s = args.batch * args.prompt_len / prefill_s
     log(f"[bench] prefill of {args.batch}x{args.prompt_len} tokens: "
-        f"{time.time()-t_pref:.2f}s")
+        f"{prefill_s:.2f}s ({prefill_tok_s/1e3:.1f}k tok/s, "
+        f"ttft {t_first*1e3:.0f} ms)")
     assert len(eng.running) == args.batch
     for _ in range(args.warmup):
         eng.decode_only_step()
@@ -176,6 +190,8 @@ def bench_serve(args) -> None:
             "prompt_len": args.prompt_len,
             "parallelism": parallelism,
             "p50_decode_latency_ms": round(ms_per_step, 3),
+            "prefill_tokens_per_s": round(prefill_tok_s, 1),
+            "ttft_ms": round((t_first or 0.0) * 1000, 1),
             **({"weights": args.quant} if args.quant else {}),
             **({"kv_cache": "fp8_e4m3"} if args.kv_dtype == "fp8" else {}),
         },
