@@ -158,9 +158,16 @@ def serve(directory: str = typer.Argument(".")):
 
 
 @app.command()
-def notebook(directory: str = typer.Argument(".")):
-    """Convert the directory's manifest to a Notebook and open it
-    (reference internal/cli/notebook.go:16-107)."""
+def notebook(directory: str = typer.Argument("."),
+             sync: bool = typer.Option(False, "--sync",
+                                       help="two-way file sync with the "
+                                            "notebook until Ctrl-C"),
+             port: int = typer.Option(0, "--port",
+                                      help="local port-forward port "
+                                           "(0 = direct endpoint)")):
+    """Convert the directory's manifest to a Notebook, open it, and
+    optionally sync files both ways (reference internal/cli/notebook.go +
+    internal/client/sync.go: nbwatch events -> copy both directions)."""
     c = _client()
     manifests = _find_manifests(
         directory, ("Model", "Dataset", "Server", "Notebook"))
@@ -176,8 +183,31 @@ def notebook(directory: str = typer.Argument(".")):
     pod = c.get("Pod", f"{name}-notebook", ns) or {}
     ep = pod.get("metadata", {}).get("annotations", {}).get(
         "substratus.ai/local-endpoint", "http://localhost:8888")
+    forwarder = None
+    if port:
+        from urllib.parse import urlparse
+        from .port_forward import PortForwarder
+        u = urlparse(ep)
+        forwarder = PortForwarder(port, target_host=u.hostname or "127.0.0.1",
+                                  target_port=u.port or 8888)
+        forwarder.start()
+        ep = f"http://localhost:{port}"
     console.print(f"notebook [bold]{name}[/bold] ready at "
                   f"{ep}/?token=default")
+    if sync:
+        from .sync import NotebookSync
+        s = NotebookSync(directory, ep, token="default")
+        n = s.push_all()
+        console.print(f"synced {n} files up; watching both ways "
+                      f"(Ctrl-C to stop)")
+        try:
+            s.run()
+        except KeyboardInterrupt:
+            pass
+        finally:
+            s.stop()
+            if forwarder:
+                forwarder.stop()
 
 
 app.command(name="nb")(notebook)

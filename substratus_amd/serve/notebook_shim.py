@@ -4,17 +4,23 @@ The production notebook image runs `notebook.sh` → JupyterLab on 8888
 (images/Dockerfile.notebook; container contract: readiness GET /api,
 NOTEBOOK_TOKEN honored). This offline environment has no jupyterlab, so the
 LocalRuntime serves the SAME contract surface with a minimal FastAPI app:
-GET /api → version info (the readiness probe), GET/PUT /files/* over the
-mounted /content tree (what the CLI's sync needs), token checked like
-Jupyter's `?token=`.
+GET /api → version info (the readiness probe), GET/PUT/DELETE /files/* over
+the mounted /content tree, and GET /events — the nbwatch change stream
+(containertools/nbwatch.py runs INSIDE the shim, the analog of the
+reference shipping the nbwatch binary into the pod and exec-ing it,
+internal/client/sync.go:28-135) that the CLI's two-way sync consumes.
+Token checked like Jupyter's `?token=`.
 """
 
 from __future__ import annotations
 
 import os
+import threading
 
 from fastapi import FastAPI, HTTPException, Request
 from fastapi.responses import JSONResponse, PlainTextResponse
+
+from ..containertools.nbwatch import watched_files
 
 
 def create_app(content_dir: str = "/content",
@@ -33,6 +39,50 @@ def create_app(content_dir: str = "/content",
     @app.get("/")
     def root():
         return {"status": "ok", "content": content_dir}
+
+    # ---- nbwatch event stream (consumed by cli/sync.py) ----------------
+    events: list[dict] = []
+    ev_lock = threading.Lock()
+    state = {"prev": None, "stop": False}
+
+    def _scan_once() -> None:
+        cur = watched_files(content_dir)
+        prev = state["prev"]
+        state["prev"] = cur
+        if prev is None:
+            return
+        with ev_lock:
+            idx = len(events)
+            for path, mtime in cur.items():
+                rel = os.path.relpath(path, content_dir)
+                if path not in prev:
+                    events.append({"index": idx, "path": rel,
+                                   "op": "CREATE"}); idx += 1
+                elif mtime != prev[path]:
+                    events.append({"index": idx, "path": rel,
+                                   "op": "WRITE"}); idx += 1
+            for path in prev:
+                if path not in cur:
+                    events.append({"index": idx,
+                                   "path": os.path.relpath(path, content_dir),
+                                   "op": "REMOVE"}); idx += 1
+
+    def _scanner():
+        import time
+        while not state["stop"]:
+            _scan_once()
+            time.sleep(0.3)
+
+    threading.Thread(target=_scanner, daemon=True).start()
+    app.state.stop_scanner = lambda: state.update(stop=True)
+
+    @app.get("/events")
+    def get_events(req: Request, since: int = 0):
+        check_token(req)
+        _scan_once()   # make tests deterministic (no poll-interval wait)
+        with ev_lock:
+            return {"events": [e for e in events if e["index"] >= since],
+                    "next": len(events)}
 
     @app.get("/files")
     def list_files(req: Request):
@@ -65,6 +115,14 @@ def create_app(content_dir: str = "/content",
         with open(full, "wb") as f:
             f.write(body)
         return JSONResponse({"written": path})
+
+    @app.delete("/files/{path:path}")
+    def delete_file(path: str, req: Request):
+        check_token(req)
+        full = os.path.join(content_dir, path)
+        if os.path.isfile(full):
+            os.remove(full)
+        return JSONResponse({"deleted": path})
 
     return app
 
