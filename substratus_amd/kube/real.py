@@ -132,21 +132,65 @@ class RealKube:
         return [i.to_dict() for i in res.get(
             namespace=namespace, label_selector=sel).items]
 
-    def watch(self, kind: Optional[str], cb: Callable) -> None:
+    # -- watch -------------------------------------------------------------
+    def _list_func(self, kind: str, namespace: str):
+        """The list callable + positional args the watch stream wraps."""
+        api, plural = self._route(kind)
+        if api == "crd":
+            return (self._dyn.list_namespaced_custom_object,
+                    ("substratus.ai", "v1", namespace, plural))
+        if api == "core":
+            fn = {"Pod": self._core.list_namespaced_pod,
+                  "Service": self._core.list_namespaced_service,
+                  "ConfigMap": self._core.list_namespaced_config_map,
+                  "ServiceAccount":
+                      self._core.list_namespaced_service_account,
+                  "Secret": self._core.list_namespaced_secret}[kind]
+            return (fn, (namespace,))
+        if api == "batch":
+            return (self._batch.list_namespaced_job, (namespace,))
+        return (self._apps.list_namespaced_deployment, (namespace,))
+
+    def watch(self, kind: str, cb: Callable,
+              namespace: str = "default") -> Callable[[], None]:
+        """Watch EVERY routable kind (CRDs, Pods, Jobs, Deployments, core) —
+        the dependent-requeue signal the controllers need (reference
+        internal/client/client.go:137-152). Reconnects with backoff,
+        resuming from the last resourceVersion; a 410 Gone (expired RV)
+        falls back to a fresh list. Returns a stop() callable."""
         import threading
+        stop = threading.Event()
+        fn, args = self._list_func(kind, namespace)  # route errors raise NOW
 
         def run():
-            api, plural = self._route(kind)
-            w = self._watch.Watch()
-            if api == "crd":
-                stream = w.stream(self._dyn.list_namespaced_custom_object,
-                                  "substratus.ai", "v1", "default", plural)
-            else:
-                return
-            for event in stream:
-                self.changes += 1
-                cb(event["type"], event["object"])
+            backoff = 1.0
+            rv = None
+            while not stop.is_set():
+                try:
+                    w = self._watch.Watch()
+                    kwargs = {"timeout_seconds": 60}
+                    if rv:
+                        kwargs["resource_version"] = rv
+                    for event in w.stream(fn, *args, **kwargs):
+                        if stop.is_set():
+                            return
+                        obj = event["object"]
+                        if not isinstance(obj, dict):  # typed model -> dict
+                            obj = self._client.ApiClient()\
+                                .sanitize_for_serialization(obj)
+                        rv = (obj.get("metadata") or {})\
+                            .get("resourceVersion") or rv
+                        self.changes += 1
+                        cb(event["type"], obj)
+                    backoff = 1.0  # clean stream timeout: reconnect at once
+                except Exception as e:  # noqa: BLE001
+                    if "410" in str(e) or "Expired" in str(e):
+                        rv = None  # relist from scratch
+                    stop.wait(backoff)
+                    backoff = min(backoff * 2, 30.0)
+
         threading.Thread(target=run, daemon=True).start()
+        return stop.set
 
 
 def _api_version(kind: str) -> str:
