@@ -190,10 +190,19 @@ def test_sample_greedy_matches_argmax():
 
 
 def test_sample_temperature_cdf_matches_ref():
+    """Seeded comparison against the fp32 CPU reference (VERDICT r01 weak
+    #6: the old 75%-of-8-rows floor couldn't catch a broken CDF scan).
+    256 rows; >=95% must bit-match the reference token, and EVERY
+    disagreement must be a genuine CDF-boundary case: the kernel's chosen
+    token's fp32 CDF interval contains the draw within EPS=1e-2 (the
+    kernel's parallel f32 scan over bf16-staged probabilities reorders
+    additions vs the reference's serial cumsum; 1e-2 bounds that
+    difference with wide margin — a mis-scanned CDF lands tokens far
+    outside it)."""
     from substratus_amd import ops
     from substratus_amd.ops import ref
     torch.manual_seed(10)
-    bsz, vocab = 8, 32000
+    bsz, vocab = 256, 32000
     logits = torch.randn(bsz, vocab, device=DEV).bfloat16() * 3
     temp = torch.full((bsz,), 0.8, device=DEV)
     tk = torch.zeros(bsz, dtype=torch.int32, device=DEV)
@@ -202,13 +211,20 @@ def test_sample_temperature_cdf_matches_ref():
     got = ops.softmax_sample(logits, temp, tk, tp, u)
     want = ref.softmax_sample(logits.cpu(), temp.cpu(), tk.cpu(), tp.cpu(),
                               u.cpu())
-    # bf16 LDS staging may flip the choice at CDF boundaries; tokens must
-    # agree for nearly all rows and all must have non-trivial probability
     agree = (got.cpu() == want).float().mean().item()
-    assert agree >= 0.75, f"only {agree:.2f} of sampled tokens agree with ref"
-    probs = torch.softmax(logits.float() / 0.8, dim=-1)
-    drawn = probs.gather(1, got[:, None]).squeeze(1)
-    assert (drawn > 1e-8).all()
+    assert agree >= 0.95, f"only {agree:.3f} of sampled tokens agree"
+    EPS = 1e-2
+    probs = torch.softmax(logits.float().cpu() / 0.8, dim=-1)
+    cdf = probs.cumsum(dim=-1)
+    gc = got.cpu()
+    for b in torch.nonzero(gc != want).flatten().tolist():
+        tok = int(gc[b])
+        hi = float(cdf[b, tok])
+        lo = hi - float(probs[b, tok])
+        ub = float(u[b].cpu())
+        assert lo - EPS <= ub <= hi + EPS, (
+            f"row {b}: kernel token {tok} CDF [{lo:.4f},{hi:.4f}] vs draw "
+            f"{ub:.4f} — not a boundary case (ref token {int(want[b])})")
 
 
 def test_sample_topk_membership():
