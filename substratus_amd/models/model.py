@@ -130,6 +130,51 @@ class Block(nn.Module):
         return self.w_down(ops.gelu(self.w_up(y)))
 
     # -- decode -------------------------------------------------------------
+    def decode_ops(self, y: torch.Tensor, h: torch.Tensor,
+                   positions: torch.Tensor,
+                   k_cache: torch.Tensor, v_cache: torch.Tensor,
+                   slots: torch.Tensor, block_table: torch.Tensor,
+                   seq_lens: torch.Tensor):
+        """Generator form of the decode block: yields after each op group
+        (qkv GEMM | rope+attention | o GEMM | up GEMM | down GEMM) so the
+        dual-stream runner (serve/graph_runner.py) can interleave two
+        half-batches op-by-op — the trailing half's GEMMs re-read weights
+        the leading half just streamed (LLC-hot) while its attention
+        overlaps the leader's compute. forward_decode drives this
+        generator to completion, so single-stream numerics are by
+        construction identical."""
+        qkv = self.wqkv(y)
+        yield
+        q, k, v = self._split_qkv(qkv)
+        q = ops.rope_kv_append(q, k, v, positions, k_cache, v_cache, slots,
+                               self.rope_theta)
+        a = ops.paged_attn_decode(q, k_cache, v_cache, block_table, seq_lens,
+                                  self.scale)
+        yield
+        attn_out = self.wo(a.reshape(a.shape[0], -1))
+        yield
+        if self.parallel_attn:
+            mlp_in = self.norm2(h, self.eps) if self.parallel_norms else y
+            if self.cfg.activation == "silu":
+                gu = self.w_gateup(mlp_in)
+                yield
+                out = attn_out + self.w_down(ops.silu_mul(gu))
+            else:
+                u = self.w_up(mlp_in)
+                yield
+                out = attn_out + self.w_down(ops.gelu(u))
+            return out, h
+        y2, h = fused_norm_residual(attn_out, h, self.norm2, self.eps)
+        if self.cfg.activation == "silu":
+            gu = self.w_gateup(y2)
+            yield
+            out = self.w_down(ops.silu_mul(gu))
+        else:
+            u = self.w_up(y2)
+            yield
+            out = self.w_down(ops.gelu(u))
+        return out, h
+
     def forward_decode(self, y: torch.Tensor, h: torch.Tensor,
                        positions: torch.Tensor,
                        k_cache: torch.Tensor, v_cache: torch.Tensor,
@@ -139,17 +184,13 @@ class Block(nn.Module):
         caller's fused residual+norm); h: the residual stream. Returns
         (delta, h) where delta is this block's output to be added by the
         NEXT norm's fused residual."""
-        q, k, v = self._split_qkv(self.wqkv(y))
-        q = ops.rope_kv_append(q, k, v, positions, k_cache, v_cache, slots,
-                               self.rope_theta)
-        a = ops.paged_attn_decode(q, k_cache, v_cache, block_table, seq_lens,
-                                  self.scale)
-        attn_out = self.wo(a.reshape(a.shape[0], -1))
-        if self.parallel_attn:
-            mlp_in = self.norm2(h, self.eps) if self.parallel_norms else y
-            return attn_out + self._mlp(mlp_in), h
-        y2, h = fused_norm_residual(attn_out, h, self.norm2, self.eps)
-        return self._mlp(y2), h
+        gen = self.decode_ops(y, h, positions, k_cache, v_cache, slots,
+                              block_table, seq_lens)
+        while True:
+            try:
+                next(gen)
+            except StopIteration as e:
+                return e.value
 
     # -- prefill ------------------------------------------------------------
     def forward_prefill(self, h: torch.Tensor, positions: torch.Tensor,
@@ -299,23 +340,40 @@ class CausalLM(nn.Module):
                 p.fill_(1.0)
 
     # -- decode --------------------------------------------------------------
-    def forward_decode(self, tokens: torch.Tensor, positions: torch.Tensor,
-                       kv_caches: list[tuple[torch.Tensor, torch.Tensor]],
-                       slots: torch.Tensor, block_table: torch.Tensor,
-                       seq_lens: torch.Tensor) -> torch.Tensor:
+    def decode_ops(self, tokens: torch.Tensor, positions: torch.Tensor,
+                   kv_caches: list[tuple[torch.Tensor, torch.Tensor]],
+                   slots: torch.Tensor, block_table: torch.Tensor,
+                   seq_lens: torch.Tensor):
+        """Whole-model decode as an op-group generator (see
+        TransformerBlock.decode_ops); returns the logits."""
         h = self.embed_tokens(tokens)
         if self.cfg.learned_pos_emb:
             h = h + self.embed_positions(positions.long())
         # fused residual+norm chain: each block's output delta is added to
         # the residual stream inside the NEXT norm kernel
         y = self.blocks[0].norm1(h, self.eps)
+        yield
         for i, (blk, (kc, vc)) in enumerate(zip(self.blocks, kv_caches)):
-            delta, h = blk.forward_decode(y, h, positions, kc, vc, slots,
-                                          block_table, seq_lens)
+            delta, h = yield from blk.decode_ops(y, h, positions, kc, vc,
+                                                 slots, block_table,
+                                                 seq_lens)
             nxt = (self.blocks[i + 1].norm1 if i + 1 < len(self.blocks)
                    else self.final_norm)
             y, h = fused_norm_residual(delta, h, nxt, self.eps)
+            yield
         return self._logits(y)
+
+    def forward_decode(self, tokens: torch.Tensor, positions: torch.Tensor,
+                       kv_caches: list[tuple[torch.Tensor, torch.Tensor]],
+                       slots: torch.Tensor, block_table: torch.Tensor,
+                       seq_lens: torch.Tensor) -> torch.Tensor:
+        gen = self.decode_ops(tokens, positions, kv_caches, slots,
+                              block_table, seq_lens)
+        while True:
+            try:
+                next(gen)
+            except StopIteration as e:
+                return e.value
 
     # -- prefill -------------------------------------------------------------
     def forward_prefill(self, tokens: torch.Tensor, positions: torch.Tensor,

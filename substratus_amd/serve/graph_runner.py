@@ -28,6 +28,8 @@ from typing import Optional
 
 import torch
 
+import os
+
 from .kv_cache import SequenceState
 
 BUCKETS = (1, 2, 4, 8, 16, 32, 64, 96, 128, 192, 256)
@@ -60,6 +62,19 @@ class GraphDecodeRunner:
         self.top_k = torch.zeros(B, dtype=torch.int32, device=dev)
         self.top_p = torch.ones(B, dtype=torch.float32, device=dev)
         self.graphs: dict[int, torch.cuda.CUDAGraph] = {}
+        # Dual-stream decode is OPT-IN (SUB_SPLIT_STREAMS=1): both the
+        # free-running split (-5.5%) and the op-skewed interleave (-10%)
+        # measured SLOWER than single-stream at B=512 — the KV stream
+        # (multi-GB per step) flushes the 256 MiB LLC between the halves'
+        # weight reads, so split GEMMs pay double HBM weight traffic, and
+        # the halves' attention kernels contend for the same HBM the
+        # single-stream schedule already saturates. Kept (with the
+        # op-generator plumbing in models/model.py decode_ops) as the
+        # harness for future overlap experiments.
+        self.split_streams = os.environ.get("SUB_SPLIT_STREAMS",
+                                            "0") == "1"
+        self.side = torch.cuda.Stream() if engine.device.type == "cuda" \
+            else None
         self.active: list[SequenceState] = []
         self.host_lens: list[int] = []
         self.primed_bucket = 0
@@ -98,24 +113,75 @@ class GraphDecodeRunner:
         self.primed_bucket = b
 
     # ------------------------------------------------------------- capture
+    def _decode_gen(self, lo: int, hi: int):
+        sl = slice(lo, hi)
+        return self.model.decode_ops(
+            self.tokens[sl], self.positions[sl], self.eng.kv.layers,
+            self.slots[sl], self.block_table[sl], self.seq_lens[sl])
+
     def _step_body(self, b: int) -> None:
+        # Op-skewed dual-stream decode: split the batch in half and
+        # interleave the two forwards op-group by op-group on two HIP
+        # streams, the trailer waiting on the leader's matching op event.
+        # The skew makes every overlapped pair COMPLEMENTARY: the
+        # trailer's qkv/up/down GEMM runs right after the leader streamed
+        # those weights (LLC-hot — a free-running split doubles the HBM
+        # weight traffic and measured -5.5%), while the trailer's
+        # HBM-bound paged attention overlaps the leader's MFMA GEMMs.
+        # Halves touch disjoint sequences/KV blocks — no data hazards;
+        # the RNG draw happens once on the main stream before the fork.
         from substratus_amd import _C
-        tokens = self.tokens[:b]
+        uniform = torch.rand(self.max_batch, device=self.device)
+
+        def sample(lo, hi, logits):
+            sl = slice(lo, hi)
+            _C.softmax_sample(self.tokens[sl], logits.contiguous(),
+                              self.temp[sl], self.top_k[sl],
+                              self.top_p[sl], uniform[sl])
+
+        if self.split_streams and b >= 32:
+            h = b // 2
+            cur = torch.cuda.current_stream()
+            self.side.wait_stream(cur)
+            ga, gb = self._decode_gen(0, h), self._decode_gen(h, b)
+            logits_a = logits_b = None
+            while logits_a is None or logits_b is None:
+                ev = None
+                if logits_a is None:
+                    try:
+                        next(ga)
+                    except StopIteration as e:
+                        logits_a = e.value
+                        sample(0, h, logits_a)
+                    ev = torch.cuda.Event()
+                    ev.record(cur)
+                if logits_b is None:
+                    with torch.cuda.stream(self.side):
+                        if ev is not None:
+                            self.side.wait_event(ev)
+                        try:
+                            next(gb)
+                        except StopIteration as e:
+                            logits_b = e.value
+                            sample(h, b, logits_b)
+            cur.wait_stream(self.side)
+        else:
+            gen = self._decode_gen(0, b)
+            while True:
+                try:
+                    next(gen)
+                except StopIteration as e:
+                    sample(0, b, e.value)
+                    break
+        # device-side advance for the next iteration
         pos = self.positions[:b]
         lens = self.seq_lens[:b]
-        slots = self.slots[:b]
         bt = self.block_table[:b]
-        logits = self.model.forward_decode(tokens, pos, self.eng.kv.layers,
-                                           slots, bt, lens)
-        uniform = torch.rand(b, device=self.device)
-        _C.softmax_sample(tokens, logits.contiguous(), self.temp[:b],
-                          self.top_k[:b], self.top_p[:b], uniform)
-        # device-side advance for the next iteration
         pos += 1
         lens += 1
         blk_col = torch.div(pos, self.bs, rounding_mode="floor").long()
         blk = bt.gather(1, blk_col.unsqueeze(1)).squeeze(1)
-        slots.copy_(blk * self.bs + pos % self.bs)
+        self.slots[:b].copy_(blk * self.bs + pos % self.bs)
 
     def _get_graph(self, b: int) -> torch.cuda.CUDAGraph:
         if b in self.graphs:
@@ -137,8 +203,19 @@ class GraphDecodeRunner:
         torch.cuda.current_stream().wait_stream(s)
         torch.cuda.synchronize()
         g = torch.cuda.CUDAGraph()
-        with torch.cuda.graph(g):
-            self._step_body(b)
+        # The op-generator step creates collectible cycles (suspended
+        # frames + StopIteration tracebacks); a Python GC firing MID-
+        # capture frees earlier tensors into foreign mempools (hipFree
+        # during stream capture aborts the process on ROCm). Drain garbage
+        # first, then hold GC off for the capture.
+        import gc
+        gc.collect()
+        gc.disable()
+        try:
+            with torch.cuda.graph(g):
+                self._step_body(b)
+        finally:
+            gc.enable()
         for dst, src in zip((self.tokens, self.positions, self.seq_lens,
                              self.slots), saved):
             dst.copy_(src)
