@@ -29,7 +29,7 @@ from .. import ops
 from ..models.config import get_config
 from ..models.model import CausalLM
 from ..parallel.comm import barrier, state
-from .ddp import GradBuckets
+from .ddp import GradBuckets, ZeroShardedDDP
 
 
 @dataclass
@@ -52,6 +52,10 @@ class TrainConfig:
     seed: int = 0
     dtype: torch.dtype = torch.bfloat16
     grad_clip: float = 1.0
+    # ZeRO-1 sharded optimizer (reduce-scatter grads, shard AdamW,
+    # all-gather params) whenever dp_size > 1; False = plain bucketed
+    # all-reduce DDP (train/ddp.py GradBuckets)
+    zero: bool = True
 
 
 class AdamWState:
@@ -109,8 +113,15 @@ class Trainer:
         params = [p for p in self.model.parameters()]
         for p in params:
             p.requires_grad_(True)
-        self.opt = AdamWState(params)
-        self.buckets = GradBuckets(params)
+        self.use_zero = bool(cfg.zero) and self.st.dp_size > 1
+        if self.use_zero:
+            self.zero = ZeroShardedDDP(params)
+            self.opt = AdamWState([])   # kept for interface; state in zero
+            self.buckets = None
+        else:
+            self.zero = None
+            self.opt = AdamWState(params)
+            self.buckets = GradBuckets(params)
         self.global_step = 0
         self.tokens_seen = 0
 
@@ -124,18 +135,29 @@ class Trainer:
         flat = logits.reshape(-1, logits.shape[-1])
         loss, dlogits = ops.cross_entropy_fwd_bwd(flat.detach(), targets)
         flat.backward(gradient=dlogits)
-        self.buckets.finalize()
-        gscale = 1.0
-        if self.cfg.grad_clip > 0:
-            # norm only; the clip SCALE is folded into the fused AdamW
-            # kernel's gradient read (saves a foreach_mul pass over every
-            # grad — ~10 ms/step at falcon-7b scale, profiles/train)
-            grads = [p.grad for p in self.opt.params if p.grad is not None]
-            norms = torch._foreach_norm(grads, 2.0)
-            total = torch.linalg.vector_norm(torch.stack(norms))
-            gscale = min(1.0, self.cfg.grad_clip / (float(total) + 1e-6))
-        self.opt.step(self.cfg, grad_scale=gscale)
-        self.opt.zero_grad()
+        if self.use_zero:
+            self.zero.finalize()
+            gscale = 1.0
+            if self.cfg.grad_clip > 0:
+                total = self.zero.global_grad_norm()
+                gscale = min(1.0, self.cfg.grad_clip / (total + 1e-6))
+            self.zero.step(self.cfg.lr, self.cfg.beta1, self.cfg.beta2,
+                           self.cfg.eps, self.cfg.weight_decay, gscale)
+            self.zero.zero_grad()
+        else:
+            self.buckets.finalize()
+            gscale = 1.0
+            if self.cfg.grad_clip > 0:
+                # norm only; the clip SCALE is folded into the fused AdamW
+                # kernel's gradient read (saves a foreach_mul pass over
+                # every grad — ~10 ms/step at falcon-7b scale)
+                grads = [p.grad for p in self.opt.params
+                         if p.grad is not None]
+                norms = torch._foreach_norm(grads, 2.0)
+                total = torch.linalg.vector_norm(torch.stack(norms))
+                gscale = min(1.0, self.cfg.grad_clip / (float(total) + 1e-6))
+            self.opt.step(self.cfg, grad_scale=gscale)
+            self.opt.zero_grad()
         self.global_step += 1
         self.tokens_seen += inputs.numel() * max(self.st.dp_size, 1)
         return float(loss)
@@ -144,14 +166,21 @@ class Trainer:
     def save_checkpoint(self, step: Optional[int] = None) -> str:
         step = step if step is not None else self.global_step
         out = os.path.join(self.cfg.artifacts_dir, f"checkpoint-{step}")
+        os.makedirs(out, exist_ok=True)
+        if self.use_zero:
+            # each DP rank owns 1/world of the optimizer state — every
+            # rank persists its own shard (resume requires the same world)
+            torch.save(self.zero.state_dict(),
+                       os.path.join(out,
+                                    f"optimizer-rank{self.st.dp_rank}.pt"))
         if self.st.rank == 0:
-            os.makedirs(out, exist_ok=True)
             from safetensors.torch import save_file
             tensors = {k: v.detach().cpu().contiguous()
                        for k, v in self.model.state_dict().items()}
             save_file(tensors, os.path.join(out, "model.safetensors"))
-            torch.save(self.opt.state_dict(),
-                       os.path.join(out, "optimizer.pt"))
+            if not self.use_zero:
+                torch.save(self.opt.state_dict(),
+                           os.path.join(out, "optimizer.pt"))
             with open(os.path.join(out, "trainer_state.json"), "w") as f:
                 json.dump({"global_step": step,
                            "tokens_seen": self.tokens_seen,
@@ -187,9 +216,15 @@ class Trainer:
         sd = load_file(os.path.join(path, "model.safetensors"))
         self.model.load_state_dict(
             {k: v.to(self.device) for k, v in sd.items()})
-        opt_path = os.path.join(path, "optimizer.pt")
-        if os.path.exists(opt_path):
-            self.opt.load_state_dict(torch.load(opt_path), self.device)
+        if self.use_zero:
+            opt_path = os.path.join(
+                path, f"optimizer-rank{self.st.dp_rank}.pt")
+            if os.path.exists(opt_path):
+                self.zero.load_state_dict(torch.load(opt_path), self.device)
+        else:
+            opt_path = os.path.join(path, "optimizer.pt")
+            if os.path.exists(opt_path):
+                self.opt.load_state_dict(torch.load(opt_path), self.device)
         with open(os.path.join(path, "trainer_state.json")) as f:
             st = json.load(f)
         self.global_step = st["global_step"]
