@@ -229,18 +229,52 @@ def apply(file: str = typer.Option(..., "-f", "--filename")):
 @app.command()
 def get(scope: str = typer.Argument("all"),
         watch: bool = typer.Option(False, "--watch", "-w",
-                                   help="refresh until interrupted")):
+                                   help="refresh until interrupted"),
+        logs: bool = typer.Option(False, "--logs",
+                                  help="with --watch: stream the workload "
+                                       "logs of not-ready objects under "
+                                       "the table")):
     """List objects: `sub get models`, `sub get models/name`, `sub get all`
     (reference internal/cli/get.go; scope grammar tui/get.go:219-287;
-    --watch mirrors the reference's live watch-driven listing)."""
+    --watch mirrors the live watch-driven listing; --watch --logs adds the
+    reference TUI's pod-log viewports, internal/tui/pods.go:222-246)."""
     c = _client()
     if watch:
         from rich.live import Live
-        with Live(_get_table(c, scope), refresh_per_second=2) as live:
+        with Live(_watch_view(c, scope, logs), refresh_per_second=2) as live:
             while True:
                 time.sleep(1)
-                live.update(_get_table(c, scope))
+                live.update(_watch_view(c, scope, logs))
     console.print(_get_table(c, scope))
+
+
+_WORKLOAD_SUFFIX = {"Model": "-modeller", "Dataset": "-data-loader",
+                    "Server": "-server", "Notebook": "-notebook"}
+
+
+def _watch_view(c, scope: str, show_logs: bool):
+    table = _get_table(c, scope)
+    if not show_logs:
+        return table
+    from rich.console import Group
+    from rich.panel import Panel
+    panels = [table]
+    for kind in apiv1.KINDS:
+        for obj in c.list(kind):
+            if apiv1.get_status_ready(obj):
+                continue
+            name = obj["metadata"]["name"]
+            ns = obj["metadata"].get("namespace", "default")
+            wl = name + _WORKLOAD_SUFFIX[kind]
+            try:
+                r = c.http.get(f"/api/logs/{ns}/{wl}", params={"tail": 8})
+                if r.status_code == 200 and r.json()["lines"]:
+                    panels.append(Panel("\n".join(r.json()["lines"]),
+                                        title=f"{kind.lower()}/{name} "
+                                              f"({wl})", height=10))
+            except Exception:
+                pass
+    return Group(*panels)
 
 
 def _get_table(c, scope: str):
@@ -280,16 +314,63 @@ def delete(target: str):
 
 @app.command()
 def logs(name: str, tail: int = typer.Option(100, "--tail"),
-         namespace: str = typer.Option("default", "-n")):
+         namespace: str = typer.Option("default", "-n"),
+         follow: bool = typer.Option(False, "--follow", "-f",
+                                     help="stream new lines until Ctrl-C"),
+         poll: float = typer.Option(0.5, "--poll", hidden=True),
+         max_polls: int = typer.Option(-1, "--max-polls", hidden=True)):
     """Logs of an object's workload: `sub logs myserver-server` /
-    `sub logs mymodel-modeller` (reference: pod log streaming in the TUI)."""
+    `sub logs -f mymodel-modeller` (reference: pod log streaming in the
+    TUI, internal/tui/pods.go:222-246)."""
     c = _client()
     r = c.http.get(f"/api/logs/{namespace}/{name}", params={"tail": tail})
     if r.status_code == 404:
         console.print(f"[red]no logs for {name}[/red]")
         raise typer.Exit(1)
-    for line in r.json()["lines"]:
+    body = r.json()
+    for line in body["lines"]:
         console.print(line, highlight=False)
+    if not follow:
+        return
+    cursor = body.get("next", 0)
+    n = 0
+    try:
+        while max_polls < 0 or n < max_polls:
+            time.sleep(poll)
+            n += 1
+            r = c.http.get(f"/api/logs/{namespace}/{name}",
+                           params={"since": cursor})
+            if r.status_code != 200:
+                continue
+            body = r.json()
+            for line in body["lines"]:
+                console.print(line, highlight=False)
+            cursor = body.get("next", cursor)
+    except KeyboardInterrupt:
+        pass
+
+
+@app.command()
+def suspend(name: str, namespace: str = typer.Option("default", "-n")):
+    """Suspend a Notebook (reaps its pod; the reference TUI's 'q' menu
+    suspend action, internal/tui/notebook.go)."""
+    c = _client()
+    if c.get("Notebook", name, namespace) is None:
+        console.print(f"[red]notebook {name} not found[/red]")
+        raise typer.Exit(1)
+    c.patch("Notebook", name, {"spec": {"suspend": True}}, namespace)
+    console.print(f"notebook [bold]{name}[/bold] suspended")
+
+
+@app.command()
+def resume(name: str, namespace: str = typer.Option("default", "-n")):
+    """Resume a suspended Notebook."""
+    c = _client()
+    if c.get("Notebook", name, namespace) is None:
+        console.print(f"[red]notebook {name} not found[/red]")
+        raise typer.Exit(1)
+    c.patch("Notebook", name, {"spec": {"suspend": False}}, namespace)
+    console.print(f"notebook [bold]{name}[/bold] resumed")
 
 
 @app.command()

@@ -93,9 +93,12 @@ def create_api(kube: MemKube, platform: Optional["Platform"] = None) -> FastAPI:
             raise HTTPException(404, "not found")
 
     @app.get("/api/logs/{namespace}/{name}")
-    def logs(namespace: str, name: str, tail: int = 200):
+    def logs(namespace: str, name: str, tail: int = 200, since: int = -1):
         # pod-log streaming equivalent (the reference TUI streams the
-        # default container's logs — internal/tui/pods.go:222-246)
+        # default container's logs — internal/tui/pods.go:222-246).
+        # since >= 0: return lines from that absolute line index (the
+        # follow cursor `sub logs -f` polls with); response carries
+        # `next`, the cursor for the following poll.
         if platform is None:
             raise HTTPException(404, "no runtime")
         path = platform.runtime.log_path(namespace, name)
@@ -103,7 +106,9 @@ def create_api(kube: MemKube, platform: Optional["Platform"] = None) -> FastAPI:
             raise HTTPException(404, f"no logs for {namespace}/{name}")
         with open(path, "rb") as f:
             lines = f.read().decode(errors="replace").splitlines()
-        return {"lines": lines[-tail:]}
+        if since >= 0:
+            return {"lines": lines[since:], "next": len(lines)}
+        return {"lines": lines[-tail:], "next": len(lines)}
 
     @app.delete("/api/objects/{kind}/{namespace}/{name}")
     def delete_obj(kind: str, namespace: str, name: str):
@@ -160,6 +165,10 @@ class Platform:
 
     def stop(self) -> None:
         self._stop = True
+        # let the reconcile loop observe _stop before tearing the runtime
+        # down under it
+        if getattr(self, "_thread", None) is not None:
+            self._thread.join(timeout=5)
         self.runtime.stop()
         self._sci_server.stop(grace=None)
         self._sci_httpd.shutdown()

@@ -324,10 +324,14 @@ class LocalRuntime:
                                 stdout=lf, stderr=lf)
 
     def stop(self) -> None:
-        for proc in self.procs.values():
+        # snapshot: the platform's reconcile thread may still be spawning/
+        # reaping concurrently (e.g. a Notebook suspend mid-shutdown) —
+        # iterating the live dict races ("dictionary changed size")
+        procs = list(self.procs.values())
+        for proc in procs:
             if proc.poll() is None:
                 proc.terminate()
-        for proc in self.procs.values():
+        for proc in procs:
             try:
                 proc.wait(timeout=10)
             except subprocess.TimeoutExpired:

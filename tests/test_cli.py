@@ -193,3 +193,47 @@ def test_run_replace_flag(daemon, tmp_path):
     obj = platform.kube.get("Dataset", "cli-replace")
     assert obj is not None  # same name, recreated/replaced
     assert platform.kube.try_get("Dataset", "cli-replace-2") is None
+
+
+def test_logs_follow_cursor(daemon):
+    """`sub logs -f`: the since-cursor API returns only new lines; the CLI
+    prints what arrives between polls (reference TUI log streaming)."""
+    platform, url = daemon
+    path = platform.runtime.log_path("default", "follow-me")
+    os.makedirs(os.path.dirname(path), exist_ok=True)
+    with open(path, "w") as f:
+        f.write("line-1\nline-2\n")
+
+    def append_later():
+        time.sleep(0.4)
+        with open(path, "a") as f:
+            f.write("line-3-late\n")
+    t = threading.Thread(target=append_later)
+    t.start()
+    r = runner.invoke(cli.app, ["logs", "follow-me", "-f",
+                                "--poll", "0.3", "--max-polls", "3"])
+    t.join()
+    assert r.exit_code == 0, r.output
+    assert "line-1" in r.output and "line-3-late" in r.output
+    # the late line must appear exactly once (cursor, not re-tail)
+    assert r.output.count("line-3-late") == 1
+
+
+def test_suspend_resume_notebook(daemon, tmp_path):
+    platform, url = daemon
+    m = tmp_path / "nb.yaml"
+    m.write_text(yaml.safe_dump({
+        "apiVersion": "substratus.ai/v1", "kind": "Notebook",
+        "metadata": {"name": "pausable", "namespace": "default"},
+        "spec": {"image": "substratus-amd/notebook"}}))
+    r = runner.invoke(cli.app, ["apply", "-f", str(m)])
+    assert r.exit_code == 0, r.output
+    r = runner.invoke(cli.app, ["suspend", "pausable"])
+    assert r.exit_code == 0, r.output
+    import httpx
+    obj = httpx.get(f"{url}/api/objects/Notebook/default/pausable").json()
+    assert obj["spec"]["suspend"] is True
+    r = runner.invoke(cli.app, ["resume", "pausable"])
+    assert r.exit_code == 0, r.output
+    obj = httpx.get(f"{url}/api/objects/Notebook/default/pausable").json()
+    assert obj["spec"]["suspend"] is False
