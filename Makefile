@@ -26,6 +26,19 @@ test-system:  ## end-to-end Model+Server -> completion request (CPU)
 test-gpu:  ## kernel parity + engine tests (run on MI355X)
 	$(PY) -m pytest tests/ -q -m gpu
 
+.PHONY: test-gpu-sanitize
+test-gpu-sanitize:  ## GPU tests under serialized launches + uncached allocator (sanitizer tier)
+	bash scripts/sanitize_gpu.sh
+
+.PHONY: release
+release:  ## dist/: sdist + wheel + container-tools tarball + checksums (goreleaser analog)
+	bash scripts/release.sh
+
+.PHONY: dev
+dev: build  ## rebuild the extension and restart the local platform (skaffold-loop analog)
+	-$(PY) -m substratus_amd.cli.main down
+	$(PY) -m substratus_amd.cli.main up
+
 .PHONY: bench
 bench: build
 	$(PY) bench.py

@@ -77,9 +77,11 @@ def up(foreground: bool = typer.Option(False, "--foreground", "-f")):
     log = os.path.expanduser("~/.substratus/daemon.log")
     os.makedirs(os.path.dirname(log), exist_ok=True)
     with open(log, "a") as lf:
-        subprocess.Popen(
+        proc = subprocess.Popen(
             [sys.executable, "-m", "substratus_amd.kube.apiserver"],
             stdout=lf, stderr=lf, start_new_session=True)
+    with open(os.path.expanduser("~/.substratus/daemon.pid"), "w") as pf:
+        pf.write(str(proc.pid))
     for _ in range(50):
         try:
             import httpx
@@ -90,6 +92,25 @@ def up(foreground: bool = typer.Option(False, "--foreground", "-f")):
             time.sleep(0.2)
     console.print(f"[red]daemon did not come up; see {log}[/red]")
     raise typer.Exit(1)
+
+
+@app.command()
+def down():
+    """Stop the local platform daemon started by `sub up` (by its recorded
+    PID — never by process-name pattern)."""
+    import signal
+    pidfile = os.path.expanduser("~/.substratus/daemon.pid")
+    if not os.path.exists(pidfile):
+        console.print("no daemon pidfile; nothing to stop")
+        return
+    with open(pidfile) as f:
+        pid = int(f.read().strip())
+    try:
+        os.kill(pid, signal.SIGTERM)
+        console.print(f"stopped daemon (pid {pid})")
+    except ProcessLookupError:
+        console.print(f"daemon (pid {pid}) already gone")
+    os.remove(pidfile)
 
 
 @app.command()
