@@ -18,4 +18,6 @@ export AMD_SERIALIZE_KERNEL=3
 export AMD_SERIALIZE_COPY=3
 export HIP_LAUNCH_BLOCKING=1
 export PYTORCH_NO_HIP_MEMORY_CACHING=1
+# serialized launches invalidate hipGraph stream capture -> eager decode
+export SUB_EAGER_DECODE=1
 exec python -m pytest tests -m gpu -q "$@"

@@ -205,8 +205,12 @@ class Engine:
         # hipGraph capture of a forward containing RCCL collectives
         # (dist.all_reduce inside the row-parallel layers) is unvalidated on
         # ROCm — TP decodes eagerly until graph+RCCL capture is proven.
+        # SUB_EAGER_DECODE=1 forces eager globally (the sanitizer tier runs
+        # with serialized launches, which invalidate stream capture).
+        import os as _os
         if (self.cfg.use_graphs and self.device.type == "cuda"
-                and state().tp_size <= 1):
+                and state().tp_size <= 1
+                and _os.environ.get("SUB_EAGER_DECODE") != "1"):
             return self._decode_graph()
         return self._decode_eager()
 
