@@ -20,7 +20,9 @@ import torch
 import torch.nn as nn
 import torch.nn.functional as F
 
-from .gguf import (QK, dequantize_q4_0, dequantize_q8_0, quantize_q4_0,
+from .gguf import (QK, QK_K, Q4_K_BLOCK_BYTES, Q6_K_BLOCK_BYTES,
+                   QUANT_CODECS, _pack_scale_min_k4, _unpack_scale_min_k4,
+                   dequantize_q4_0, dequantize_q8_0, quantize_q4_0,
                    quantize_q8_0)
 
 
@@ -38,6 +40,56 @@ def repack(packed: torch.Tensor, k: int, quant: str) -> tuple[torch.Tensor, torc
     scales = blocks[:, :, :2].contiguous().reshape(n, -1).view(torch.float16)
     qs = blocks[:, :, 2:].contiguous().reshape(n, k // QK * payload)
     return scales.contiguous(), qs.contiguous()
+
+
+def repack_k(packed: torch.Tensor, k: int, quant: str) -> tuple[torch.Tensor, torch.Tensor]:
+    """K-quant GGUF block stream -> aligned SoA for the GPU kernels
+    (quant_k.hip header documents the layout; llama.cpp's 144/210-byte
+    blocks are not naturally aligned for vector loads)."""
+    n = packed.shape[0]
+    sb = k // QK_K
+    if quant == "q4_K":
+        blocks = packed.reshape(n, sb, Q4_K_BLOCK_BYTES)
+        scales = blocks[:, :, 0:4].contiguous().reshape(n, -1)\
+            .view(torch.float16)                       # [N, 2*SB] d,dmin
+        sc, mn = _unpack_scale_min_k4(
+            blocks[:, :, 4:16].reshape(-1, 12))        # [N*SB, 8]
+        qs = torch.cat([sc.to(torch.uint8).reshape(n, sb * 8),
+                        mn.to(torch.uint8).reshape(n, sb * 8),
+                        blocks[:, :, 16:144].reshape(n, sb * 128)], dim=1)
+        return scales.contiguous(), qs.contiguous()
+    if quant == "q6_K":
+        blocks = packed.reshape(n, sb, Q6_K_BLOCK_BYTES)
+        scales = blocks[:, :, 208:210].contiguous().reshape(n, -1)\
+            .view(torch.float16)                       # [N, SB] d
+        qs = torch.cat([blocks[:, :, 0:128].reshape(n, sb * 128),
+                        blocks[:, :, 128:192].reshape(n, sb * 64),
+                        blocks[:, :, 192:208].reshape(n, sb * 16)], dim=1)
+        return scales.contiguous(), qs.contiguous()
+    raise ValueError(quant)
+
+
+def unrepack_k(scales: torch.Tensor, qs: torch.Tensor, k: int,
+               quant: str) -> torch.Tensor:
+    """Inverse of repack_k: rebuild the GGUF block stream (CPU dequant)."""
+    n = scales.shape[0]
+    sb = k // QK_K
+    scales = scales.cpu()
+    qs = qs.cpu()
+    if quant == "q4_K":
+        out = torch.zeros(n, sb, Q4_K_BLOCK_BYTES, dtype=torch.uint8)
+        out[:, :, 0:4] = scales.view(torch.uint8).reshape(n, sb, 4)
+        sc = qs[:, 0:sb * 8].reshape(n * sb, 8)
+        mn = qs[:, sb * 8:sb * 16].reshape(n * sb, 8)
+        out[:, :, 4:16] = _pack_scale_min_k4(sc, mn).reshape(n, sb, 12)
+        out[:, :, 16:144] = qs[:, sb * 16:].reshape(n, sb, 128)
+        return out.reshape(n, sb * Q4_K_BLOCK_BYTES)
+    out = torch.zeros(n, sb, Q6_K_BLOCK_BYTES, dtype=torch.uint8)
+    out[:, :, 0:128] = qs[:, 0:sb * 128].reshape(n, sb, 128)
+    out[:, :, 128:192] = qs[:, sb * 128:sb * 192].reshape(n, sb, 64)
+    out[:, :, 192:208] = qs[:, sb * 192:sb * 208].reshape(n, sb, 16)
+    out[:, :, 208:210] = scales.view(torch.uint8).reshape(n, sb, 2)
+    return out.reshape(n, sb * Q6_K_BLOCK_BYTES)
 
 
 class QuantLinear(nn.Module):
@@ -60,9 +112,10 @@ class QuantLinear(nn.Module):
         # quantize on the weight's own device (GPU for big models: a 13b
         # CPU quantize would take minutes)
         w = lin.weight.detach().float()
-        qfn = quantize_q4_0 if quant == "q4_0" else quantize_q8_0
-        packed = qfn(w)
-        scales, qs = repack(packed, w.shape[1], quant)
+        qfn = QUANT_CODECS[quant][2]
+        packed = qfn(w.cpu()) if quant.endswith("_K") else qfn(w)
+        rp = repack_k if quant.endswith("_K") else repack
+        scales, qs = rp(packed, w.shape[1], quant)
         dev = lin.weight.device
         bias = None
         if getattr(lin, "bias", None) is not None:
@@ -72,6 +125,9 @@ class QuantLinear(nn.Module):
 
     def _dequant_cpu(self) -> torch.Tensor:
         n, k = self.out_features, self.in_features
+        if self.quant.endswith("_K"):
+            blocks = unrepack_k(self.scales, self.qs, k, self.quant)
+            return QUANT_CODECS[self.quant][3](blocks, k)
         if self.quant == "q4_0":
             blocks = torch.cat([
                 self.scales.view(torch.uint8).reshape(n, k // QK, 2).cpu(),
@@ -88,7 +144,11 @@ class QuantLinear(nn.Module):
             from substratus_amd import _C
             out = torch.empty(self.out_features, self.in_features,
                               dtype=torch.bfloat16, device=self.scales.device)
-            _C.dequant(out, self.scales, self.qs, self.quant == "q8_0")
+            if self.quant.endswith("_K"):
+                _C.dequant_k(out, self.scales, self.qs,
+                             self.quant == "q6_K")
+            else:
+                _C.dequant(out, self.scales, self.qs, self.quant == "q8_0")
             return out
         return self._dequant_cpu()
 
@@ -100,8 +160,12 @@ class QuantLinear(nn.Module):
             from substratus_amd import _C
             y = torch.empty(m, self.out_features, dtype=torch.bfloat16,
                             device=x.device)
-            _C.quant_gemv(y, x2.contiguous().bfloat16(), self.scales,
-                          self.qs, self.quant == "q8_0")
+            if self.quant.endswith("_K"):
+                _C.quant_gemv_k(y, x2.contiguous().bfloat16(), self.scales,
+                                self.qs, self.quant == "q6_K")
+            else:
+                _C.quant_gemv(y, x2.contiguous().bfloat16(), self.scales,
+                              self.qs, self.quant == "q8_0")
             y = y.to(x.dtype)
         else:
             w = self.dequant_weight().to(x.dtype)
@@ -150,14 +214,21 @@ def _llamacpp_to_native(tensors: dict, cfg) -> dict:
 
     def cat_rows(parts: list[GGUFTensor], name: str) -> GGUFTensor:
         gt0 = parts[0]
-        for p in parts:
-            if p.ggml_type != gt0.ggml_type:
-                raise ValueError(
-                    f"mixed ggml types in fused tensor {name}: "
-                    f"{[p.ggml_type for p in parts]} (mixed-precision "
-                    "K-quant files are unsupported; re-quantize to pure "
-                    "q4_0 or q8_0)")
         shape = [sum(p.shape[0] for p in parts)] + list(gt0.shape[1:])
+        if any(p.ggml_type != gt0.ggml_type for p in parts):
+            # Q4_K_M files mix q4_K and q6_K WITHIN a fusion group (e.g.
+            # attn_v is q6_K while attn_q/attn_k are q4_K). The fused
+            # QuantLinear needs one codec, so dequantize the parts and
+            # re-encode the fused tensor at the widest member format —
+            # a one-time re-quantization of already-quantized data
+            # (error bounded by the codec step, not the weights).
+            from .gguf import GGML_Q6_K, GGML_Q8_0, QUANT_CODECS
+            types = {p.ggml_type for p in parts}
+            target = GGML_Q8_0 if GGML_Q8_0 in types else GGML_Q6_K
+            tname = {v[0]: k for k, v in QUANT_CODECS.items()}[target]
+            full = torch.cat([p.dequantize() for p in parts], dim=0)
+            return GGUFTensor(name, shape, target,
+                              QUANT_CODECS[tname][2](full))
         return GGUFTensor(name, shape, gt0.ggml_type,
                           torch.cat([p.data for p in parts], dim=0))
 
@@ -208,26 +279,25 @@ def load_gguf_into_model(model: nn.Module, path: str, device,
     """Load a GGUF checkpoint: quantized 2-D tensors become QuantLinear
     buffers in place; everything else is dequantized/copied. Returns the
     quant name."""
-    from .gguf import GGML_Q4_0, GGML_Q8_0, read_gguf
+    from .gguf import (GGML_Q4_0, GGML_Q4_K, GGML_Q6_K, GGML_Q8_0,
+                       read_gguf)
     meta, tensors = read_gguf(path)
+    gt_names = {GGML_Q4_0: "q4_0", GGML_Q8_0: "q8_0",
+                GGML_Q4_K: "q4_K", GGML_Q6_K: "q6_K"}
     quant = meta.get("substratus.quant", "q4_0")
     if any(n.startswith("blk.") or n == "token_embd.weight"
            for n in tensors):
         # llama.cpp-named file (ecosystem GGUFs)
         tensors = _llamacpp_to_native(tensors, model.cfg)
         for gt in tensors.values():
-            if gt.ggml_type == GGML_Q4_0:
-                quant = "q4_0"
-                break
-            if gt.ggml_type == GGML_Q8_0:
-                quant = "q8_0"
+            if gt.ggml_type in gt_names:
+                quant = gt_names[gt.ggml_type]
                 break
     sd = {}
     qmap = {}
     for name, gt in tensors.items():
-        if gt.ggml_type in (GGML_Q4_0, GGML_Q8_0):
-            qname = "q4_0" if gt.ggml_type == GGML_Q4_0 else "q8_0"
-            qmap[name] = (gt, qname)
+        if gt.ggml_type in gt_names:
+            qmap[name] = (gt, gt_names[gt.ggml_type])
         else:
             sd[name] = gt.dequantize()
     # first load full-precision leftovers
@@ -240,7 +310,8 @@ def load_gguf_into_model(model: nn.Module, path: str, device,
         mod = mods.get(mod_name)
         from ..parallel.tp import Linear
         if isinstance(mod, Linear):
-            scales, qs = repack(gt.data, gt.shape[-1], qname)
+            rp = repack_k if qname.endswith("_K") else repack
+            scales, qs = rp(gt.data, gt.shape[-1], qname)
             parent_name, attr = (mod_name.rsplit(".", 1) + [""])[:2] \
                 if "." in mod_name else ("", mod_name)
             parent = mods[parent_name] if parent_name else model

@@ -30,6 +30,8 @@ void launch_adamw(bf16_t*, const bf16_t*, float*, float*, long, float, float, fl
 void launch_adamw_f32(float*, const float*, float*, float*, long, float, float, float, float, float, int, float, hipStream_t);
 void launch_quant_gemv(bf16_t*, const bf16_t*, const void*, const unsigned char*, int, int, int, bool, hipStream_t);
 void launch_dequant(bf16_t*, const void*, const unsigned char*, long, int, bool, hipStream_t);
+void launch_quant_gemv_k(bf16_t*, const bf16_t*, const void*, const unsigned char*, int, int, int, bool, hipStream_t);
+void launch_dequant_k(bf16_t*, const void*, const unsigned char*, int, int, bool, hipStream_t);
 int decode_gemm_splitk(int, int);
 bool decode_gemm_supported(int, int, int);
 void launch_decode_gemm(bf16_t*, const bf16_t*, const bf16_t*, float*, int, int, int, int, hipStream_t);
@@ -297,6 +299,31 @@ long mfma_gemm_pick_splitk(long m, long n, long k) {
   return mfma_gemm_splitk((int)m, (int)n, (int)k);
 }
 
+void quant_gemv_k(torch::Tensor y, torch::Tensor x, torch::Tensor scales,
+                  torch::Tensor qs, bool q6) {
+  check_bf16(y, "y"); check_bf16(x, "x");
+  TORCH_CHECK(scales.scalar_type() == torch::kHalf, "scales must be fp16");
+  TORCH_CHECK(qs.scalar_type() == torch::kByte, "qs must be uint8");
+  int m = x.size(0), k = x.size(1), n = y.size(1);
+  TORCH_CHECK(m >= 1 && m <= 8, "quant_gemv_k is the M<=8 decode path");
+  TORCH_CHECK(k % 256 == 0, "K must be a multiple of 256 (K-quants)");
+  launch_quant_gemv_k(bf(y), bfc(x), scales.data_ptr(),
+                      qs.data_ptr<unsigned char>(), m, n, k, q6, stream());
+  post_launch_check("quant_gemv_k");
+}
+
+void dequant_k(torch::Tensor out, torch::Tensor scales, torch::Tensor qs,
+               bool q6) {
+  check_bf16(out, "out");
+  TORCH_CHECK(scales.scalar_type() == torch::kHalf, "scales must be fp16");
+  TORCH_CHECK(qs.scalar_type() == torch::kByte, "qs must be uint8");
+  int n = out.size(0), k = out.size(1);
+  TORCH_CHECK(k % 256 == 0, "K must be a multiple of 256 (K-quants)");
+  launch_dequant_k(bf(out), scales.data_ptr(),
+                   qs.data_ptr<unsigned char>(), n, k, q6, stream());
+  post_launch_check("dequant_k");
+}
+
 torch::Tensor mfma_probe(torch::Tensor a, torch::Tensor b) {
   check_bf16(a, "a"); check_bf16(b, "b");
   TORCH_CHECK(a.size(0) == 16 && a.size(1) == 32 && b.size(0) == 32 &&
@@ -448,6 +475,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("attn_prefill", &attn_prefill);
   m.def("attn_train_fwd", &attn_train_fwd);
   m.def("mfma_gemm", &mfma_gemm);
+  m.def("quant_gemv_k", &quant_gemv_k);
+  m.def("dequant_k", &dequant_k);
   m.def("mfma_gemm_pick_splitk", &mfma_gemm_pick_splitk);
   m.def("attn_train_bwd", &attn_train_bwd);
   m.def("mfma_probe", &mfma_probe);

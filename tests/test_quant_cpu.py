@@ -227,3 +227,185 @@ def test_gguf_nondefault_alignment_roundtrip(tmp_path):
     assert meta["general.alignment"] == 64
     assert torch.equal(tensors["a"].dequantize(), a)
     assert torch.equal(tensors["b"].dequantize(), b)
+
+
+def _kq_model():
+    import torch
+    from substratus_amd.models.config import ModelConfig, register
+    from substratus_amd.models.model import CausalLM
+    cfg = register(ModelConfig(
+        name="ktest-tiny", family="llama", hidden_size=256,
+        intermediate_size=512, num_layers=2, num_heads=4, num_kv_heads=2,
+        vocab_size=512, max_position_embeddings=256))
+    torch.manual_seed(11)
+    m = CausalLM(cfg, device="meta", dtype=torch.float32)
+    m = m.to_empty(device="cpu")
+    m.init_random(7)
+    return cfg, m
+
+
+def test_kquant_codec_error_bounds():
+    """q4_K beats q4_0 (sub-block mins) and q6_K beats both."""
+    import torch
+    from substratus_amd.models.gguf import QUANT_CODECS
+    torch.manual_seed(5)
+    w = torch.randn(16, 1024) * 0.04
+    rms = {}
+    for q in ("q4_0", "q4_K", "q6_K"):
+        _, _, qf, df = QUANT_CODECS[q]
+        back = df(qf(w), 1024)
+        rms[q] = float((back - w).pow(2).mean().sqrt() /
+                       w.pow(2).mean().sqrt())
+    assert rms["q4_K"] < rms["q4_0"]
+    assert rms["q6_K"] < rms["q4_K"] / 2
+    assert rms["q6_K"] < 0.03
+
+
+def test_q4k_ecosystem_file_generates_and_matches_dequant(tmp_path):
+    """A llama.cpp-named q4_K file (ecosystem Q4_K class) loads through
+    the K-quant path and generates EXACTLY the tokens of a model whose
+    weights are the codec's dequantized values (the QuantLinear CPU path
+    is F.linear over the same dequant)."""
+    import torch
+    from substratus_amd.models import gguf
+    from substratus_amd.models.model import CausalLM
+    from substratus_amd.models.quant import QuantLinear, load_gguf_into_model
+    from substratus_amd.serve.engine import Engine, EngineConfig
+
+    cfg, ref = _kq_model()
+    sd = {k: v.float() for k, v in ref.state_dict().items()}
+    hq, hkv, d = cfg.num_heads, cfg.num_kv_heads, cfg.head_dim
+
+    def perm_fwd(w, n_head):
+        n, k = w.shape
+        return (w.reshape(n_head, 2, d // 2, k).swapaxes(1, 2)
+                 .reshape(n, k))
+
+    tensors = {}
+
+    def q4k(name, w):
+        tensors[name] = (gguf.quantize_q4_K(w), gguf.GGML_Q4_K,
+                         list(w.shape))
+
+    def f32(name, w):
+        tensors[name] = (w.contiguous(), gguf.GGML_F32, list(w.shape))
+
+    f32("token_embd.weight", sd["embed_tokens.weight"])
+    f32("output_norm.weight", sd["final_norm.weight"])
+    q4k("output.weight", sd["lm_head.weight"])
+    for i in range(cfg.num_layers):
+        p = f"blocks.{i}."
+        f32(f"blk.{i}.attn_norm.weight", sd[p + "norm1.weight"])
+        f32(f"blk.{i}.ffn_norm.weight", sd[p + "norm2.weight"])
+        wq, wk, wv = sd[p + "wqkv.weight"].split(
+            [hq * d, hkv * d, hkv * d], dim=0)
+        q4k(f"blk.{i}.attn_q.weight", perm_fwd(wq, hq))
+        q4k(f"blk.{i}.attn_k.weight", perm_fwd(wk, hkv))
+        q4k(f"blk.{i}.attn_v.weight", wv)
+        q4k(f"blk.{i}.attn_output.weight", sd[p + "wo.weight"])
+        g, u = sd[p + "w_gateup.weight"].chunk(2, dim=0)
+        q4k(f"blk.{i}.ffn_gate.weight", g)
+        q4k(f"blk.{i}.ffn_up.weight", u)
+        q4k(f"blk.{i}.ffn_down.weight", sd[p + "w_down.weight"])
+    path = str(tmp_path / "ktest.Q4_K.gguf")
+    gguf.write_gguf(path, tensors)
+
+    qm = CausalLM(cfg, device="meta", dtype=torch.float32)
+    qm = qm.to_empty(device="cpu")
+    qm.init_random(7)
+    quant = load_gguf_into_model(qm, path, "cpu", torch.float32)
+    assert quant == "q4_K"
+    assert isinstance(qm.blocks[0].wqkv, QuantLinear)
+    assert qm.blocks[0].wqkv.quant == "q4_K"
+
+    # reference: plain model with the dequantized weights
+    dm = CausalLM(cfg, device="meta", dtype=torch.float32)
+    dm = dm.to_empty(device="cpu")
+    dm.init_random(7)
+    dsd = dm.state_dict()
+    for name, mod in qm.named_modules():
+        if isinstance(mod, QuantLinear):
+            dsd[name + ".weight"] = mod._dequant_cpu()
+    dm.load_state_dict(dsd)
+
+    toks = {}
+    for tag, model in (("quant", qm), ("dequant", dm)):
+        eng = Engine(EngineConfig(model="ktest-tiny", dtype=torch.float32,
+                                  max_batch=1, max_seq_len=64,
+                                  kv_blocks=16), device="cpu", model=model)
+        eng.add_request([3, 1, 4, 1, 5], max_new_tokens=8)
+        out = []
+        while eng.has_work():
+            out.extend(o.token for o in eng.step())
+        toks[tag] = out
+    assert toks["quant"] == toks["dequant"], toks
+
+
+def test_mixed_kquant_fusion_loads(tmp_path):
+    """Q4_K_M-style mixing: attn_v is q6_K while attn_q/k are q4_K — the
+    fused wqkv re-encodes at q6_K and the weights stay within codec
+    error of the originals."""
+    import torch
+    from substratus_amd.models import gguf
+    from substratus_amd.models.model import CausalLM
+    from substratus_amd.models.quant import QuantLinear, load_gguf_into_model
+
+    cfg, ref = _kq_model()
+    sd = {k: v.float() for k, v in ref.state_dict().items()}
+    hq, hkv, d = cfg.num_heads, cfg.num_kv_heads, cfg.head_dim
+
+    def perm_fwd(w, n_head):
+        n, k = w.shape
+        return (w.reshape(n_head, 2, d // 2, k).swapaxes(1, 2)
+                 .reshape(n, k))
+
+    tensors = {"token_embd.weight": (sd["embed_tokens.weight"].contiguous(),
+                                     gguf.GGML_F32, [cfg.vocab_size, 256]),
+               "output_norm.weight": (sd["final_norm.weight"].contiguous(),
+                                      gguf.GGML_F32, [256]),
+               "output.weight": (gguf.quantize_q6_K(sd["lm_head.weight"]),
+                                 gguf.GGML_Q6_K,
+                                 list(sd["lm_head.weight"].shape))}
+    for i in range(cfg.num_layers):
+        p = f"blocks.{i}."
+        tensors[f"blk.{i}.attn_norm.weight"] = (
+            sd[p + "norm1.weight"].contiguous(), gguf.GGML_F32, [256])
+        tensors[f"blk.{i}.ffn_norm.weight"] = (
+            sd[p + "norm2.weight"].contiguous(), gguf.GGML_F32, [256])
+        wq, wk, wv = sd[p + "wqkv.weight"].split(
+            [hq * d, hkv * d, hkv * d], dim=0)
+        tensors[f"blk.{i}.attn_q.weight"] = (
+            gguf.quantize_q4_K(perm_fwd(wq, hq)), gguf.GGML_Q4_K,
+            list(wq.shape))
+        tensors[f"blk.{i}.attn_k.weight"] = (
+            gguf.quantize_q4_K(perm_fwd(wk, hkv)), gguf.GGML_Q4_K,
+            list(wk.shape))
+        tensors[f"blk.{i}.attn_v.weight"] = (
+            gguf.quantize_q6_K(wv), gguf.GGML_Q6_K, list(wv.shape))
+        tensors[f"blk.{i}.attn_output.weight"] = (
+            gguf.quantize_q4_K(sd[p + "wo.weight"]), gguf.GGML_Q4_K,
+            list(sd[p + "wo.weight"].shape))
+        g, u = sd[p + "w_gateup.weight"].chunk(2, dim=0)
+        tensors[f"blk.{i}.ffn_gate.weight"] = (
+            gguf.quantize_q4_K(g), gguf.GGML_Q4_K, list(g.shape))
+        tensors[f"blk.{i}.ffn_up.weight"] = (
+            gguf.quantize_q4_K(u), gguf.GGML_Q4_K, list(u.shape))
+        tensors[f"blk.{i}.ffn_down.weight"] = (
+            gguf.quantize_q6_K(sd[p + "w_down.weight"]), gguf.GGML_Q6_K,
+            list(sd[p + "w_down.weight"].shape))
+    path = str(tmp_path / "ktest.Q4_K_M.gguf")
+    gguf.write_gguf(path, tensors)
+
+    qm = CausalLM(cfg, device="meta", dtype=torch.float32)
+    qm = qm.to_empty(device="cpu")
+    qm.init_random(7)
+    load_gguf_into_model(qm, path, "cpu", torch.float32)
+    wqkv = qm.blocks[0].wqkv
+    assert isinstance(wqkv, QuantLinear) and wqkv.quant == "q6_K"
+    back = wqkv._dequant_cpu()
+    orig = sd["blocks.0.wqkv.weight"]
+    rel = float((back - orig).pow(2).mean().sqrt() /
+                orig.pow(2).mean().sqrt())
+    assert rel < 0.12, rel
+    # down stayed pure q6_K (no re-encode)
+    assert qm.blocks[0].w_down.quant == "q6_K"
