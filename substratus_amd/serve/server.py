@@ -363,6 +363,93 @@ def create_app(engine: Optional[Engine] = None,
             },
         })
 
+    @app.post("/v1/chat/completions")
+    async def chat_completions(req: Request):
+        """OpenAI chat API on the same engine (beyond the reference's
+        basaran surface, which was /v1/completions only): messages render
+        through a minimal chat template and the completion flow runs
+        unchanged."""
+        try:
+            body = await req.json()
+        except Exception:
+            return _bad_request("request body is not valid JSON")
+        msgs = body.get("messages")
+        if not isinstance(msgs, list) or not msgs:
+            return _bad_request("messages must be a non-empty list")
+        parts = []
+        for m_ in msgs:
+            if not isinstance(m_, dict) or "content" not in m_:
+                return _bad_request("each message needs role and content")
+            parts.append(f"{m_.get('role', 'user')}: {m_['content']}")
+        parts.append("assistant:")
+        prompt = "\n".join(parts)
+        prompt_tokens = tok_obj.encode(prompt)
+        try:
+            max_tokens = int(body.get("max_tokens", 64))
+            temperature = float(body.get("temperature", 1.0))
+            top_p = float(body.get("top_p", 1.0))
+            stream = bool(body.get("stream", False))
+        except (TypeError, ValueError) as e:
+            return _bad_request(f"invalid parameter: {e}")
+        limit = engine.cfg.max_seq_len
+        if len(prompt_tokens) + max_tokens > limit:
+            return _bad_request("messages exceed the server max_length")
+        t_start = time.time()
+        sid, out_q = worker.submit(prompt_tokens, max_tokens, temperature,
+                                   0, top_p, eos_token=tok_obj.eos)
+        if stream:
+            def gen():
+                while True:
+                    o: StepOutput = out_q.get(timeout=600)
+                    if o.error:
+                        yield ("data: " + json.dumps(
+                            {"error": {"message": o.error,
+                                       "type": "server_error"}}) + "\n\n")
+                        yield "data: [DONE]\n\n"
+                        return
+                    chunk = {
+                        "id": f"chatcmpl-{sid}",
+                        "object": "chat.completion.chunk",
+                        "model": model_id,
+                        "choices": [{"index": 0, "delta": {
+                            "content": tok_obj.decode([o.token])},
+                            "finish_reason": "stop" if o.finished
+                            else None}],
+                    }
+                    yield f"data: {json.dumps(chunk)}\n\n"
+                    if o.finished:
+                        yield "data: [DONE]\n\n"
+                        return
+            return StreamingResponse(gen(), media_type="text/event-stream")
+        toks = []
+        while True:
+            o: StepOutput = out_q.get(timeout=600)
+            if o.error:
+                return JSONResponse({"error": {"message": o.error,
+                                               "type": "server_error"}},
+                                    status_code=500)
+            toks.append(o.token)
+            if o.finished:
+                break
+        return JSONResponse({
+            "id": f"chatcmpl-{sid}",
+            "object": "chat.completion",
+            "created": int(t_start),
+            "model": model_id,
+            "choices": [{
+                "index": 0,
+                "message": {"role": "assistant",
+                            "content": tok_obj.decode(toks)},
+                "finish_reason": "length" if len(toks) >= max_tokens
+                                 else "stop",
+            }],
+            "usage": {
+                "prompt_tokens": len(prompt_tokens),
+                "completion_tokens": len(toks),
+                "total_tokens": len(prompt_tokens) + len(toks),
+            },
+        })
+
     return app
 
 

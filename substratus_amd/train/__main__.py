@@ -35,7 +35,11 @@ def main() -> None:
     on_gpu = torch.cuda.is_available()
     cfg = TrainConfig(
         model=model_name,
-        lr=float(params.get("lr", 2e-5)),
+        lr=float(params.get("lr", params.get("learning_rate", 2e-5))),
+        warmup_steps=int(params.get("warmup_steps", 0)),
+        lr_schedule=str(params.get("lr_scheduler_type",
+                                   params.get("lr_schedule", "constant"))),
+        min_lr=float(params.get("min_lr", 0.0)),
         micro_batch=int(params.get("micro_batch", 4)),
         seq_len=int(params.get("seq_len", params.get("max_length", 2048))),
         num_train_epochs=int(params.get("num_train_epochs", 1)),

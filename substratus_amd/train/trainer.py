@@ -18,6 +18,7 @@ over RCCL, fused AdamW HIP kernel for the update.
 from __future__ import annotations
 
 import json
+import math
 import os
 import time
 from dataclasses import dataclass, field
@@ -52,6 +53,13 @@ class TrainConfig:
     seed: int = 0
     dtype: torch.dtype = torch.bfloat16
     grad_clip: float = 1.0
+    # LR schedule (HF TrainingArguments parity: warmup_steps +
+    # lr_scheduler_type): linear warmup to lr, then constant | cosine |
+    # linear decay toward min_lr over max_steps (or endlessly constant
+    # when max_steps is 0)
+    warmup_steps: int = 0
+    lr_schedule: str = "constant"   # constant | cosine | linear
+    min_lr: float = 0.0
     # ZeRO-1 sharded optimizer (reduce-scatter grads, shard AdamW,
     # all-gather params) whenever dp_size > 1; False = plain bucketed
     # all-reduce DDP (train/ddp.py GradBuckets)
@@ -68,12 +76,14 @@ class AdamWState:
         self.v = [torch.zeros_like(m) for m in self.m]
         self.step_num = 0
 
-    def step(self, cfg: TrainConfig, grad_scale: float = 1.0) -> None:
+    def step(self, cfg: TrainConfig, grad_scale: float = 1.0,
+             lr: Optional[float] = None) -> None:
         self.step_num += 1
+        lr = cfg.lr if lr is None else lr
         for p, m, v in zip(self.params, self.m, self.v):
             if p.grad is None:
                 continue
-            ops.adamw_step(p.data, p.grad, m, v, cfg.lr, cfg.beta1, cfg.beta2,
+            ops.adamw_step(p.data, p.grad, m, v, lr, cfg.beta1, cfg.beta2,
                            cfg.eps, cfg.weight_decay, self.step_num,
                            grad_scale)
 
@@ -125,6 +135,22 @@ class Trainer:
         self.global_step = 0
         self.tokens_seen = 0
 
+    def lr_at(self, step: int) -> float:
+        """Scheduled LR for 1-based step `step`."""
+        cfg = self.cfg
+        if cfg.warmup_steps > 0 and step <= cfg.warmup_steps:
+            return cfg.lr * step / cfg.warmup_steps
+        if cfg.lr_schedule == "constant" or cfg.max_steps <= 0:
+            return cfg.lr
+        total = max(cfg.max_steps - cfg.warmup_steps, 1)
+        t = min(max(step - cfg.warmup_steps, 0) / total, 1.0)
+        if cfg.lr_schedule == "cosine":
+            return cfg.min_lr + (cfg.lr - cfg.min_lr) * 0.5 * (
+                1.0 + math.cos(math.pi * t))
+        if cfg.lr_schedule == "linear":
+            return cfg.min_lr + (cfg.lr - cfg.min_lr) * (1.0 - t)
+        raise ValueError(f"unknown lr_schedule {cfg.lr_schedule!r}")
+
     # ------------------------------------------------------------- one step
     def train_step(self, tokens: torch.Tensor) -> float:
         """tokens: [micro_batch, seq_len+1] int64 (inputs ‖ shifted targets).
@@ -135,13 +161,14 @@ class Trainer:
         flat = logits.reshape(-1, logits.shape[-1])
         loss, dlogits = ops.cross_entropy_fwd_bwd(flat.detach(), targets)
         flat.backward(gradient=dlogits)
+        lr = self.lr_at(self.global_step + 1)
         if self.use_zero:
             self.zero.finalize()
             gscale = 1.0
             if self.cfg.grad_clip > 0:
                 total = self.zero.global_grad_norm()
                 gscale = min(1.0, self.cfg.grad_clip / (total + 1e-6))
-            self.zero.step(self.cfg.lr, self.cfg.beta1, self.cfg.beta2,
+            self.zero.step(lr, self.cfg.beta1, self.cfg.beta2,
                            self.cfg.eps, self.cfg.weight_decay, gscale)
             self.zero.zero_grad()
         else:
@@ -156,7 +183,7 @@ class Trainer:
                 norms = torch._foreach_norm(grads, 2.0)
                 total = torch.linalg.vector_norm(torch.stack(norms))
                 gscale = min(1.0, self.cfg.grad_clip / (float(total) + 1e-6))
-            self.opt.step(self.cfg, grad_scale=gscale)
+            self.opt.step(self.cfg, grad_scale=gscale, lr=lr)
             self.opt.zero_grad()
         self.global_step += 1
         self.tokens_seen += inputs.numel() * max(self.st.dp_size, 1)

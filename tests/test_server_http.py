@@ -392,3 +392,30 @@ def test_engine_failure_blast_radius():
     assert "KV blocks" in rb.json()["error"]["message"]
     assert rg.status_code == 200
     assert rg.json()["usage"]["completion_tokens"] >= 1
+
+
+def test_chat_completions(client):
+    r = client.post("/v1/chat/completions", json={
+        "messages": [{"role": "system", "content": "be brief"},
+                     {"role": "user", "content": "hello"}],
+        "max_tokens": 4, "temperature": 0.0})
+    assert r.status_code == 200
+    body = r.json()
+    assert body["object"] == "chat.completion"
+    assert body["choices"][0]["message"]["role"] == "assistant"
+    assert body["usage"]["completion_tokens"] >= 1
+
+
+def test_chat_completions_stream(client):
+    with client.stream("POST", "/v1/chat/completions", json={
+            "messages": [{"role": "user", "content": "hi"}],
+            "max_tokens": 3, "stream": True}) as r:
+        assert r.status_code == 200
+        events = [l for l in r.iter_lines() if l.startswith("data:")]
+    assert events[-1].strip() == "data: [DONE]"
+    assert "chat.completion.chunk" in events[0]
+
+
+def test_chat_completions_bad_messages(client):
+    r = client.post("/v1/chat/completions", json={"messages": "nope"})
+    assert r.status_code == 400

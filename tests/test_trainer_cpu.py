@@ -264,3 +264,22 @@ def test_ddp_modes_checkpoint_resume_2rank(tmp_path, zero, port):
         p.join(timeout=60)
     for k in results[0]:
         torch.testing.assert_close(results[0][k], results[1][k])
+
+
+def test_lr_schedule():
+    """Warmup + cosine/linear/constant schedules (HF TrainingArguments
+    warmup_steps / lr_scheduler_type parity)."""
+    cfg = TrainConfig(model="llama-tiny", lr=1e-3, warmup_steps=4,
+                      lr_schedule="cosine", max_steps=12, min_lr=1e-4,
+                      dtype=torch.float32, artifacts_dir="/tmp/x")
+    tr = Trainer.__new__(Trainer)   # lr_at needs only cfg
+    tr.cfg = cfg
+    assert tr.lr_at(1) == pytest.approx(2.5e-4)
+    assert tr.lr_at(4) == pytest.approx(1e-3)
+    assert tr.lr_at(12) == pytest.approx(1e-4)          # decayed to min
+    assert tr.lr_at(8) == pytest.approx(
+        1e-4 + (1e-3 - 1e-4) * 0.5, rel=1e-6)           # cosine midpoint
+    cfg.lr_schedule = "linear"
+    assert tr.lr_at(8) == pytest.approx(1e-4 + (1e-3 - 1e-4) * 0.5)
+    cfg.lr_schedule = "constant"
+    assert tr.lr_at(9) == pytest.approx(1e-3)
