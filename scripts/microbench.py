@@ -161,6 +161,35 @@ def bench_wstream(m=256):
               f"parity={'OK' if ok else f'FAIL err={err:.4f}/{scale:.1f}'}")
 
 
+def bench_attn_v5():
+    """V4 vs V5 prefill attention, interleaved rounds (guide rule 24), on
+    the serving (llama2-7b) and training (falcon) shapes."""
+    from substratus_amd import _C
+    shapes = [("llama S512", 8, 32, 32, 128, 512),
+              ("llama S2048", 4, 32, 32, 128, 2048),
+              ("falcon S2048", 4, 71, 1, 64, 2048)]
+    for name, B, hq, hkv, d, S in shapes:
+        torch.manual_seed(0)
+        t = B * S
+        q = (torch.rand(t, hq, d, device="cuda") * 2 - 1).bfloat16()
+        k = (torch.rand(t, hkv, d, device="cuda") * 2 - 1).bfloat16()
+        v = (torch.rand(t, hkv, d, device="cuda") * 2 - 1).bfloat16()
+        ss = torch.arange(0, (B + 1) * S, S, dtype=torch.int32,
+                          device="cuda")
+        out = torch.empty_like(q)
+        scale = d ** -0.5
+        t4, t5 = [], []
+        for _ in range(5):
+            t4.append(timeit(lambda: _C.attn_prefill(
+                out, q, k, v, ss, S, scale), iters=30))
+            t5.append(timeit(lambda: _C.attn_v5(
+                out, q, k, v, ss, S, scale, None), iters=30))
+        flops = 2 * 2 * B * hq * S * S * d * 0.5
+        print(f"attn {name}: v4 {min(t4)*1e6:.0f} us "
+              f"({flops/min(t4)/1e12:.0f} TF) vs v5 {min(t5)*1e6:.0f} us "
+              f"({flops/min(t5)/1e12:.0f} TF)")
+
+
 def bench_mgemm(m=512):
     """mfma_gemm v3 (256x256 glds tile) vs hipBLASLt, interleaved rounds
     (guide rule 24) + parity vs fp32 matmul. Shapes: llama2-7b B=512 decode
@@ -236,6 +265,8 @@ if __name__ == "__main__":
     if a.which in ("all", "wstream"):
         bench_wstream()
         bench_wstream(m=64)
+    if a.which in ("all", "attnv5"):
+        bench_attn_v5()
     if a.which in ("all", "mgemm"):
         m = int(os.environ.get("MGEMM_M", "512"))
         bench_mgemm(m)

@@ -133,6 +133,12 @@ def attn_prefill(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
         return ref.attn_prefill(q, k, v, seq_start, scale)
     out = torch.empty_like(q)
     max_len = int((seq_start[1:] - seq_start[:-1]).max())
+    if os.environ.get("SUB_ATTN_V5", "1") != "0":
+        # V5: 8-wave swapped-QK^T in-register-softmax structure —
+        # 293-516 TF vs V4's 167-250 (scripts/microbench.py --which attnv5)
+        ext.attn_v5(out, q.contiguous(), k.contiguous(), v.contiguous(),
+                    seq_start, max_len, scale, None)
+        return out
     ext.attn_prefill(out, q.contiguous(), k.contiguous(), v.contiguous(),
                      seq_start, max_len, scale)
     return out
@@ -152,7 +158,10 @@ class _FlashAttnTrain(torch.autograd.Function):
         out = torch.empty_like(q)
         lse = torch.empty(q.shape[0], q.shape[1], dtype=torch.float32,
                           device=q.device)
-        ext.attn_train_fwd(out, lse, q, k, v, seq_start, max_len, scale)
+        if os.environ.get("SUB_ATTN_V5", "1") != "0":
+            ext.attn_v5(out, q, k, v, seq_start, max_len, scale, lse)
+        else:
+            ext.attn_train_fwd(out, lse, q, k, v, seq_start, max_len, scale)
         ctx.save_for_backward(q, k, v, out, lse, seq_start)
         ctx.max_len = max_len
         ctx.scale = scale

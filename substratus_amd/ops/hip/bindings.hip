@@ -19,11 +19,14 @@ int paged_attn_nsplit(int, int, int);
 bool paged_attn_decode_supported(int, int);
 void launch_attn_prefill(bf16_t*, const bf16_t*, const bf16_t*, const bf16_t*, const int*, int, int, int, int, int, float, hipStream_t, float*);
 bool attn_prefill_supported(int);
+void launch_attn_v5(bf16_t*, const bf16_t*, const bf16_t*, const bf16_t*, const int*, int, int, int, int, int, float, hipStream_t, float*);
+bool attn_v5_supported(int);
 void launch_attn_bwd_delta(float*, const bf16_t*, const bf16_t*, long, int, hipStream_t);
 void launch_attn_bwd(bf16_t*, float*, float*, const bf16_t*, const bf16_t*, const bf16_t*, const bf16_t*, const float*, const float*, const int*, int, int, int, int, int, float, hipStream_t);
 bool attn_bwd_supported(int);
 void launch_attn_prefill_probe(int, bf16_t*, const bf16_t*, const bf16_t*, const bf16_t*, const int*, int, int, int, int, float, hipStream_t);
 void launch_mfma_probe(float*, const bf16_t*, const bf16_t*, hipStream_t);
+void launch_mfma32_probe(float*, const bf16_t*, const bf16_t*, hipStream_t);
 void launch_softmax_sample(long*, const bf16_t*, const float*, const int*, const float*, const float*, int, int, hipStream_t);
 void launch_cross_entropy(float*, bf16_t*, const bf16_t*, const long*, long, int, long, float, hipStream_t);
 void launch_adamw(bf16_t*, const bf16_t*, float*, float*, long, float, float, float, float, float, int, float, hipStream_t);
@@ -211,6 +214,27 @@ void attn_prefill(torch::Tensor out, torch::Tensor q, torch::Tensor k,
   post_launch_check("attn_prefill");
 }
 
+void attn_v5(torch::Tensor out, torch::Tensor q, torch::Tensor k,
+             torch::Tensor v, torch::Tensor seq_start, long max_len,
+             double scale, c10::optional<torch::Tensor> lse) {
+  check_bf16(out, "out"); check_bf16(q, "q"); check_bf16(k, "k");
+  check_bf16(v, "v");
+  TORCH_CHECK(seq_start.scalar_type() == torch::kInt, "seq_start int32");
+  int hq = q.size(1), d = q.size(2), hkv = k.size(1);
+  int num_seqs = seq_start.size(0) - 1;
+  TORCH_CHECK(attn_v5_supported(d), "unsupported head_dim ", d);
+  TORCH_CHECK(hq % hkv == 0, "hq must be divisible by hkv");
+  float* lp = nullptr;
+  if (lse.has_value()) {
+    TORCH_CHECK(lse->scalar_type() == torch::kFloat, "lse f32");
+    lp = lse->data_ptr<float>();
+  }
+  launch_attn_v5(bf(out), bfc(q), bfc(k), bfc(v),
+                 seq_start.data_ptr<int>(), num_seqs, (int)max_len, hq,
+                 hkv, d, (float)scale, stream(), lp);
+  post_launch_check("attn_v5");
+}
+
 // forward with saved log2-domain LSE (training path)
 void attn_train_fwd(torch::Tensor out, torch::Tensor lse, torch::Tensor q,
                     torch::Tensor k, torch::Tensor v,
@@ -322,6 +346,16 @@ void dequant_k(torch::Tensor out, torch::Tensor scales, torch::Tensor qs,
   launch_dequant_k(bf(out), scales.data_ptr(),
                    qs.data_ptr<unsigned char>(), n, k, q6, stream());
   post_launch_check("dequant_k");
+}
+
+torch::Tensor mfma32_probe(torch::Tensor a, torch::Tensor b) {
+  check_bf16(a, "a"); check_bf16(b, "b");
+  TORCH_CHECK(a.size(0) == 32 && a.size(1) == 16 &&
+              b.size(0) == 16 && b.size(1) == 32, "probe is 32x16 x 16x32");
+  auto d = torch::empty({32, 32}, a.options().dtype(torch::kFloat));
+  launch_mfma32_probe(d.data_ptr<float>(), bfc(a), bfc(b), stream());
+  post_launch_check("mfma32_probe");
+  return d;
 }
 
 torch::Tensor mfma_probe(torch::Tensor a, torch::Tensor b) {
@@ -476,6 +510,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("attn_train_fwd", &attn_train_fwd);
   m.def("mfma_gemm", &mfma_gemm);
   m.def("quant_gemv_k", &quant_gemv_k);
+  m.def("mfma32_probe", &mfma32_probe);
+  m.def("attn_v5", &attn_v5);
   m.def("dequant_k", &dequant_k);
   m.def("mfma_gemm_pick_splitk", &mfma_gemm_pick_splitk);
   m.def("attn_train_bwd", &attn_train_bwd);

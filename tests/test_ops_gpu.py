@@ -791,3 +791,70 @@ def test_kquant_engine_generates_gpu(tmp_path):
     assert len(toks[DEV]) == 8
     agree = sum(a == b for a, b in zip(toks["cpu"], toks[DEV]))
     assert agree >= 6, toks   # bf16 vs fp32 activations may diverge late
+
+
+def test_mfma32_fragment_layout():
+    """v_mfma_f32_32x32x16_bf16 fragment maps (the V5 attention structure
+    builds on them) vs torch.matmul on asymmetric data."""
+    ext = _ext()
+    torch.manual_seed(0)
+    a = torch.randn(32, 16, device=DEV).bfloat16()
+    b = torch.randn(16, 32, device=DEV).bfloat16()
+    d = ext.mfma32_probe(a.contiguous(), b.contiguous())
+    ref = a.float() @ b.float()
+    assert_close_bf16(d, ref, atol=5e-2, rtol=5e-2)
+
+
+@pytest.mark.parametrize("hq,hkv,d", [(4, 4, 128), (8, 2, 128), (4, 1, 64)])
+def test_attn_v5_parity(hq, hkv, d):
+    """V5 (8-wave swapped-QK^T, in-register softmax) vs the fp32 torch
+    reference — same harness as the V4 prefill parity test, plus the LSE
+    output checked against logsumexp."""
+    from substratus_amd import _C
+    from substratus_amd.ops import ref
+    torch.manual_seed(8)
+    lens = [1, 63, 64, 200, 300, 513]
+    t = sum(lens)
+    starts = [0]
+    for n in lens:
+        starts.append(starts[-1] + n)
+    q = torch.randn(t, hq, d, device=DEV).bfloat16()
+    k = torch.randn(t, hkv, d, device=DEV).bfloat16()
+    v = torch.randn(t, hkv, d, device=DEV).bfloat16()
+    ss = torch.tensor(starts, dtype=torch.int32)
+    scale = 1.0 / math.sqrt(d)
+    out = torch.empty_like(q)
+    lse = torch.empty(t, hq, dtype=torch.float32, device=DEV)
+    _C.attn_v5(out, q.contiguous(), k.contiguous(), v.contiguous(),
+               ss.to(DEV), max(lens), scale, lse)
+    want = ref.attn_prefill(q.cpu(), k.cpu(), v.cpu(), ss, scale)
+    assert_close_bf16(out, want, atol=3e-2, rtol=3e-2)
+    # LSE (log2 domain, scale folded) vs torch logsumexp for one config
+    qf = q.float().cpu()
+    kf = k.float().cpu()
+    rep = hq // hkv
+    for b0, n in ((starts[3], lens[3]),):
+        sm = torch.einsum("qhd,khd->hqk", qf[b0:b0 + n],
+                          kf[b0:b0 + n].repeat_interleave(rep, 1)) * scale
+        mask = torch.full((n, n), float("-inf")).triu(1)
+        want_lse = (sm + mask).logsumexp(dim=-1) / math.log(2)  # [h, q]
+        got = lse[b0:b0 + n].cpu().t()
+        torch.testing.assert_close(got, want_lse, atol=2e-2, rtol=2e-2)
+
+
+def test_attn_v5_ragged_tail_rows():
+    """A tile whose wave has only ONE live q row: dead lanes must not
+    contaminate live lanes' output through shared MFMA fragments (the
+    A-row of every lane feeds every output column)."""
+    from substratus_amd import _C
+    from substratus_amd.ops import ref
+    torch.manual_seed(44)
+    for n in (257, 289, 513):     # one row into a fresh 32-row wave/tile
+        q = torch.randn(n, 4, 128, device=DEV).bfloat16()
+        k = torch.randn(n, 4, 128, device=DEV).bfloat16()
+        v = torch.randn(n, 4, 128, device=DEV).bfloat16()
+        ss = torch.tensor([0, n], dtype=torch.int32)
+        out = torch.empty_like(q)
+        _C.attn_v5(out, q, k, v, ss.to(DEV), n, 128 ** -0.5, None)
+        want = ref.attn_prefill(q.cpu(), k.cpu(), v.cpu(), ss, 128 ** -0.5)
+        assert_close_bf16(out, want, atol=3e-2, rtol=3e-2)
