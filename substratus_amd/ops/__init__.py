@@ -178,8 +178,12 @@ class _FlashAttnTrain(torch.autograd.Function):
         dk32 = torch.zeros(k.shape, dtype=torch.float32, device=k.device)
         dv32 = torch.zeros(v.shape, dtype=torch.float32, device=v.device)
         delta = torch.empty_like(lse)
-        ext.attn_train_bwd(dq, dk32, dv32, q, k, v, out, dout, lse, delta,
-                           seq_start, ctx.max_len, ctx.scale)
+        if os.environ.get("SUB_ATTN_V5", "1") != "0":
+            ext.attn_train_bwd_v5(dq, dk32, dv32, q, k, v, out, dout, lse,
+                                  delta, seq_start, ctx.max_len, ctx.scale)
+        else:
+            ext.attn_train_bwd(dq, dk32, dv32, q, k, v, out, dout, lse,
+                               delta, seq_start, ctx.max_len, ctx.scale)
         return dq, dk32.to(k.dtype), dv32.to(v.dtype), None, None, None
 
 

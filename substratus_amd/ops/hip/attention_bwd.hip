@@ -513,3 +513,23 @@ void launch_attn_bwd(bf16_t* dq, float* dk, float* dv, const bf16_t* q,
 }
 
 bool attn_bwd_supported(int d) { return d == 128 || d == 64; }
+
+// dkv alone (the mixed-dispatch backward pairs the V5 dq kernel with THIS
+// dkv: the V5-structure dkv measured 2.3x SLOWER — 2104 vs 918 us/call on
+// falcon MQA — despite winning on dq/fwd; see attention_v5.hip STATUS)
+void launch_attn_bwd_dkv_only(float* dk, float* dv, const bf16_t* q,
+                              const bf16_t* k, const bf16_t* v,
+                              const bf16_t* dout, const float* lse,
+                              const float* delta, const int* seq_start,
+                              int num_seqs, int max_len, int hq, int hkv,
+                              int d, float scale, hipStream_t stream) {
+  const int tiles = (max_len + BW_TILE - 1) / BW_TILE;
+  dim3 block(BW_WAVES * WAVE);
+  dim3 grid_kv(num_seqs, hq, tiles);
+  if (d == 128)
+    attnbwd::attn_bwd_dkv_kernel<128><<<grid_kv, block, 0, stream>>>(
+        dk, dv, q, k, v, dout, lse, delta, seq_start, hq, hkv, scale);
+  else
+    attnbwd::attn_bwd_dkv_kernel<64><<<grid_kv, block, 0, stream>>>(
+        dk, dv, q, k, v, dout, lse, delta, seq_start, hq, hkv, scale);
+}

@@ -21,6 +21,9 @@ void launch_attn_prefill(bf16_t*, const bf16_t*, const bf16_t*, const bf16_t*, c
 bool attn_prefill_supported(int);
 void launch_attn_v5(bf16_t*, const bf16_t*, const bf16_t*, const bf16_t*, const int*, int, int, int, int, int, float, hipStream_t, float*);
 bool attn_v5_supported(int);
+void launch_attn_bwd_dq_v5(bf16_t*, const bf16_t*, const bf16_t*, const bf16_t*, const bf16_t*, const float*, const float*, const int*, int, int, int, int, int, float, hipStream_t);
+void launch_attn_bwd_dkv_v5(float*, float*, const bf16_t*, const bf16_t*, const bf16_t*, const bf16_t*, const float*, const float*, const int*, int, int, int, int, int, float, hipStream_t);
+void launch_attn_bwd_dkv_only(float*, float*, const bf16_t*, const bf16_t*, const bf16_t*, const bf16_t*, const float*, const float*, const int*, int, int, int, int, int, float, hipStream_t);
 void launch_attn_bwd_delta(float*, const bf16_t*, const bf16_t*, long, int, hipStream_t);
 void launch_attn_bwd(bf16_t*, float*, float*, const bf16_t*, const bf16_t*, const bf16_t*, const bf16_t*, const float*, const float*, const int*, int, int, int, int, int, float, hipStream_t);
 bool attn_bwd_supported(int);
@@ -282,6 +285,50 @@ void attn_train_bwd(torch::Tensor dq, torch::Tensor dk, torch::Tensor dv,
                   hkv, d, (float)scale, stream());
   post_launch_check("attn_train_bwd");
 }
+void attn_train_bwd_v5(torch::Tensor dq, torch::Tensor dk,
+                       torch::Tensor dv, torch::Tensor q, torch::Tensor k,
+                       torch::Tensor v, torch::Tensor out,
+                       torch::Tensor dout, torch::Tensor lse,
+                       torch::Tensor delta, torch::Tensor seq_start,
+                       long max_len, double scale) {
+  check_bf16(dq, "dq");
+  TORCH_CHECK(dk.scalar_type() == torch::kFloat &&
+              dv.scalar_type() == torch::kFloat, "dk/dv f32");
+  check_bf16(q, "q"); check_bf16(k, "k"); check_bf16(v, "v");
+  check_bf16(out, "out"); check_bf16(dout, "dout");
+  TORCH_CHECK(lse.scalar_type() == torch::kFloat &&
+              delta.scalar_type() == torch::kFloat, "lse/delta f32");
+  TORCH_CHECK(seq_start.scalar_type() == torch::kInt, "seq_start int32");
+  int hq = q.size(1), d = q.size(2), hkv = k.size(1);
+  int num_seqs = seq_start.size(0) - 1;
+  TORCH_CHECK(attn_v5_supported(d), "unsupported head_dim ", d);
+  TORCH_CHECK(hq % hkv == 0, "hq must be divisible by hkv");
+  launch_attn_bwd_delta(delta.data_ptr<float>(), bfc(dout), bfc(out),
+                        (long)q.size(0) * hq, d, stream());
+  launch_attn_bwd_dq_v5(bf(dq), bfc(q), bfc(k), bfc(v), bfc(dout),
+                        lse.data_ptr<float>(), delta.data_ptr<float>(),
+                        seq_start.data_ptr<int>(), num_seqs, (int)max_len,
+                        hq, hkv, d, (float)scale, stream());
+  // dkv: the 16-row kernel (the V5-structure dkv measured 2.3x slower;
+  // SUB_ATTN_DKV_V5=1 re-enables it for measurement)
+  const char* e = getenv("SUB_ATTN_DKV_V5");
+  if (e && e[0] == '1')
+    launch_attn_bwd_dkv_v5(dk.data_ptr<float>(), dv.data_ptr<float>(),
+                           bfc(q), bfc(k), bfc(v), bfc(dout),
+                           lse.data_ptr<float>(), delta.data_ptr<float>(),
+                           seq_start.data_ptr<int>(), num_seqs,
+                           (int)max_len, hq, hkv, d, (float)scale,
+                           stream());
+  else
+    launch_attn_bwd_dkv_only(dk.data_ptr<float>(), dv.data_ptr<float>(),
+                             bfc(q), bfc(k), bfc(v), bfc(dout),
+                             lse.data_ptr<float>(), delta.data_ptr<float>(),
+                             seq_start.data_ptr<int>(), num_seqs,
+                             (int)max_len, hq, hkv, d, (float)scale,
+                             stream());
+  post_launch_check("attn_train_bwd_v5");
+}
+
 
 void attn_prefill_probe(long variant, torch::Tensor out, torch::Tensor q,
                         torch::Tensor k, torch::Tensor v,
@@ -512,6 +559,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("quant_gemv_k", &quant_gemv_k);
   m.def("mfma32_probe", &mfma32_probe);
   m.def("attn_v5", &attn_v5);
+  m.def("attn_train_bwd_v5", &attn_train_bwd_v5);
   m.def("dequant_k", &dequant_k);
   m.def("mfma_gemm_pick_splitk", &mfma_gemm_pick_splitk);
   m.def("attn_train_bwd", &attn_train_bwd);
