@@ -212,3 +212,42 @@ def test_server_reconcile_through_realkube(realkube, tmp_path):
         mgr._enqueue(("Server", "s1", "default"))
         mgr.run_once()
     assert apiv1.get_status_ready(rk.get("Server", "s1"))
+
+
+@pytest.mark.timeout(120)
+def test_dataset_and_notebook_through_realkube(realkube, tmp_path):
+    """Dataset (loader Job -> Ready) and Notebook (pod, suspend reap)
+    through the kubernetes-client surface."""
+    rk, store = realkube
+    cloud = new_cloud({"CLOUD": "kind", "CLUSTER_NAME": "rt",
+                       "BUCKET_DIR": str(tmp_path / "bucket")})
+    from substratus_amd.controller.manager import Manager
+    mgr = Manager(rk, cloud, FakeSCIClient())
+
+    rk.create(apiv1.new_object("Dataset", "d1", {
+        "image": "loader:v1", "params": {"urls": "file:///x.jsonl"}}))
+    for _ in range(3):
+        mgr._enqueue(("Dataset", "d1", "default"))
+        mgr.run_once()
+    job = rk.try_get("Job", "d1-data-loader")
+    assert job is not None and job["spec"]["backoffLimit"] == 2
+    rk.patch("Job", "d1-data-loader",
+             {"status": {"conditions": [{"type": "Complete",
+                                         "status": "True"}]}})
+    for _ in range(3):
+        mgr._enqueue(("Dataset", "d1", "default"))
+        mgr.run_once()
+    assert apiv1.get_status_ready(rk.get("Dataset", "d1"))
+
+    rk.create(apiv1.new_object("Notebook", "nb1", {"image": "nb:v1"}))
+    for _ in range(3):
+        mgr._enqueue(("Notebook", "nb1", "default"))
+        mgr.run_once()
+    pod = rk.try_get("Pod", "nb1-notebook")
+    assert pod is not None
+    # suspend reaps the pod (reference notebook_controller suspend path)
+    rk.patch("Notebook", "nb1", {"spec": {"suspend": True}})
+    for _ in range(3):
+        mgr._enqueue(("Notebook", "nb1", "default"))
+        mgr.run_once()
+    assert rk.try_get("Pod", "nb1-notebook") is None
