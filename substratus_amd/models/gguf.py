@@ -111,7 +111,7 @@ def _pack_scale_min_k4(sc: torch.Tensor, mn: torch.Tensor) -> torch.Tensor:
     j<4:  q[j]   = sc&63 | (upper 2 bits of sc[j+4] later)
     The exact inverse of _unpack_scale_min_k4."""
     B = sc.shape[0]
-    out = torch.zeros(B, 12, dtype=torch.uint8)
+    out = torch.zeros(B, 12, dtype=torch.uint8, device=sc.device)
     out[:, 0:4] = (sc[:, 0:4] & 63) | (((sc[:, 4:8] >> 4) & 0x3) << 6)
     out[:, 4:8] = (mn[:, 0:4] & 63) | (((mn[:, 4:8] >> 4) & 0x3) << 6)
     out[:, 8:12] = (sc[:, 4:8] & 0xF) | ((mn[:, 4:8] & 0xF) << 4)
@@ -122,7 +122,7 @@ def _unpack_scale_min_k4(scales: torch.Tensor) -> tuple[torch.Tensor, torch.Tens
     """scales: [B, 12] uint8 -> (sc, mn) [B, 8] uint8 — llama.cpp
     get_scale_min_k4 for all 8 sub-blocks."""
     q = scales.to(torch.int32)
-    sc = torch.empty(q.shape[0], 8, dtype=torch.int32)
+    sc = torch.empty(q.shape[0], 8, dtype=torch.int32, device=scales.device)
     mn = torch.empty_like(sc)
     sc[:, 0:4] = q[:, 0:4] & 63
     mn[:, 0:4] = q[:, 4:8] & 63
@@ -150,7 +150,8 @@ def quantize_q4_K(w: torch.Tensor) -> torch.Tensor:
     eff_m = dmin * mn.float()
     q = torch.round((blocks + eff_m[:, :, None]) / eff_d[:, :, None])
     q = q.clamp(0, 15).to(torch.uint8)       # [B, 8, 32]
-    out = torch.zeros(B, Q4_K_BLOCK_BYTES, dtype=torch.uint8)
+    out = torch.zeros(B, Q4_K_BLOCK_BYTES, dtype=torch.uint8,
+                      device=w.device)
     out[:, 0:2] = d.to(torch.float16).view(torch.uint8).reshape(B, 2)
     out[:, 2:4] = dmin.to(torch.float16).view(torch.uint8).reshape(B, 2)
     out[:, 4:16] = _pack_scale_min_k4(sc, mn)
@@ -189,13 +190,14 @@ def quantize_q6_K(w: torch.Tensor) -> torch.Tensor:
     eff = d * sc.float().clamp(min=1.0)      # sc >= 0 by construction
     q = torch.round(blocks / eff[:, :, None]).clamp(-32, 31).to(torch.int32)
     q = (q + 32).reshape(B, 2, 128)          # [B, half, 128] in [0, 63]
-    out = torch.zeros(B, Q6_K_BLOCK_BYTES, dtype=torch.uint8)
+    out = torch.zeros(B, Q6_K_BLOCK_BYTES, dtype=torch.uint8,
+                      device=w.device)
     # per 128-half: ql = low 4 bits of (l, l+64) pairs? llama.cpp layout:
     # element groups within a half: e[l], e[l+32], e[l+64], e[l+96] for
     # l in [0,32) pack as ql[l]=lo4(e0)|lo4(e2)<<4, ql[l+32]=lo4(e1)|lo4(e3)<<4,
     # qh[l] = hi2(e0) | hi2(e1)<<2 | hi2(e2)<<4 | hi2(e3)<<6
     e = q.reshape(B, 2, 4, 32)               # [B, half, quarter, 32]
-    ql = torch.zeros(B, 2, 64, dtype=torch.int32)
+    ql = torch.zeros(B, 2, 64, dtype=torch.int32, device=w.device)
     ql[:, :, 0:32] = (e[:, :, 0] & 0xF) | ((e[:, :, 2] & 0xF) << 4)
     ql[:, :, 32:64] = (e[:, :, 1] & 0xF) | ((e[:, :, 3] & 0xF) << 4)
     qh = (e[:, :, 0] >> 4) | ((e[:, :, 1] >> 4) << 2) |          ((e[:, :, 2] >> 4) << 4) | ((e[:, :, 3] >> 4) << 6)

@@ -113,7 +113,7 @@ class QuantLinear(nn.Module):
         # CPU quantize would take minutes)
         w = lin.weight.detach().float()
         qfn = QUANT_CODECS[quant][2]
-        packed = qfn(w.cpu()) if quant.endswith("_K") else qfn(w)
+        packed = qfn(w)
         rp = repack_k if quant.endswith("_K") else repack
         scales, qs = rp(packed, w.shape[1], quant)
         dev = lin.weight.device
