@@ -60,7 +60,7 @@ class EngineConfig:
         return cls(**d)
 
 
-@dataclass
+@dataclass(slots=True)
 class StepOutput:
     seq_id: int
     token: int

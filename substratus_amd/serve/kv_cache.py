@@ -66,7 +66,7 @@ class PagedKVCache:
         return max(bytes_budget // per_block, 1)
 
 
-@dataclass
+@dataclass(slots=True)
 class SequenceState:
     """Server-side state of one running request."""
     seq_id: int
