@@ -29,12 +29,13 @@
 // consistent because read and write both key on the LOCAL row index).
 //
 // STATUS (round 3 conclusion, profiles/mgemm_r02.md): correct (parity vs
-// fp32 matmul on all five llama2-7b decode shapes at M=256/512) and at its
-// structure's measured ceiling (~1.1 PF normalized for grid fill), 1.5-2.3x
-// v2 — but hipBLASLt's MT256/stream-k solutions run 1.3-4.1 TB/s on the
-// same shapes, i.e. at the ~1.33 PF ceiling of the 8-phase counted-vmcnt
-// schedule itself. hipBLASLt stays the production path; SUB_MGEMM=1 routes
-// through this kernel for measurement.
+// fp32 matmul on all five llama2-7b decode shapes at M=256/512). The
+// counted-vmcnt upgrade (vmcnt(GLDS_PER_TILE) + raw barrier instead of
+// the __syncthreads drain) added +12-13% on the 256-tile
+// (qkv 612 -> 687 TF, gate-up 740 -> 839, logits 984 -> 1011), but
+// hipBLASLt's MT256/stream-k solutions still run 1.3-2.9 TB/s on the
+// same shapes (0.6-0.75x everywhere). hipBLASLt stays the production
+// path; SUB_MGEMM=1 routes through this kernel for measurement.
 #include "common.h"
 
 typedef __attribute__((ext_vector_type(8))) short short8;
@@ -139,12 +140,27 @@ void mfma_gemm_kernel(bf16_t* __restrict__ y,      // [M, N]  (splitk == 1)
     }
   };
 
+  constexpr int GLDS_PER_TILE =
+      (APIECES + BPIECES) / NWAVES;   // per-thread glds per K-tile
+
   stage(0, t0);
-  __syncthreads();   // carries the vmcnt(0) drain of the in-flight glds
 
   for (int t = t0; t < t1; ++t) {
     const int cur = (t - t0) & 1;
-    if (t + 1 < t1) stage(cur ^ 1, t + 1);
+    // Counted-vmcnt pipeline (guide T3+T4: counted-vs-drain0 is +38-73%):
+    // issue the NEXT tile's glds, then wait only until THIS tile's have
+    // landed (the next tile's stay in flight across the raw barrier — a
+    // plain __syncthreads would emit vmcnt(0) and drain them). One
+    // barrier per tile is overwrite-safe: the barrier at the END of
+    // iteration t-1 means every wave finished reading buf^1 before any
+    // wave's stage(t+1) rewrites it.
+    if (t + 1 < t1) {
+      stage(cur ^ 1, t + 1);
+      asm volatile("s_waitcnt vmcnt(%0)" :: "i"(GLDS_PER_TILE) : "memory");
+    } else {
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    }
+    __builtin_amdgcn_s_barrier();
 
     // B-fragments for the wave's n-tiles × 2 k-steps (held whole tile)
     const uint4* btile = &tiles[cur][BM * MG_BK * 2 / 16];
@@ -174,8 +190,9 @@ void mfma_gemm_kernel(bf16_t* __restrict__ y,      // [M, N]  (splitk == 1)
             af1, bf[nf][1], acc[fm][nf], 0, 0, 0);
       }
     }
-    __syncthreads();  // drains the next tile's glds; releases cur for reuse
+    __builtin_amdgcn_s_barrier();   // end-of-iter: releases cur for reuse
   }
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");  // drain before stores
 
   // ---- epilogue ---------------------------------------------------------
   #pragma unroll
@@ -232,14 +249,12 @@ bool mfma_gemm_supported(int m, int n, int k) {
 // Pick the biggest tile whose grid still fills the chip; env
 // SUB_MGEMM_FORCE={0,1} overrides for A/B measurement.
 static int pick_cfg(int m, int n) {
+  // With the counted-vmcnt pipeline the 256-tile wins even underfilled
+  // (gate-up: 839 TF at 172 blocks vs 641 TF on a 688-block 128 grid)
   const int t256 = (n % 256 == 0)
       ? ((m + 255) / 256) * (n / 256) : 0;
-  if (t256 >= 192) return 0;
-  const int t128 = ((m + 127) / 128) * (n / 128);
-  // 128-tiles run 2 blocks/CU -> want >= ~384 for full overlap, but any
-  // fill beats an under-filled 256 grid
-  if (t128 >= t256 * 3) return 1;
-  return 0;
+  if (t256 >= 32) return 0;
+  return 1;
 }
 
 int mfma_gemm_splitk(int m, int n, int k) {
