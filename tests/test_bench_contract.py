@@ -82,3 +82,21 @@ def test_bench_torchrun_tp2_gloo():
     r = _last_json(out.stdout)
     assert r["config"]["parallelism"] == "tp2"
     assert r["config"]["global_batch"] == 2  # ONE shared batch, not 2x
+
+
+@pytest.mark.timeout(600)
+def test_bench_train_torchrun_2rank_gloo():
+    """Train bench under the driver's N>1 launch shape — exercises the
+    ZeRO-1 path (reduce-scatter/shard-AdamW/all-gather) end to end through
+    bench.py on gloo."""
+    out = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+         "--master-port", "29914", "bench.py", "--gpus", "2", "--mode",
+         "train", "--model", "llama-tiny", "--steps", "2", "--warmup", "1",
+         "--seq-len", "32"],
+        cwd=ROOT, capture_output=True, text=True, timeout=580)
+    assert out.returncode == 0, out.stderr[-2000:]
+    r = _last_json(out.stdout)
+    assert r["metric"] == "fine_tune_tokens_per_s"
+    assert r["n_gpus"] == 2

@@ -83,3 +83,46 @@ def _hf_tok():
 def test_hf_byte_level_tokenizer_roundtrip(text):
     t = _hf_tok()
     assert t.decode(t.encode(text)) == text
+
+
+@given(st.integers(0, 2**32 - 1),
+       st.sampled_from(["q4_K", "q6_K"]),
+       st.integers(1, 4))
+@settings(max_examples=20, deadline=None)
+def test_kquant_codec_roundtrip_bounds(seed, quant, rows):
+    """K-quant codecs: dequant(quant(w)) stays within the codec's step
+    bound for every sub-block (q4_K: d*sc covers [min,max] in 15 steps;
+    q6_K: 31 steps of d*sc per 16-group)."""
+    import torch
+    from substratus_amd.models.gguf import QUANT_CODECS
+    g = torch.Generator().manual_seed(seed)
+    w = torch.randn(rows, 512, generator=g) * 0.1
+    _, _, qf, df = QUANT_CODECS[quant]
+    back = df(qf(w), 512)
+    if quant == "q4_K":
+        grp = w.reshape(-1, 32)
+        span = (grp.amax(1) - grp.amin(1)).clamp(min=1e-6)
+        step = span / 15.0
+        err = (back.reshape(-1, 32) - grp).abs().amax(1)
+        # quantization error <= one step (+ 6-bit scale/min rounding slack)
+        assert bool((err <= step * 1.2 + 1e-3).all())
+    else:
+        grp = w.reshape(-1, 16)
+        step = grp.abs().amax(1).clamp(min=1e-6) / 31.0
+        err = (back.reshape(-1, 16) - grp).abs().amax(1)
+        assert bool((err <= step * 1.2 + 1e-3).all())
+
+
+@given(st.integers(0, 2**32 - 1), st.sampled_from(["q4_K", "q6_K"]))
+@settings(max_examples=10, deadline=None)
+def test_kquant_repack_involution(seed, quant):
+    """repack_k (SoA for the GPU kernels) and unrepack_k are exact
+    inverses on random valid blocks."""
+    import torch
+    from substratus_amd.models.gguf import QUANT_CODECS
+    from substratus_amd.models.quant import repack_k, unrepack_k
+    g = torch.Generator().manual_seed(seed)
+    w = torch.randn(4, 768, generator=g)
+    packed = QUANT_CODECS[quant][2](w)
+    s, qs = repack_k(packed, 768, quant)
+    assert torch.equal(unrepack_k(s, qs, 768, quant), packed)

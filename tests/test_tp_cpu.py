@@ -143,3 +143,25 @@ def test_shard_reconstruction_roundtrip():
         else:
             for p in parts:
                 assert torch.equal(p, t), name
+
+
+def test_tp_forces_eager_decode(monkeypatch):
+    """With tp_size>1 the engine must NOT take the hipGraph path (RCCL
+    collectives inside stream capture are unvalidated on ROCm) — the
+    guard is in Engine._decode."""
+    import torch
+    from substratus_amd.parallel import comm
+    from substratus_amd.serve.engine import Engine, EngineConfig
+    eng = Engine(EngineConfig(model="llama-tiny", dtype=torch.float32,
+                              max_batch=2, max_seq_len=64, kv_blocks=16,
+                              use_graphs=True), device="cpu")
+    eng.add_request([1, 2, 3], max_new_tokens=3)
+    # pretend we're TP rank 0 of 2 (no process group needed for the guard)
+    monkeypatch.setattr(comm, "_STATE",
+                        comm.ParallelState(initialized=True, tp_size=2))
+    called = {"graph": False}
+    monkeypatch.setattr(eng, "_decode_graph",
+                        lambda: called.__setitem__("graph", True) or [])
+    while eng.has_work():
+        eng.step()
+    assert called["graph"] is False
