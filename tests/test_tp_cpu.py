@@ -156,9 +156,11 @@ def test_tp_forces_eager_decode(monkeypatch):
                               max_batch=2, max_seq_len=64, kv_blocks=16,
                               use_graphs=True), device="cpu")
     eng.add_request([1, 2, 3], max_new_tokens=3)
-    # pretend we're TP rank 0 of 2 (no process group needed for the guard)
+    # pretend we're TP rank 0 of 2 (no process group: stub the collective)
     monkeypatch.setattr(comm, "_STATE",
                         comm.ParallelState(initialized=True, tp_size=2))
+    from substratus_amd.parallel import tp as tpmod
+    monkeypatch.setattr(tpmod, "tp_all_reduce_", lambda t: t)
     called = {"graph": False}
     monkeypatch.setattr(eng, "_decode_graph",
                         lambda: called.__setitem__("graph", True) or [])
