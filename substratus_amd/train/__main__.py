@@ -41,6 +41,8 @@ def main() -> None:
                                    params.get("lr_schedule", "constant"))),
         min_lr=float(params.get("min_lr", 0.0)),
         micro_batch=int(params.get("micro_batch", 4)),
+        grad_accum=int(params.get("gradient_accumulation_steps",
+                                  params.get("grad_accum", 1))),
         seq_len=int(params.get("seq_len", params.get("max_length", 2048))),
         num_train_epochs=int(params.get("num_train_epochs", 1)),
         save_steps=int(params.get("save_steps", 0)),

@@ -28,12 +28,17 @@ class GradBuckets:
         self._pending_bytes = 0
         self._handles: list = []
         self._hooks = []
+        # while True, grads accumulate locally (gradient accumulation);
+        # hooks no-op and the final micro-batch's backward flushes the sum
+        self.accumulating = False
         if self.st.dp_size > 1:
             for p in self.params:
                 h = p.register_post_accumulate_grad_hook(self._on_grad)
                 self._hooks.append(h)
 
     def _on_grad(self, p: torch.Tensor) -> None:
+        if self.accumulating:
+            return
         g = p.grad
         if g is None:
             return
@@ -109,6 +114,7 @@ class ZeroShardedDDP:
         self.step_num = 0
         self.params = [p for p in params if p.requires_grad]
         self.bucket_bytes = bucket_bytes
+        self.accumulating = False
         self.buckets: list[dict] = []
         self._param_bucket: dict[int, int] = {}
         self._handles: list = []
@@ -162,6 +168,8 @@ class ZeroShardedDDP:
 
     # ------------------------------------------------------------ backward
     def _on_grad(self, p: torch.Tensor) -> None:
+        if self.accumulating:
+            return
         b = self.buckets[self._param_bucket[id(p)]]
         b["ready"].add(id(p))
         if len(b["ready"]) == len(b["params"]):

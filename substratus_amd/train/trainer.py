@@ -154,13 +154,27 @@ class Trainer:
     # ------------------------------------------------------------- one step
     def train_step(self, tokens: torch.Tensor) -> float:
         """tokens: [micro_batch, seq_len+1] int64 (inputs ‖ shifted targets).
-        Returns the scalar loss."""
-        inputs = tokens[:, :-1].to(self.device)
-        targets = tokens[:, 1:].to(self.device).reshape(-1)
-        logits = self.model.forward_train(inputs)
-        flat = logits.reshape(-1, logits.shape[-1])
-        loss, dlogits = ops.cross_entropy_fwd_bwd(flat.detach(), targets)
-        flat.backward(gradient=dlogits)
+        Returns the scalar loss (mean over accumulation chunks).
+
+        cfg.grad_accum > 1 (HF gradient_accumulation_steps): the batch is
+        split into that many forward/backward chunks whose gradients
+        accumulate locally (the DDP hooks hold their reduce until the
+        final chunk), then one optimizer step runs."""
+        accum = max(int(self.cfg.grad_accum), 1)
+        chunks = tokens.chunk(accum) if accum > 1 else [tokens]
+        ddp = self.zero if self.use_zero else self.buckets
+        losses = []
+        for ci, chunk in enumerate(chunks):
+            if ddp is not None:
+                ddp.accumulating = ci < len(chunks) - 1
+            inputs = chunk[:, :-1].to(self.device)
+            targets = chunk[:, 1:].to(self.device).reshape(-1)
+            logits = self.model.forward_train(inputs)
+            flat = logits.reshape(-1, logits.shape[-1])
+            loss, dlogits = ops.cross_entropy_fwd_bwd(flat.detach(), targets)
+            flat.backward(gradient=dlogits / len(chunks))
+            losses.append(float(loss))
+        loss = sum(losses) / len(losses)
         lr = self.lr_at(self.global_step + 1)
         if self.use_zero:
             self.zero.finalize()
@@ -186,7 +200,8 @@ class Trainer:
             self.opt.step(self.cfg, grad_scale=gscale, lr=lr)
             self.opt.zero_grad()
         self.global_step += 1
-        self.tokens_seen += inputs.numel() * max(self.st.dp_size, 1)
+        self.tokens_seen += (tokens.shape[0] * (tokens.shape[1] - 1) *
+                             max(self.st.dp_size, 1))
         return float(loss)
 
     # ---------------------------------------------------------- checkpoints

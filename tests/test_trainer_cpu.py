@@ -283,3 +283,26 @@ def test_lr_schedule():
     assert tr.lr_at(8) == pytest.approx(1e-4 + (1e-3 - 1e-4) * 0.5)
     cfg.lr_schedule = "constant"
     assert tr.lr_at(9) == pytest.approx(1e-3)
+
+
+def test_grad_accum_matches_full_batch(tmp_path):
+    """grad_accum=2 over a 4-row batch lands on the same params as one
+    4-row step (fp32, same data; hooks must hold the reduce until the
+    final chunk)."""
+    torch.manual_seed(0)
+    g = torch.Generator().manual_seed(99)
+    batch = torch.randint(0, 512, (4, 17), generator=g)
+    params = {}
+    for accum in (1, 2):
+        torch.manual_seed(0)
+        cfg = TrainConfig(model="llama-tiny", micro_batch=4, seq_len=16,
+                          dtype=torch.float32, grad_accum=accum,
+                          artifacts_dir=str(tmp_path / f"a{accum}"),
+                          lr=1e-3)
+        tr = Trainer(cfg, device="cpu")
+        tr.train_step(batch)
+        params[accum] = {k: v.detach().clone()
+                        for k, v in tr.model.state_dict().items()}
+    for k in params[1]:
+        torch.testing.assert_close(params[1][k], params[2][k],
+                                   atol=1e-5, rtol=1e-4)
