@@ -189,84 +189,79 @@ void attn_v5_kernel(bf16_t* __restrict__ out,      // [T, HQ, D]
     stage_load(kc + 1);              // loads fly over this chunk's MFMAs
     const int kbase = kc * V5_KCHUNK;
     if (wave_live && kbase <= min(q0 + 31, n - 1)) {
+      // ---- S^T = K·Q^T over the WHOLE 64-key chunk (both 32-subtiles
+      // back-to-back: one online-softmax step per chunk instead of two —
+      // halves the rescale/stat passes and gives the scheduler two
+      // independent MFMA chains)
+      float16v sacc[2];
       #pragma unroll
       for (int sc = 0; sc < 2; ++sc) {
-        // ---- S^T = K·Q^T over this 32-key subtile -----------------------
-        float16v sacc = (float16v){};
+        sacc[sc] = (float16v){};
         #pragma unroll
         for (int kk = 0; kk < KK; ++kk) {
           const int key = sc * 32 + lo5;
           const uint4 ka4 = k_tile[cur][swz<D * 2>(key, kk * 2 + hi)];
-          sacc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
-              *reinterpret_cast<const short8*>(&ka4), qreg[kk], sacc,
+          sacc[sc] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+              *reinterpret_cast<const short8*>(&ka4), qreg[kk], sacc[sc],
               0, 0, 0);
         }
-        // ---- online softmax (32 keys), all in-register ------------------
-        float sv[16];
-        float mloc = -INFINITY;
+      }
+      // ---- one in-register online-softmax step over 64 keys ------------
+      float sv[2][16];
+      float mloc = -INFINITY;
+      #pragma unroll
+      for (int sc = 0; sc < 2; ++sc) {
         #pragma unroll
         for (int r = 0; r < 16; ++r) {
           const int kpos = kbase + sc * 32 + (r & 3) + 8 * (r >> 2) + 4 * hi;
           const bool live = (kpos <= qpos) && (kpos < n) && (qpos < n);
-          sv[r] = live ? sacc[r] * scale2 : -INFINITY;
-          mloc = fmaxf(mloc, sv[r]);
+          sv[sc][r] = live ? sacc[sc][r] * scale2 : -INFINITY;
+          mloc = fmaxf(mloc, sv[sc][r]);
         }
-        mloc = fmaxf(mloc, __shfl_xor(mloc, 32, WAVE));
-        // All control flow below is WAVE-UNIFORM (__all / compile-time):
-        // an MFMA under a divergent branch reads operand fragments from
-        // EXEC-masked lanes (a lane's A-row serves every lane's output),
-        // so dead-q lanes must run the same instruction stream with their
-        // P values guarded to zero — not branch around it.
-        //
-        // defer-max (guide T13, +5%): skip the O-wide rescale while the
-        // running max grows by <= THR (log2 domain: P bounded by 2^THR =
-        // 256, safe in f32 accumulation). The decision covers THIS
-        // subtile's P entirely (textbook order: exponentiate after the
-        // decision, l-update with the same alpha) — the T13 hazard of
-        // splitting a pending tile's P·V does not arise. Dead lanes
-        // (mloc == -inf) read as "never rescale".
-        constexpr float THR = 8.0f;
-        {
-          float alpha = 1.0f;
-          if (!__all((mloc == -INFINITY) || (mloc - m <= THR))) {
-            const float mn = fmaxf(m, mloc);
-            alpha = (m == -INFINITY) ? 0.f
-                                     : __builtin_amdgcn_exp2f(m - mn);
-            m = mn;
+      }
+      mloc = fmaxf(mloc, __shfl_xor(mloc, 32, WAVE));
+      // wave-uniform control flow only (see the divergent-MFMA note in
+      // the git history of the two-subtile version); defer-max as before
+      constexpr float THR = 8.0f;
+      {
+        float alpha = 1.0f;
+        if (!__all((mloc == -INFINITY) || (mloc - m <= THR))) {
+          const float mn = fmaxf(m, mloc);
+          alpha = (m == -INFINITY) ? 0.f
+                                   : __builtin_amdgcn_exp2f(m - mn);
+          m = mn;
+          #pragma unroll
+          for (int c = 0; c < DC; ++c)
             #pragma unroll
-            for (int c = 0; c < DC; ++c)
-              #pragma unroll
-              for (int r = 0; r < 16; ++r) o_t[c][r] *= alpha;
-          }
-          float rowsum = 0.f;
-          float p[16];
+            for (int r = 0; r < 16; ++r) o_t[c][r] *= alpha;
+        }
+        float rowsum = 0.f;
+        unsigned w[2][8];
+        #pragma unroll
+        for (int sc = 0; sc < 2; ++sc) {
           #pragma unroll
-          for (int r = 0; r < 16; ++r) {
-            p[r] = (sv[r] == -INFINITY)
-                       ? 0.f : __builtin_amdgcn_exp2f(sv[r] - m);
-            rowsum += p[r];
+          for (int i = 0; i < 8; ++i) {
+            float p0 = (sv[sc][2 * i] == -INFINITY) ? 0.f
+                : __builtin_amdgcn_exp2f(sv[sc][2 * i] - m);
+            float p1 = (sv[sc][2 * i + 1] == -INFINITY) ? 0.f
+                : __builtin_amdgcn_exp2f(sv[sc][2 * i + 1] - m);
+            rowsum += p0 + p1;
+            w[sc][i] = cvt_pk_bf16(p0, p1);
           }
-          rowsum += __shfl_xor(rowsum, 32, WAVE);
-          l = l * alpha + rowsum;
-          // ---- P -> PV B-frags: cvt_pk pairs + permlane32_swap ---------
-          // w[i] packs this lane's keys {4hi + 8*(i/2)*?}: see header map
-          unsigned w[8];
-          #pragma unroll
-          for (int i = 0; i < 8; ++i)
-            w[i] = cvt_pk_bf16(p[2 * i], p[2 * i + 1]);
-          short8 pb[2];   // B-frag per 16-key step g
+        }
+        rowsum += __shfl_xor(rowsum, 32, WAVE);
+        l = l * alpha + rowsum;
+        // ---- O^T += V^T · P over the chunk's four 16-key steps ---------
+        #pragma unroll
+        for (int sc = 0; sc < 2; ++sc) {
           #pragma unroll
           for (int g = 0; g < 2; ++g) {
             auto r02 = __builtin_amdgcn_permlane32_swap(
-                w[4 * g + 0], w[4 * g + 2], false, false);
+                w[sc][4 * g + 0], w[sc][4 * g + 2], false, false);
             auto r13 = __builtin_amdgcn_permlane32_swap(
-                w[4 * g + 1], w[4 * g + 3], false, false);
+                w[sc][4 * g + 1], w[sc][4 * g + 3], false, false);
             uint4 pw = make_uint4(r02[0], r13[0], r02[1], r13[1]);
-            pb[g] = *reinterpret_cast<short8*>(&pw);
-          }
-          // ---- O^T += V^T · P over the subtile's two 16-key steps ------
-          #pragma unroll
-          for (int g = 0; g < 2; ++g) {
+            const short8 pb = *reinterpret_cast<short8*>(&pw);
             #pragma unroll
             for (int c = 0; c < DC; ++c) {
               const short4v lo_ =
@@ -277,7 +272,7 @@ void attn_v5_kernel(bf16_t* __restrict__ out,      // [T, HQ, D]
               vt[0] = lo_[0]; vt[1] = lo_[1]; vt[2] = lo_[2]; vt[3] = lo_[3];
               vt[4] = hi_[0]; vt[5] = hi_[1]; vt[6] = hi_[2]; vt[7] = hi_[3];
               o_t[c] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
-                  vt, pb[g], o_t[c], 0, 0, 0);
+                  vt, pb, o_t[c], 0, 0, 0);
             }
           }
         }
