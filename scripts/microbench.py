@@ -190,6 +190,38 @@ def bench_attn_v5():
               f"({flops/min(t5)/1e12:.0f} TF)")
 
 
+def bench_attn_bwd():
+    """Backward kernels old-vs-V5, interleaved (falcon MQA + llama
+    shapes) — the dkv_v5 slowdown diagnosis harness."""
+    from substratus_amd import _C
+    shapes = [("falcon", 4, 71, 1, 64, 2048), ("llama", 4, 32, 32, 128, 2048)]
+    for name, B, hq, hkv, d, S in shapes:
+        torch.manual_seed(0)
+        t = B * S
+        q = (torch.rand(t, hq, d, device="cuda") * 2 - 1).bfloat16()
+        k = (torch.rand(t, hkv, d, device="cuda") * 2 - 1).bfloat16()
+        v = (torch.rand(t, hkv, d, device="cuda") * 2 - 1).bfloat16()
+        do = torch.randn_like(q)
+        ss = torch.arange(0, (B + 1) * S, S, dtype=torch.int32,
+                          device="cuda")
+        out = torch.empty_like(q)
+        lse = torch.empty(t, hq, dtype=torch.float32, device="cuda")
+        scale = d ** -0.5
+        _C.attn_v5(out, q, k, v, ss, S, scale, lse)
+        dq = torch.empty_like(q)
+        dk = torch.zeros(t, hkv, d, dtype=torch.float32, device="cuda")
+        dv = torch.zeros_like(dk)
+        delta = torch.empty_like(lse)
+        import os as _os
+        for tag, env in (("mixed", None), ("dkv_v5", "1")):
+            if env: _os.environ["SUB_ATTN_DKV_V5"] = env
+            else: _os.environ.pop("SUB_ATTN_DKV_V5", None)
+            dt = timeit(lambda: _C.attn_train_bwd_v5(
+                dq, dk, dv, q, k, v, out, do, lse, delta, ss, S, scale),
+                iters=20)
+            print(f"bwd {name} [{tag}]: {dt*1e6:.0f} us")
+
+
 def bench_mgemm(m=512):
     """mfma_gemm v3 (256x256 glds tile) vs hipBLASLt, interleaved rounds
     (guide rule 24) + parity vs fp32 matmul. Shapes: llama2-7b B=512 decode
@@ -267,6 +299,8 @@ if __name__ == "__main__":
         bench_wstream(m=64)
     if a.which in ("all", "attnv5"):
         bench_attn_v5()
+    if a.which == "attnbwd":
+        bench_attn_bwd()
     if a.which in ("all", "mgemm"):
         m = int(os.environ.get("MGEMM_M", "512"))
         bench_mgemm(m)

@@ -309,10 +309,13 @@ void attn_train_bwd_v5(torch::Tensor dq, torch::Tensor dk,
                         lse.data_ptr<float>(), delta.data_ptr<float>(),
                         seq_start.data_ptr<int>(), num_seqs, (int)max_len,
                         hq, hkv, d, (float)scale, stream());
-  // dkv: the 16-row kernel (the V5-structure dkv measured 2.3x slower;
-  // SUB_ATTN_DKV_V5=1 re-enables it for measurement)
+  // dkv dispatch by GQA group size: the V5-structure dkv WINS at grp==1
+  // (llama MHA: 1010 vs 1113 us) but loses 2x under MQA/GQA — its fewer,
+  // wider blocks collide harder on the per-kv-row atomic adds (falcon
+  // 71:1: 2477 vs 918 us). SUB_ATTN_DKV_V5=0/1 overrides.
   const char* e = getenv("SUB_ATTN_DKV_V5");
-  if (e && e[0] == '1')
+  const bool use_v5 = e ? (e[0] == '1') : (hq == hkv);
+  if (use_v5)
     launch_attn_bwd_dkv_v5(dk.data_ptr<float>(), dv.data_ptr<float>(),
                            bfc(q), bfc(k), bfc(v), bfc(dout),
                            lse.data_ptr<float>(), delta.data_ptr<float>(),
