@@ -108,6 +108,18 @@ class Engine:
             max_needed = (cfg.max_batch *
                           ((cfg.max_seq_len + cfg.block_size - 1) // cfg.block_size))
             num_blocks = min(num_blocks, max_needed)
+            if state().tp_size > 1:
+                # TP ranks must agree on the pool size: schedules are
+                # lockstep and preemption fires when the allocator runs
+                # out — a rank with less free HBM (fragmentation, other
+                # allocations) would preempt at a different step and
+                # desync the first collective after it. Take the MIN.
+                import torch.distributed as dist
+                t = torch.tensor([num_blocks], dtype=torch.int64,
+                                 device=self.device)
+                dist.all_reduce(t, op=dist.ReduceOp.MIN,
+                                group=state().tp_group)
+                num_blocks = int(t.item())
         else:
             num_blocks = cfg.max_batch * (
                 (cfg.max_seq_len + cfg.block_size - 1) // cfg.block_size)
