@@ -404,15 +404,23 @@ def export_model_gguf(path: str, state_dict: dict, model_name: str,
     """Write a CausalLM state dict as GGUF: 2-D weights quantized with
     `quant`, 1-D tensors as F32."""
     gtype, _bb, qfn, _ = QUANT_CODECS[quant]
+    blk = QK_K if quant.endswith("_K") else QK
     tensors: dict = {}
     for name, t in state_dict.items():
         t = t.detach().float().cpu()
         # projections quantize; embeddings/norms stay F32 (matches the
-        # in-memory quantize_model policy)
-        if t.dim() >= 2 and t.shape[-1] % QK == 0 and "embed" not in name:
-            tensors[name] = (qfn(t), gtype, list(t.shape))
-        else:
-            tensors[name] = (t, GGML_F32, list(t.shape))
+        # in-memory quantize_model policy). K-quants need K % 256 == 0;
+        # non-divisible projections fall back to q8_0 like llama.cpp's
+        # mixed-type files do for awkward shapes.
+        if t.dim() >= 2 and "embed" not in name:
+            if t.shape[-1] % blk == 0:
+                tensors[name] = (qfn(t), gtype, list(t.shape))
+                continue
+            if quant.endswith("_K") and t.shape[-1] % QK == 0:
+                tensors[name] = (quantize_q8_0(t), GGML_Q8_0,
+                                 list(t.shape))
+                continue
+        tensors[name] = (t, GGML_F32, list(t.shape))
     write_gguf(path, tensors, metadata={
         "general.architecture": "substratus-causallm",
         "general.name": model_name,
