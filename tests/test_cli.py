@@ -237,3 +237,29 @@ def test_suspend_resume_notebook(daemon, tmp_path):
     assert r.exit_code == 0, r.output
     obj = httpx.get(f"{url}/api/objects/Notebook/default/pausable").json()
     assert obj["spec"]["suspend"] is False
+
+
+def test_watch_view_with_logs(daemon, tmp_path):
+    """get --watch --logs renders log panes for not-ready objects."""
+    platform, url = daemon
+    import yaml as _yaml
+    m = tmp_path / "m.yaml"
+    m.write_text(_yaml.safe_dump({
+        "apiVersion": "substratus.ai/v1", "kind": "Model",
+        "metadata": {"name": "watchme", "namespace": "default"},
+        "spec": {"image": "substratus-amd/model-loader",
+                 "params": {"name": "definitely-not-a-model"}}}))
+    runner.invoke(cli.app, ["apply", "-f", str(m)])
+    # give the runtime a moment to spawn + write logs for the modeller job
+    path = platform.runtime.log_path("default", "watchme-modeller")
+    os.makedirs(os.path.dirname(path), exist_ok=True)
+    with open(path, "a") as f:
+        f.write("loading definitely-not-a-model...\n")
+    c = cli._client()
+    view = cli._watch_view(c, "models/watchme", True)
+    from rich.console import Console as _C
+    buf = _C(record=True, width=100)
+    buf.print(view)
+    text = buf.export_text()
+    assert "watchme" in text
+    assert "definitely-not-a-model" in text   # the log pane rendered
