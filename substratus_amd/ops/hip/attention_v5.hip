@@ -78,7 +78,7 @@ DEV_INLINE int swz(int row, int chunk) {
   return row * NCH + (chunk ^ (row & (NCH - 1)));
 }
 
-template <int D>
+template <int D, int CH = V5_KCHUNK>
 __global__ __launch_bounds__(V5_WAVES * WAVE)
 void attn_v5_kernel(bf16_t* __restrict__ out,      // [T, HQ, D]
                     const bf16_t* __restrict__ q,  // [T, HQ, D]
@@ -128,22 +128,22 @@ void attn_v5_kernel(bf16_t* __restrict__ out,      // [T, HQ, D]
   // LDS: K rows (QK^T A-frags) + V tr16 images (PV A-frags), dbuf
   //   v_img[c(32d)][sc][g2(16k)][sel][16][16]:
   //   img[( (kk>=8)*8 + dhalf*4 + kk%4 )][d%16], sel = (kk%8)>=4
-  __shared__ uint4 k_tile[2][V5_KCHUNK * D * 2 / 16];
-  __shared__ short v_img[2][DC][2][2][2][16][16];
+  __shared__ uint4 k_tile[2][CH * D * 2 / 16];
+  __shared__ short v_img[2][DC][CH / 32][2][2][16][16];
 
   const int kmax_tile = min(q0_tile + V5_QTILE, n);
-  const int nchunks = (kmax_tile + V5_KCHUNK - 1) / V5_KCHUNK;
+  const int nchunks = (kmax_tile + CH - 1) / CH;
   const float scale2 = scale * 1.44269504088896340736f;
 
   // T14 async-STAGE split (guide G15, +17% on attention): ISSUE the next
   // chunk's global loads before the compute phase, WRITE them to LDS
   // after it — HBM latency hides under the MFMA stream.
-  constexpr int LOADS = V5_KCHUNK * D / 8 / (V5_WAVES * WAVE);
+  constexpr int LOADS = CH * D / 8 / (V5_WAVES * WAVE);
   uint4 st_k[LOADS], st_v[LOADS];
 
   auto stage_load = [&](int kc2) {
     if (kc2 >= nchunks) return;
-    const int kbase = kc2 * V5_KCHUNK;
+    const int kbase = kc2 * CH;
     #pragma unroll
     for (int i = 0; i < LOADS; ++i) {
       const int flat = (i * V5_WAVES * WAVE + threadIdx.x) * 8;
@@ -187,15 +187,15 @@ void attn_v5_kernel(bf16_t* __restrict__ out,      // [T, HQ, D]
   for (int kc = 0; kc < nchunks; ++kc) {
     const int cur = kc & 1;
     stage_load(kc + 1);              // loads fly over this chunk's MFMAs
-    const int kbase = kc * V5_KCHUNK;
+    const int kbase = kc * CH;
     if (wave_live && kbase <= min(q0 + 31, n - 1)) {
       // ---- S^T = K·Q^T over the WHOLE 64-key chunk (both 32-subtiles
       // back-to-back: one online-softmax step per chunk instead of two —
       // halves the rescale/stat passes and gives the scheduler two
       // independent MFMA chains)
-      float16v sacc[2];
+      float16v sacc[CH / 32];
       #pragma unroll
-      for (int sc = 0; sc < 2; ++sc) {
+      for (int sc = 0; sc < CH / 32; ++sc) {
         sacc[sc] = (float16v){};
         #pragma unroll
         for (int kk = 0; kk < KK; ++kk) {
@@ -207,10 +207,10 @@ void attn_v5_kernel(bf16_t* __restrict__ out,      // [T, HQ, D]
         }
       }
       // ---- one in-register online-softmax step over 64 keys ------------
-      float sv[2][16];
+      float sv[CH / 32][16];
       float mloc = -INFINITY;
       #pragma unroll
-      for (int sc = 0; sc < 2; ++sc) {
+      for (int sc = 0; sc < CH / 32; ++sc) {
         #pragma unroll
         for (int r = 0; r < 16; ++r) {
           const int kpos = kbase + sc * 32 + (r & 3) + 8 * (r >> 2) + 4 * hi;
@@ -236,9 +236,9 @@ void attn_v5_kernel(bf16_t* __restrict__ out,      // [T, HQ, D]
             for (int r = 0; r < 16; ++r) o_t[c][r] *= alpha;
         }
         float rowsum = 0.f;
-        unsigned w[2][8];
+        unsigned w[CH / 32][8];
         #pragma unroll
-        for (int sc = 0; sc < 2; ++sc) {
+        for (int sc = 0; sc < CH / 32; ++sc) {
           #pragma unroll
           for (int i = 0; i < 8; ++i) {
             float p0 = (sv[sc][2 * i] == -INFINITY) ? 0.f
@@ -253,7 +253,7 @@ void attn_v5_kernel(bf16_t* __restrict__ out,      // [T, HQ, D]
         l = l * alpha + rowsum;
         // ---- O^T += V^T · P over the chunk's four 16-key steps ---------
         #pragma unroll
-        for (int sc = 0; sc < 2; ++sc) {
+        for (int sc = 0; sc < CH / 32; ++sc) {
           #pragma unroll
           for (int g = 0; g < 2; ++g) {
             auto r02 = __builtin_amdgcn_permlane32_swap(
@@ -314,10 +314,12 @@ void launch_attn_v5(bf16_t* out, const bf16_t* q, const bf16_t* k,
   dim3 grid(num_seqs, hq, tiles);
   dim3 block(V5_WAVES * WAVE);
   if (d == 128) {
-    attnv5::attn_v5_kernel<128><<<grid, block, 0, stream>>>(
+    attnv5::attn_v5_kernel<128, 64><<<grid, block, 0, stream>>>(
         out, q, k, v, seq_start, hq, hkv, scale, lse);
   } else if (d == 64) {
-    attnv5::attn_v5_kernel<64><<<grid, block, 0, stream>>>(
+    // CH=128 measured -21% at d=64 (fewer barriers did not pay for the
+    // longer serial softmax span); CH stays 64
+    attnv5::attn_v5_kernel<64, 64><<<grid, block, 0, stream>>>(
         out, q, k, v, seq_start, hq, hkv, scale, lse);
   }
 }
