@@ -131,6 +131,10 @@ class Engine:
         self.waiting: list[SequenceState] = []
         self.running: list[SequenceState] = []
         self._gen = torch.Generator(device="cpu").manual_seed(cfg.seed)
+        import os as _os
+        # sanitizer tier (scripts/sanitize_gpu.sh) forces eager: serialized
+        # launches invalidate hipGraph stream capture
+        self._force_eager = _os.environ.get("SUB_EAGER_DECODE") == "1"
         self.graph_runner = None
         self.steps = 0
         self.tokens_generated = 0
@@ -217,12 +221,8 @@ class Engine:
         # hipGraph capture of a forward containing RCCL collectives
         # (dist.all_reduce inside the row-parallel layers) is unvalidated on
         # ROCm — TP decodes eagerly until graph+RCCL capture is proven.
-        # SUB_EAGER_DECODE=1 forces eager globally (the sanitizer tier runs
-        # with serialized launches, which invalidate stream capture).
-        import os as _os
         if (self.cfg.use_graphs and self.device.type == "cuda"
-                and state().tp_size <= 1
-                and _os.environ.get("SUB_EAGER_DECODE") != "1"):
+                and state().tp_size <= 1 and not self._force_eager):
             return self._decode_graph()
         return self._decode_eager()
 
