@@ -536,7 +536,7 @@ namespace attnv5 {
 // in-register to B-fragments, dV^T = dO^T·P and dK^T = Q^T·dS accumulated
 // transposed (per-lane key-scalar epilogue; GQA heads atomically add into
 // the fp32 dk/dv buffers, grid is per Q head like the 16-row dkv).
-template <int D>
+template <int D, bool NOATOMIC = false>  // NOATOMIC: timing-only probe
 __global__ __launch_bounds__(V5_WAVES * WAVE)
 void attn_bwd_dkv_v5_kernel(float* __restrict__ dk,       // [T, HKV, D] f32
                             float* __restrict__ dv,
@@ -727,7 +727,7 @@ void attn_bwd_dkv_v5_kernel(float* __restrict__ dk,       // [T, HKV, D] f32
       #pragma unroll
       for (int r = 0; r < 16; ++r) {
         const int d = c * 32 + (r & 3) + 8 * (r >> 2) + 4 * hi;
-        if (grp == 1) {
+        if (NOATOMIC || grp == 1) {
           dk[roff + d] = dk_t[c][r];
           dv[roff + d] = dv_t[c][r];
         } else {
@@ -750,10 +750,23 @@ void launch_attn_bwd_dkv_v5(float* dk, float* dv, const bf16_t* q,
   const int tiles = (max_len + V5_QTILE - 1) / V5_QTILE;
   dim3 grid(num_seqs, hq, tiles);
   dim3 block(V5_WAVES * WAVE);
-  if (d == 128)
-    attnv5::attn_bwd_dkv_v5_kernel<128><<<grid, block, 0, stream>>>(
-        dk, dv, q, k, v, dout, lse, delta, seq_start, hq, hkv, scale);
-  else
-    attnv5::attn_bwd_dkv_v5_kernel<64><<<grid, block, 0, stream>>>(
-        dk, dv, q, k, v, dout, lse, delta, seq_start, hq, hkv, scale);
+  const char* na = getenv("SUB_DKV_NOATOMIC");   // timing-only probe
+  const bool noat = na && na[0] == '1';
+  if (d == 128) {
+    if (noat)
+      attnv5::attn_bwd_dkv_v5_kernel<128, true><<<grid, block, 0,
+          stream>>>(dk, dv, q, k, v, dout, lse, delta, seq_start, hq,
+                    hkv, scale);
+    else
+      attnv5::attn_bwd_dkv_v5_kernel<128><<<grid, block, 0, stream>>>(
+          dk, dv, q, k, v, dout, lse, delta, seq_start, hq, hkv, scale);
+  } else {
+    if (noat)
+      attnv5::attn_bwd_dkv_v5_kernel<64, true><<<grid, block, 0,
+          stream>>>(dk, dv, q, k, v, dout, lse, delta, seq_start, hq,
+                    hkv, scale);
+    else
+      attnv5::attn_bwd_dkv_v5_kernel<64><<<grid, block, 0, stream>>>(
+          dk, dv, q, k, v, dout, lse, delta, seq_start, hq, hkv, scale);
+  }
 }
