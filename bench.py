@@ -44,6 +44,11 @@ def parse_args():
                         "18.3k at mb4; mb32 exceeds 288 GB without "
                         "gradient checkpointing)")
     p.add_argument("--seq-len", type=int, default=None)
+    p.add_argument("--prefill-chunk", type=int, default=None,
+                   help="max packed prefill tokens per step (default: "
+                        "batch*prompt_len, one chunk; cap it for models "
+                        "whose prefill activations would not fit beside "
+                        "weights+KV, e.g. 32768 for llama2-70b bf16)")
     p.add_argument("--kv-dtype", default=None, choices=[None, "fp8"],
                    help="KV-cache dtype override (fp8 e4m3; reduced-precision"
                         " secondary config, NOT the headline)")
@@ -105,7 +110,8 @@ def bench_serve(args) -> None:
     max_seq = args.prompt_len + args.warmup + args.steps + 64
     cfg = EngineConfig(model=model_name, dtype=dtype, max_batch=args.batch,
                        max_seq_len=max_seq,
-                       prefill_chunk=args.batch * args.prompt_len,
+                       prefill_chunk=(args.prefill_chunk or
+                                      args.batch * args.prompt_len),
                        quant=args.quant,
                        kv_dtype=args.kv_dtype,
                        seed=1234 + st.dp_rank)

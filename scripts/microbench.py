@@ -280,6 +280,14 @@ if __name__ == "__main__":
         bench_attn()
         bench_attn(ctx=2048)
         bench_attn(hq=64, hkv=8, ctx=512)  # llama2-70b-style GQA
+    if a.which == "attngqa":           # GQA decode sweep (LDS-staged path)
+        bench_attn(B=512, hq=32, hkv=32, ctx=512)   # MHA reference point
+        bench_attn(B=512, hq=64, hkv=8, ctx=512)    # llama2-70b
+        bench_attn(B=256, hq=64, hkv=8, ctx=512)
+        bench_attn(B=256, hq=64, hkv=8, ctx=128)    # short-ctx regime
+        bench_attn(B=512, hq=64, hkv=8, ctx=2048)
+        bench_attn(B=512, hq=128, hkv=8, d=64, ctx=512)  # GQA-16 d=64
+        bench_attn(B=64, hq=64, hkv=8, ctx=512)     # nsplit>1 regime
     if a.which in ("all", "prefill"):
         bench_prefill()
         bench_prefill(S=2048, B=4)

@@ -15,7 +15,7 @@ void launch_silu_mul(bf16_t*, const bf16_t*, long, int, hipStream_t);
 void launch_gelu(bf16_t*, const bf16_t*, long, hipStream_t);
 void launch_rope_kv_append(bf16_t*, bf16_t*, const bf16_t*, const int*, void*, void*, const int*, int, int, int, int, int, float, bool, long, long, long, hipStream_t);
 void launch_paged_attn_decode(bf16_t*, const bf16_t*, const void*, const void*, const int*, const int*, int, int, int, int, int, int, float, long, int, float*, float*, bool, hipStream_t);
-int paged_attn_nsplit(int, int, int);
+int paged_attn_nsplit(int, int, int, int);
 bool paged_attn_decode_supported(int, int);
 void launch_attn_prefill(bf16_t*, const bf16_t*, const bf16_t*, const bf16_t*, const int*, int, int, int, int, int, float, hipStream_t, float*);
 bool attn_prefill_supported(int);
@@ -182,7 +182,7 @@ void paged_attn_decode(torch::Tensor out, torch::Tensor q,
               "paged_attn_decode: unsupported head_dim ", d, " / block_size ",
               block_size);
   TORCH_CHECK(hq % hkv == 0, "hq must be divisible by hkv");
-  const int nsplit = paged_attn_nsplit(bsz, hq, hkv);
+  const int nsplit = paged_attn_nsplit(bsz, hq, hkv, d);
   float *po = nullptr, *pml = nullptr;
   torch::Tensor part_o, part_ml;
   if (nsplit > 1) {

@@ -127,7 +127,10 @@ def test_rope_kv_append(d, theta):
 
 # ---------------------------------------------------------- decode attention
 @pytest.mark.parametrize("hq,hkv,d", [(8, 8, 128), (8, 1, 128), (16, 2, 128),
-                                      (8, 8, 64), (71, 1, 64)])
+                                      (8, 8, 64), (71, 1, 64),
+                                      # LDS-staged GQA path HPW=1/2/4
+                                      (8, 2, 128), (16, 1, 128),
+                                      (16, 1, 64), (8, 2, 64)])
 def test_paged_attn_decode(hq, hkv, d):
     from substratus_amd import ops
     from substratus_amd.ops import ref
@@ -455,13 +458,15 @@ def test_prefill_matches_decode_continuation():
 
 
 # ------------------------------------------------------------- fp8 KV cache
-def test_paged_attn_decode_fp8_kv():
+@pytest.mark.parametrize("hq,hkv", [(8, 4), (8, 1), (16, 1)])
+def test_paged_attn_decode_fp8_kv(hq, hkv):
     """fp8 e4m3 KV cache vs an fp32 reference computed from the SAME
-    quantized values (the quantization error itself is checked separately)."""
+    quantized values (the quantization error itself is checked separately).
+    (8,4)=register path; (8,1)/(16,1)=LDS-staged GQA path HPW 2/4."""
     from substratus_amd.ops import ref
     from substratus_amd import ops
     torch.manual_seed(40)
-    bsz, hq, hkv, d, bs = 4, 8, 4, 128, 16
+    bsz, d, bs = 4, 128, 16
     lens = [1, 17, 64, 33]
     maxb = 4
     nb = 20
