@@ -553,6 +553,299 @@ void paged_attn_decode_gqa_kernel(
   }
 }
 
+// ---------------------------------------------------------- MFMA GQA path
+// The scalar GQA kernel above is ISSUE/LATENCY-bound, not memory-bound: it
+// spends ~400 VALU+LDS instructions per wave per 8 KB KV block (8 q heads
+// of dot/softmax/PV per byte) and measures 2.45 TB/s at occupancy 4 where
+// the 1-head MHA path reaches 6.5. This variant puts the GQA group on the
+// MFMA M dimension instead: S[group<=16, 32 tok] = Q·K^T from an
+// XOR-swizzled row-major K tile (B-frags read conflict-free straight from
+// LDS, same scheme as attention_bwd.hip's k_tile), P relayed through a
+// per-wave tr16 image (ds_read_b64_tr_b16 A-frags, same as its ds_img),
+// and O[group, D] += P·V with V staged as tr16 B-frag images. Waves split
+// the D columns, so every wave computes the SAME S/softmax (8 redundant
+// MFMAs — free next to the byte stream) and owns its O slice outright: no
+// cross-wave merge, no stats exchange. ~10 MFMA + ~60 VALU per wave per
+// 16 KB chunk.
+typedef __attribute__((ext_vector_type(8))) short short8d;
+typedef __attribute__((ext_vector_type(4))) short short4d;
+typedef __attribute__((ext_vector_type(4))) float float4d;
+
+DEV_INLINE short4d dec_tr16_read(const short* p) {
+  return __builtin_amdgcn_ds_read_tr16_b64_v4i16(
+      (__attribute__((address_space(3))) short4d*)p);
+}
+
+DEV_INLINE unsigned dec_cvt_pk_bf16(float lo, float hi) {
+  unsigned r;
+  asm("v_cvt_pk_bf16_f32 %0, %1, %2" : "=v"(r) : "v"(lo), "v"(hi));
+  return r;
+}
+
+template <int ROW_BYTES>
+DEV_INLINE int dec_swz(int row, int chunk) {
+  constexpr int NCH = ROW_BYTES / 16;
+  return row * NCH + (chunk ^ (row & (NCH - 1)));
+}
+
+#define DEC_CH 32   // tokens per chunk (2 KV blocks)
+constexpr float DEC_LOG2E = 1.44269504088896340736f;
+
+// load one 8-element source granule (16B of bf16, or 8B of fp8 converted
+// through the hardware cvt) as a short8 of bf16 for the LDS tiles
+template <bool KV8>
+DEV_INLINE short8d dec_load_granule(const char* p, bool live) {
+  if (!KV8) {
+    uint4v g = live ? nt_load_u4(p) : (uint4v)0;
+    return *reinterpret_cast<short8d*>(&g);
+  }
+  const uint2v g = live ? nt_load_u2(p) : (uint2v)0;
+  short8d r;
+  #pragma unroll
+  for (int i = 0; i < 2; ++i) {
+    const unsigned w = g[i];
+    const unsigned a = dec_cvt_pk_bf16(fp8_to_f32<0>(w), fp8_to_f32<1>(w));
+    const unsigned b = dec_cvt_pk_bf16(fp8_to_f32<2>(w), fp8_to_f32<3>(w));
+    r[i * 4 + 0] = (short)(a & 0xffff);
+    r[i * 4 + 1] = (short)(a >> 16);
+    r[i * 4 + 2] = (short)(b & 0xffff);
+    r[i * 4 + 3] = (short)(b >> 16);
+  }
+  return r;
+}
+
+template <int D, bool KV8>
+__global__ __launch_bounds__(NW * WAVE)
+void paged_attn_decode_mfma_kernel(
+    bf16_t* __restrict__ out,            // [B, HQ, D]
+    const bf16_t* __restrict__ q,        // [B, HQ, D]
+    const void* __restrict__ k_cache,    // [NB, HKV, BS, D] bf16 | fp8
+    const void* __restrict__ v_cache,
+    const int* __restrict__ block_table, // [B, max_blocks]
+    const int* __restrict__ seq_lens,    // [B]
+    int hq, int hkv, int max_blocks, float scale,
+    long q_rstride, int nsplit,
+    float* __restrict__ part_o,           // [B, HQ, NS, D]  (nsplit > 1)
+    float* __restrict__ part_ml) {        // [B, HQ, NS, 2]
+  constexpr int DC32 = D / 32;           // 32-elem k-steps of a row
+  constexpr int DC16 = D / 16;           // 16-col subtiles of a row
+  constexpr int CPW = DC16 / NW;         // out col-subtiles per wave
+  constexpr int GRAN = DEC_CH * D / 8;   // 8-elem source granules per tile
+  constexpr int GPT = GRAN / (NW * WAVE); // granules per thread (K or V)
+  constexpr int ESZ = KV8 ? 1 : 2;
+
+  const int b = blockIdx.x;
+  const int h = blockIdx.y;              // kv head
+  const int split = blockIdx.z;
+  const int group = hq / hkv;            // <= 16
+  const int wave = threadIdx.x / WAVE;
+  const int lane = threadIdx.x % WAVE;
+  const int lhi = lane >> 4;
+  const int llo = lane & 15;
+  const int tid = threadIdx.x;
+
+  const int n = seq_lens[b];
+  const int nch = (n + DEC_CH - 1) / DEC_CH;
+  const int nblk = (n + BS - 1) / BS;
+  const float scale2 = scale * DEC_LOG2E;
+
+  // Q A-fragments: row llo = q head (zero-padded past the group)
+  short8d qa[DC32];
+  {
+    const bool live = llo < group;
+    const bf16_t* qrow =
+        q + (long)b * q_rstride + (long)(h * group + (live ? llo : 0)) * D;
+    #pragma unroll
+    for (int c = 0; c < DC32; ++c) {
+      short8d v = *reinterpret_cast<const short8d*>(qrow + c * 32 + lhi * 8);
+      if (!live)
+        #pragma unroll
+        for (int j = 0; j < 8; ++j) v[j] = 0;
+      qa[c] = v;
+    }
+  }
+
+  float m2[4], l[4];                     // per C row (lhi*4+i), log2 domain
+  float4d oacc[CPW];
+  #pragma unroll
+  for (int i = 0; i < 4; ++i) { m2[i] = -INFINITY; l[i] = 0.f; }
+  #pragma unroll
+  for (int c = 0; c < CPW; ++c) oacc[c] = (float4d){0.f, 0.f, 0.f, 0.f};
+
+  __shared__ uint4v k_tile[2][DEC_CH * D * 2 / 16];      // swizzled rows
+  __shared__ short v_img[2][DC16][DEC_CH / 32][2][16][16]; // tr16 B-frags
+  __shared__ short p_img[NW][DEC_CH / 32][2][16][16];      // tr16 A-frags
+
+  const int* bt = block_table + (long)b * max_blocks;
+  const long head_off = (long)h * BS * D;       // elems
+  const long blk_stride = (long)hkv * BS * D;   // elems
+
+  // stage one 32-token chunk: granule g -> row g*8/D, col (g*8)%D
+  auto stage = [&](int ci, int buf) {
+    #pragma unroll
+    for (int i = 0; i < GPT; ++i) {
+      const int g = i * NW * WAVE + tid;
+      const int row = g * 8 / D, col = (g * 8) % D;
+      const int tok = ci * DEC_CH + row;
+      const int bl = 2 * ci + row / BS;
+      const bool live = tok < n && bl < nblk;
+      const long base =
+          live ? ((long)bt[bl] * blk_stride + head_off +
+                  (long)(row % BS) * D + col) * ESZ
+               : 0;
+      const short8d ks = dec_load_granule<KV8>(
+          reinterpret_cast<const char*>(k_cache) + base, live);
+      const short8d vs = dec_load_granule<KV8>(
+          reinterpret_cast<const char*>(v_cache) + base, live);
+      *reinterpret_cast<short8d*>(
+          &k_tile[buf][dec_swz<D * 2>(row, col / 8)]) = ks;
+      const int cc = col / 16, off = col % 16;
+      const int kk = row % 32;
+      const int sel = (kk % 8) >= 4;
+      const int rr = (kk / 8) * 4 + (kk % 4);
+      *reinterpret_cast<short8d*>(
+          &v_img[buf][cc][row / 32][sel][rr][off]) = vs;
+    }
+  };
+
+  if (split < nch) stage(split, 0);
+  __syncthreads();
+
+  int cur = 0;
+  for (int ci = split; ci < nch; ci += nsplit) {
+    const int cn = ci + nsplit;
+
+    // ---- S = Q·K^T (two 16-col subchunks), softmax in log2 domain ------
+    float4d sf[DEC_CH / 16];
+    #pragma unroll
+    for (int sc = 0; sc < DEC_CH / 16; ++sc) {
+      sf[sc] = (float4d){0.f, 0.f, 0.f, 0.f};
+      const int key = sc * 16 + llo;
+      #pragma unroll
+      for (int c = 0; c < DC32; ++c) {
+        const uint4v kb = k_tile[cur][dec_swz<D * 2>(key, c * 4 + lhi)];
+        sf[sc] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            qa[c], *reinterpret_cast<const short8d*>(&kb), sf[sc], 0, 0, 0);
+      }
+    }
+    float bm[4];
+    #pragma unroll
+    for (int i = 0; i < 4; ++i) {
+      float v = -INFINITY;
+      #pragma unroll
+      for (int sc = 0; sc < DEC_CH / 16; ++sc) {
+        const int tok = ci * DEC_CH + sc * 16 + llo;
+        if (tok < n) v = fmaxf(v, sf[sc][i] * scale2);
+      }
+      #pragma unroll
+      for (int off = 8; off > 0; off >>= 1)
+        v = fmaxf(v, __shfl_xor(v, off, WAVE));
+      bm[i] = v;
+    }
+    // P (bf16) into the per-wave tr16 image + row sums
+    #pragma unroll
+    for (int i = 0; i < 4; ++i) {
+      const float mn = fmaxf(m2[i], bm[i]);
+      if (mn == -INFINITY) continue;
+      const float alpha =
+          (m2[i] == -INFINITY) ? 0.f : __builtin_amdgcn_exp2f(m2[i] - mn);
+      m2[i] = mn;
+      l[i] *= alpha;
+      #pragma unroll
+      for (int c = 0; c < CPW; ++c) oacc[c][i] *= alpha;
+    }
+    float pv[DEC_CH / 16][4];
+    #pragma unroll
+    for (int sc = 0; sc < DEC_CH / 16; ++sc) {
+      const int tok = ci * DEC_CH + sc * 16 + llo;
+      #pragma unroll
+      for (int i = 0; i < 4; ++i) {
+        const float p =
+            (tok < n && m2[i] != -INFINITY)
+                ? __builtin_amdgcn_exp2f(sf[sc][i] * scale2 - m2[i])
+                : 0.f;
+        pv[sc][i] = p;
+      }
+    }
+    #pragma unroll
+    for (int i = 0; i < 4; ++i) {
+      float rs = 0.f;
+      #pragma unroll
+      for (int sc = 0; sc < DEC_CH / 16; ++sc) rs += pv[sc][i];
+      #pragma unroll
+      for (int off = 8; off > 0; off >>= 1)
+        rs += __shfl_xor(rs, off, WAVE);
+      l[i] += rs;
+    }
+    #pragma unroll
+    for (int sc = 0; sc < DEC_CH / 16; ++sc) {
+      const int key = sc * 16 + llo;
+      const int kk = key & 31;
+      *reinterpret_cast<uint2*>(
+          &p_img[wave][key >> 5][(kk % 8) >= 4]
+                [(kk / 8) * 4 + (kk % 4)][lhi * 4]) =
+          make_uint2(dec_cvt_pk_bf16(pv[sc][0], pv[sc][1]),
+                     dec_cvt_pk_bf16(pv[sc][2], pv[sc][3]));
+    }
+    __builtin_amdgcn_wave_barrier();
+
+    // ---- O += P·V over the chunk's 32 tokens ---------------------------
+    #pragma unroll
+    for (int ks = 0; ks < DEC_CH / 32; ++ks) {
+      const short4d alo = dec_tr16_read(&p_img[wave][ks][0][0][0] + lane * 4);
+      const short4d ahi = dec_tr16_read(&p_img[wave][ks][1][0][0] + lane * 4);
+      short8d pa;
+      pa[0] = alo[0]; pa[1] = alo[1]; pa[2] = alo[2]; pa[3] = alo[3];
+      pa[4] = ahi[0]; pa[5] = ahi[1]; pa[6] = ahi[2]; pa[7] = ahi[3];
+      #pragma unroll
+      for (int c = 0; c < CPW; ++c) {
+        const int cc = wave * CPW + c;
+        const short4d lo =
+            dec_tr16_read(&v_img[cur][cc][ks][0][0][0] + lane * 4);
+        const short4d hi =
+            dec_tr16_read(&v_img[cur][cc][ks][1][0][0] + lane * 4);
+        short8d vb;
+        vb[0] = lo[0]; vb[1] = lo[1]; vb[2] = lo[2]; vb[3] = lo[3];
+        vb[4] = hi[0]; vb[5] = hi[1]; vb[6] = hi[2]; vb[7] = hi[3];
+        oacc[c] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            pa, vb, oacc[c], 0, 0, 0);
+      }
+    }
+
+    // stage the next chunk into the other buffer; one barrier orders the
+    // writes against the next iteration's reads (see scalar GQA kernel)
+    if (cn < nch) stage(cn, cur ^ 1);
+    __syncthreads();
+    cur ^= 1;
+  }
+
+  // ---- epilogue: C row = q head, col = wave's D slice ------------------
+  #pragma unroll
+  for (int i = 0; i < 4; ++i) {
+    const int row = lhi * 4 + i;
+    if (row >= group) continue;
+    const int qhead = h * group + row;
+    if (nsplit == 1) {
+      const float invl = (l[i] > 0.f) ? 1.f / l[i] : 0.f;
+      bf16_t* orow = out + ((long)b * hq + qhead) * D;
+      #pragma unroll
+      for (int c = 0; c < CPW; ++c)
+        orow[(wave * CPW + c) * 16 + llo] = f2bf(oacc[c][i] * invl);
+    } else {
+      float* po = part_o + (((long)b * hq + qhead) * nsplit + split) * D;
+      #pragma unroll
+      for (int c = 0; c < CPW; ++c)
+        po[(wave * CPW + c) * 16 + llo] = oacc[c][i];
+      if (llo == 0 && wave == 0) {
+        float* pm = part_ml + (((long)b * hq + qhead) * nsplit + split) * 2;
+        // reduce kernel merges in the natural-log domain
+        pm[0] = (m2[i] == -INFINITY) ? -INFINITY : m2[i] / DEC_LOG2E;
+        pm[1] = l[i];
+      }
+    }
+  }
+}
+
 // GQA heads-per-wave for the LDS-staged decode variant; 0 = use the
 // register-tiled kernel (MHA group 1, falcon's prime group 71, d oddballs).
 static inline int gqa_hpw(int hq, int hkv, int d) {
@@ -630,8 +923,26 @@ void launch_paged_attn_decode(bf16_t* out, const bf16_t* q,
   const int hpw = gqa_hpw(hq, hkv, d);
   dim3 block(NW * WAVE);
   if (hpw > 0) {
-    // LDS-staged GQA path: groups 4/8/16 (llama2/3-70b, GQA-16 d=64)
     dim3 grid(bsz, hkv, nsplit);
+    // default: the MFMA variant (group on the M dim); SUB_DECODE_GQA=scalar
+    // selects the scalar LDS-staged kernel for A/B measurement
+    static const char* ge = getenv("SUB_DECODE_GQA");
+    static const bool scalar_gqa = ge && ge[0] == 's';
+    if (!scalar_gqa) {
+      #define LAUNCHM(DD, K8) \
+        paged_attn_decode_mfma_kernel<DD, K8><<<grid, block, 0, stream>>>( \
+            out, q, k_cache, v_cache, block_table, seq_lens, hq, hkv, \
+            max_blocks, scale, q_rstride, nsplit, part_o, part_ml)
+      if (!kv8) {
+        if (d == 128) LAUNCHM(128, false);
+        else LAUNCHM(64, false);
+      } else {
+        if (d == 128) LAUNCHM(128, true);
+        else LAUNCHM(64, true);
+      }
+      #undef LAUNCHM
+    } else {
+    // LDS-staged GQA path: groups 4/8/16 (llama2/3-70b, GQA-16 d=64)
     #define LAUNCHG(DD, HH, K8) \
       paged_attn_decode_gqa_kernel<DD, HH, K8><<<grid, block, 0, stream>>>( \
           out, q, k_cache, v_cache, block_table, seq_lens, hq, hkv, \
@@ -652,6 +963,7 @@ void launch_paged_attn_decode(bf16_t* out, const bf16_t* q,
       else LAUNCHG(64, 4, true);
     }
     #undef LAUNCHG
+    }
   } else {
   // GC=8 when the GQA group tiles evenly by 8 but the LDS path is out
   // (non-power-of-two d); GC=1 otherwise (MHA group=1; falcon's prime
