@@ -680,8 +680,11 @@ void paged_attn_decode_mfma_kernel(
   const long head_off = (long)h * BS * D;       // elems
   const long blk_stride = (long)hkv * BS * D;   // elems
 
-  // stage one 32-token chunk: granule g -> row g*8/D, col (g*8)%D
-  auto stage = [&](int ci, int buf) {
+  // staging is split T14-style: loads into registers issue BEFORE the
+  // chunk's compute (a full compute phase of latency slack), LDS writes
+  // land after it. granule g -> row g*8/D, col (g*8)%D.
+  short8d nk[GPT], nv[GPT];
+  auto stage_load = [&](int ci) {
     #pragma unroll
     for (int i = 0; i < GPT; ++i) {
       const int g = i * NW * WAVE + tid;
@@ -693,27 +696,38 @@ void paged_attn_decode_mfma_kernel(
           live ? ((long)bt[bl] * blk_stride + head_off +
                   (long)(row % BS) * D + col) * ESZ
                : 0;
-      const short8d ks = dec_load_granule<KV8>(
+      nk[i] = dec_load_granule<KV8>(
           reinterpret_cast<const char*>(k_cache) + base, live);
-      const short8d vs = dec_load_granule<KV8>(
+      nv[i] = dec_load_granule<KV8>(
           reinterpret_cast<const char*>(v_cache) + base, live);
+    }
+  };
+  auto stage_write = [&](int buf) {
+    #pragma unroll
+    for (int i = 0; i < GPT; ++i) {
+      const int g = i * NW * WAVE + tid;
+      const int row = g * 8 / D, col = (g * 8) % D;
       *reinterpret_cast<short8d*>(
-          &k_tile[buf][dec_swz<D * 2>(row, col / 8)]) = ks;
+          &k_tile[buf][dec_swz<D * 2>(row, col / 8)]) = nk[i];
       const int cc = col / 16, off = col % 16;
       const int kk = row % 32;
       const int sel = (kk % 8) >= 4;
       const int rr = (kk / 8) * 4 + (kk % 4);
       *reinterpret_cast<short8d*>(
-          &v_img[buf][cc][row / 32][sel][rr][off]) = vs;
+          &v_img[buf][cc][row / 32][sel][rr][off]) = nv[i];
     }
   };
 
-  if (split < nch) stage(split, 0);
+  if (split < nch) {
+    stage_load(split);
+    stage_write(0);
+  }
   __syncthreads();
 
   int cur = 0;
   for (int ci = split; ci < nch; ci += nsplit) {
     const int cn = ci + nsplit;
+    if (cn < nch) stage_load(cn);
 
     // ---- S = Q·K^T (two 16-col subchunks), softmax in log2 domain ------
     float4d sf[DEC_CH / 16];
@@ -814,7 +828,7 @@ void paged_attn_decode_mfma_kernel(
 
     // stage the next chunk into the other buffer; one barrier orders the
     // writes against the next iteration's reads (see scalar GQA kernel)
-    if (cn < nch) stage(cn, cur ^ 1);
+    if (cn < nch) stage_write(cur ^ 1);
     __syncthreads();
     cur ^= 1;
   }
