@@ -77,6 +77,9 @@ class Engine:
         if device is None:
             device = "cuda" if torch.cuda.is_available() else "cpu"
         self.device = torch.device(device)
+        if self.device.type == "cuda":
+            from ..utils.tunable import load_tunables
+            load_tunables()     # pre-tuned GEMM solutions (no-op sans table)
         if model is None:
             with torch.no_grad():
                 model = CausalLM(self.mcfg, device="meta", dtype=cfg.dtype)

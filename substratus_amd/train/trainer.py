@@ -113,6 +113,9 @@ class Trainer:
             device = (f"cuda:{self.st.local_rank}"
                       if torch.cuda.is_available() else "cpu")
         self.device = torch.device(device)
+        if self.device.type == "cuda":
+            from ..utils.tunable import load_tunables
+            load_tunables()     # pre-tuned GEMM solutions (no-op sans table)
         self.mcfg = get_config(cfg.model)
         if model is None:
             model = CausalLM(self.mcfg, device="meta", dtype=cfg.dtype)
