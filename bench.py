@@ -149,7 +149,11 @@ def bench_serve(args) -> None:
     log(f"[bench] prefill of {args.batch}x{args.prompt_len} tokens: "
         f"{prefill_s:.2f}s ({prefill_tok_s/1e3:.1f}k tok/s, "
         f"ttft {t_first*1e3:.0f} ms)")
-    assert len(eng.running) == args.batch
+    assert len(eng.running) == args.batch, (
+        f"only {len(eng.running)}/{args.batch} sequences running after "
+        f"prefill (kv_blocks={eng.kv.num_blocks}, "
+        f"free={eng.kv.allocator.free_count()}); the KV pool cannot hold "
+        f"batch x max_seq_len at this model size — lower --batch")
     for _ in range(args.warmup):
         eng.decode_only_step()
 
