@@ -136,8 +136,18 @@ def bench_serve(args) -> None:
         torch.cuda.synchronize()
     t_pref = time.time()
     t_first = None
+    stuck = 0
     while eng.waiting:
+        n_wait = len(eng.waiting)
         outs = eng.step()
+        # a too-small KV pool can preempt-admit-preempt forever: fail
+        # loudly instead of spinning until an external timeout
+        stuck = stuck + 1 if len(eng.waiting) >= n_wait else 0
+        assert stuck < 2000, (
+            f"prefill stalled: {len(eng.waiting)} waiting / "
+            f"{len(eng.running)} running, kv free "
+            f"{eng.kv.allocator.free_count()}/{eng.kv.num_blocks} — "
+            f"batch x max_seq_len exceeds the KV pool; lower --batch")
         if outs and t_first is None:
             if on_gpu:
                 torch.cuda.synchronize()
