@@ -146,7 +146,7 @@ def bench_serve(args) -> None:
         assert stuck < 2000, (
             f"prefill stalled: {len(eng.waiting)} waiting / "
             f"{len(eng.running)} running, kv free "
-            f"{eng.kv.allocator.free_count()}/{eng.kv.num_blocks} — "
+            f"{eng.kv.allocator.free_count}/{eng.kv.num_blocks} — "
             f"batch x max_seq_len exceeds the KV pool; lower --batch")
         if outs and t_first is None:
             if on_gpu:
@@ -162,7 +162,7 @@ def bench_serve(args) -> None:
     assert len(eng.running) == args.batch, (
         f"only {len(eng.running)}/{args.batch} sequences running after "
         f"prefill (kv_blocks={eng.kv.num_blocks}, "
-        f"free={eng.kv.allocator.free_count()}); the KV pool cannot hold "
+        f"free={eng.kv.allocator.free_count}); the KV pool cannot hold "
         f"batch x max_seq_len at this model size — lower --batch")
     for _ in range(args.warmup):
         eng.decode_only_step()

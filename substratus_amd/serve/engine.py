@@ -341,6 +341,28 @@ class Engine:
         self.steps += 1
         return res
 
+    def abort(self, seq_id: int) -> bool:
+        """Drop ONE request (client disconnect / cancellation). Safe
+        between steps: a running sequence is removed exactly the way
+        _retire removes a finished one (free blocks, compact running), so
+        the next decode step just sees a smaller batch. Returns False for
+        an unknown id (already finished — the common disconnect race).
+        Under TP the leader broadcasts aborts with the iteration command
+        (serve/tp.py) and every rank applies them identically."""
+        for i, s in enumerate(self.waiting):
+            if s.seq_id == seq_id:
+                del self.waiting[i]
+                return True
+        for s in self.running:
+            if s.seq_id == seq_id:
+                if s.blocks:
+                    self.kv.allocator.free(s.blocks)
+                    s.blocks = []
+                self.running = [r for r in self.running
+                                if r.seq_id != seq_id]
+                return True
+        return False
+
     def abort_all(self) -> None:
         """Drop every waiting/running request and free their KV blocks.
         Whole-engine failure recovery; deterministic, so TP leader and

@@ -64,6 +64,7 @@ class InferenceWorker:
         self.coord = TPCoordinator()
         self.engine = engine
         self.inbox: queue.Queue = queue.Queue()
+        self.abort_inbox: queue.Queue = queue.Queue()
         self.outputs: dict[int, queue.Queue] = {}
         self.lock = threading.Lock()
         self.total_tokens = 0
@@ -82,6 +83,13 @@ class InferenceWorker:
         done.wait()
         return holder["sid"], out_q
 
+    def cancel(self, sid: int) -> None:
+        """Drop a request whose client went away. Queued to the engine
+        thread; a no-op if the sequence already finished. cancel() can
+        only be called after submit() returned sid, so on TP followers
+        the add always lands in an earlier publish than the abort."""
+        self.abort_inbox.put(sid)
+
     def _run(self) -> None:
         while not self._stop:
             moved = False
@@ -91,12 +99,20 @@ class InferenceWorker:
                     pending.append(self.inbox.get_nowait())
                 except queue.Empty:
                     break
-            if self.coord.active and (pending or self.engine.has_work()):
+            abort_ids = []
+            while True:
+                try:
+                    abort_ids.append(self.abort_inbox.get_nowait())
+                except queue.Empty:
+                    break
+            if self.coord.active and (pending or abort_ids or
+                                      self.engine.has_work()):
                 # lockstep TP: publish this iteration's admissions first
                 self.coord.publish([
                     dict(prompt_tokens=pt, max_new_tokens=mnt,
                          temperature=temp, top_k=tk, top_p=tp, eos_token=eos)
-                    for (pt, mnt, temp, tk, tp, eos, *_rest) in pending])
+                    for (pt, mnt, temp, tk, tp, eos, *_rest) in pending],
+                    aborts=abort_ids)
             for (pt, mnt, temp, tk, tp, eos, out_q, holder, done) in pending:
                 sid = self.engine.add_request(pt, max_new_tokens=mnt,
                                               temperature=temp, top_k=tk,
@@ -105,6 +121,11 @@ class InferenceWorker:
                     self.outputs[sid] = out_q
                 holder["sid"] = sid
                 done.set()
+                moved = True
+            for sid in abort_ids:     # after adds: same order as followers
+                self.engine.abort(sid)
+                with self.lock:
+                    self.outputs.pop(sid, None)
                 moved = True
             if self.engine.has_work():
                 t0 = time.time()
@@ -318,26 +339,32 @@ def create_app(engine: Optional[Engine] = None,
 
         if stream:
             def gen():
-                while True:
-                    o: StepOutput = out_q.get(timeout=600)
-                    if o.error:
-                        err = {"error": {"message": o.error,
-                                         "type": "server_error"}}
-                        yield f"data: {json.dumps(err)}\n\n"
-                        yield "data: [DONE]\n\n"
-                        return
-                    chunk = {
-                        "id": f"cmpl-{sid}", "object": "text_completion",
-                        "model": model_id,
-                        "choices": [{"index": 0,
-                                     "text": tok_obj.decode([o.token]),
-                                     "finish_reason": "stop" if o.finished
-                                     else None}],
-                    }
-                    yield f"data: {json.dumps(chunk)}\n\n"
-                    if o.finished:
-                        yield "data: [DONE]\n\n"
-                        return
+                # a dropped client closes the generator (GeneratorExit at
+                # the yield); the finally cancels the request so it stops
+                # occupying batch slots and KV blocks
+                try:
+                    while True:
+                        o: StepOutput = out_q.get(timeout=600)
+                        if o.error:
+                            err = {"error": {"message": o.error,
+                                             "type": "server_error"}}
+                            yield f"data: {json.dumps(err)}\n\n"
+                            yield "data: [DONE]\n\n"
+                            return
+                        chunk = {
+                            "id": f"cmpl-{sid}", "object": "text_completion",
+                            "model": model_id,
+                            "choices": [{"index": 0,
+                                         "text": tok_obj.decode([o.token]),
+                                         "finish_reason": "stop" if o.finished
+                                         else None}],
+                        }
+                        yield f"data: {json.dumps(chunk)}\n\n"
+                        if o.finished:
+                            yield "data: [DONE]\n\n"
+                            return
+                finally:
+                    worker.cancel(sid)    # no-op if already finished
             return StreamingResponse(gen(), media_type="text/event-stream")
 
         toks, err = collect()
@@ -399,27 +426,30 @@ def create_app(engine: Optional[Engine] = None,
                                    0, top_p, eos_token=tok_obj.eos)
         if stream:
             def gen():
-                while True:
-                    o: StepOutput = out_q.get(timeout=600)
-                    if o.error:
-                        yield ("data: " + json.dumps(
-                            {"error": {"message": o.error,
-                                       "type": "server_error"}}) + "\n\n")
-                        yield "data: [DONE]\n\n"
-                        return
-                    chunk = {
-                        "id": f"chatcmpl-{sid}",
-                        "object": "chat.completion.chunk",
-                        "model": model_id,
-                        "choices": [{"index": 0, "delta": {
-                            "content": tok_obj.decode([o.token])},
-                            "finish_reason": "stop" if o.finished
-                            else None}],
-                    }
-                    yield f"data: {json.dumps(chunk)}\n\n"
-                    if o.finished:
-                        yield "data: [DONE]\n\n"
-                        return
+                try:
+                    while True:
+                        o: StepOutput = out_q.get(timeout=600)
+                        if o.error:
+                            yield ("data: " + json.dumps(
+                                {"error": {"message": o.error,
+                                           "type": "server_error"}}) + "\n\n")
+                            yield "data: [DONE]\n\n"
+                            return
+                        chunk = {
+                            "id": f"chatcmpl-{sid}",
+                            "object": "chat.completion.chunk",
+                            "model": model_id,
+                            "choices": [{"index": 0, "delta": {
+                                "content": tok_obj.decode([o.token])},
+                                "finish_reason": "stop" if o.finished
+                                else None}],
+                        }
+                        yield f"data: {json.dumps(chunk)}\n\n"
+                        if o.finished:
+                            yield "data: [DONE]\n\n"
+                            return
+                finally:
+                    worker.cancel(sid)    # no-op if already finished
             return StreamingResponse(gen(), media_type="text/event-stream")
         toks = []
         while True:

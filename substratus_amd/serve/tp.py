@@ -38,11 +38,15 @@ class TPCoordinator:
         # tp_group only when running pure-gloo (CPU tests)
         return self.st.tp_ctrl_group or self.st.tp_group
 
-    def publish(self, adds: list[dict], stop: bool = False) -> None:
-        """Leader: broadcast this iteration's admissions to followers."""
+    def publish(self, adds: list[dict], aborts: list[int] | None = None,
+                stop: bool = False) -> None:
+        """Leader: broadcast this iteration's admissions + aborts to
+        followers. Seq ids are assigned by the same monotonic counter on
+        every rank (followers replay add_request in publish order), so an
+        abort id names the same sequence everywhere."""
         if not self.active:
             return
-        obj = [{"adds": adds, "stop": stop}]
+        obj = [{"adds": adds, "aborts": aborts or [], "stop": stop}]
         dist.broadcast_object_list(obj, src=self.st.rank - self.st.tp_rank,
                                    group=self._ctrl)
 
@@ -76,6 +80,8 @@ def follower_loop(engine) -> None:
             return
         for add in cmd["adds"]:
             engine.add_request(**add)
+        for sid in cmd.get("aborts", []):
+            engine.abort(sid)
         if engine.has_work():
             try:
                 engine.step()

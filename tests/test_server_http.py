@@ -188,6 +188,40 @@ def test_worker_survives_engine_failure(tmp_path):
         assert ok.json()["usage"]["completion_tokens"] == 2
 
 
+def test_streaming_client_disconnect_aborts(tmp_path):
+    """Dropping a streaming client mid-generation cancels the request: the
+    engine stops spending steps on it, frees its KV blocks, and the next
+    request is served normally (Engine.abort via worker.cancel from the
+    generator's finally)."""
+    import time
+    import torch
+    from starlette.testclient import TestClient
+    from substratus_amd.serve.engine import Engine, EngineConfig
+    from substratus_amd.serve.server import create_app
+    eng = Engine(EngineConfig(model="llama-tiny", dtype=torch.float32,
+                              max_batch=2, max_seq_len=256, kv_blocks=32))
+    app = create_app(engine=eng)
+    with TestClient(app) as c:
+        with c.stream("POST", "/v1/completions",
+                      json={"prompt": [1, 2, 3], "max_tokens": 200,
+                            "stream": True}) as r:
+            it = r.iter_lines()
+            first = next(l for l in it if l.startswith("data:"))
+            assert "text" in first
+            # leave the context WITHOUT reading the other ~199 chunks:
+            # the server-side generator is closed -> cancel
+        deadline = time.time() + 10
+        while time.time() < deadline and eng.has_work():
+            time.sleep(0.05)
+        assert not eng.has_work(), "aborted request still occupies the engine"
+        assert eng.kv.allocator.free_count == eng.kv.num_blocks, \
+            "disconnect leaked KV blocks"
+        ok = c.post("/v1/completions", json={"prompt": [4, 5, 6],
+                                             "max_tokens": 3})
+        assert ok.status_code == 200
+        assert ok.json()["usage"]["completion_tokens"] == 3
+
+
 def test_hf_named_checkpoint_serves(tmp_path, monkeypatch):
     """An HF-layout llama safetensors checkpoint (model.layers.N.self_attn.
     q_proj...) loads via the name mapping and produces the SAME generation

@@ -83,6 +83,79 @@ def test_tp2_http_server_end_to_end(tmp_path):
     assert fstats["tokens"] == 4, fstats
 
 
+def _abort_rank_worker(rank, world, q):
+    import traceback
+    try:
+        _abort_rank_inner(rank, world, q)
+    except Exception:
+        q.put((f"ERR{rank}", traceback.format_exc()))
+
+
+def _abort_rank_inner(rank, world, q):
+    import torch
+    import torch.distributed as dist
+    os.environ.update({"RANK": str(rank), "WORLD_SIZE": str(world),
+                       "LOCAL_RANK": str(rank),
+                       "MASTER_ADDR": "127.0.0.1", "MASTER_PORT": "29783"})
+    from substratus_amd.parallel import comm
+    comm._STATE = comm.ParallelState()
+    comm.init_distributed(tp_size=world, backend="gloo")
+    from substratus_amd.serve.engine import Engine, EngineConfig
+    from substratus_amd.serve.tp import TPCoordinator, follower_loop
+    engine = Engine(EngineConfig(model="llama-tiny", dtype=torch.float32,
+                                 max_batch=2, max_seq_len=64, kv_blocks=16),
+                    device="cpu")
+    coord = TPCoordinator()
+    add = dict(prompt_tokens=[3, 1, 4], max_new_tokens=40, temperature=0.0,
+               top_k=0, top_p=1.0, eos_token=-1)
+    if rank == 0:
+        coord.publish([add])
+        sid = engine.add_request(**add)
+        engine.step()
+        for _ in range(3):
+            coord.publish([])
+            engine.step()
+        # abort mid-generation; follower must mirror it or the next
+        # collective hangs / the schedules desync
+        coord.publish([], aborts=[sid])
+        assert engine.abort(sid)
+        assert not engine.has_work()
+        coord.publish([], stop=True)
+    else:
+        follower_loop(engine)
+    q.put((rank, {"steps": engine.steps,
+                  "free": engine.kv.allocator.free_count,
+                  "blocks": engine.kv.num_blocks,
+                  "has_work": engine.has_work()}))
+    dist.destroy_process_group()
+
+
+@pytest.mark.timeout(300)
+def test_tp2_abort_propagates_to_follower():
+    """Per-request abort rides the iteration command: after the leader
+    cancels, BOTH ranks drain, free every KV block, and agree on the step
+    count (a desync here would hang the next all-reduce)."""
+    import torch.multiprocessing as mp
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    procs = [ctx.Process(target=_abort_rank_worker, args=(r, 2, q))
+             for r in range(2)]
+    for p in procs:
+        p.start()
+    results = {}
+    for _ in range(2):
+        rank, stats = q.get(timeout=240)
+        assert not str(rank).startswith("ERR"), stats
+        results[rank] = stats
+    for p in procs:
+        p.join(timeout=60)
+    assert results[0]["steps"] == results[1]["steps"] == 4
+    for r in (0, 1):
+        assert not results[r]["has_work"]
+        assert results[r]["free"] == results[r]["blocks"], \
+            f"rank {r} leaked KV blocks after abort"
+
+
 def test_llama2_70b_tp8_shardable():
     """BASELINE config 5 sanity without weights: llama2-70b head counts and
     FFN divide by tp=8, and the bf16 shard + KV fit comfortably in 8×288 GB
