@@ -352,6 +352,9 @@ class Engine:
         for i, s in enumerate(self.waiting):
             if s.seq_id == seq_id:
                 del self.waiting[i]
+                if s.blocks:        # waiting seqs hold none today (fresh
+                    self.kv.allocator.free(s.blocks)   # and preempted both
+                    s.blocks = []                      # freed) — stay safe
                 return True
         for s in self.running:
             if s.seq_id == seq_id:
