@@ -26,6 +26,9 @@ def main() -> None:
     ap.add_argument("--requests", type=int, default=400)
     ap.add_argument("--max-batch", type=int, default=128)
     ap.add_argument("--max-seq-len", type=int, default=1024)
+    ap.add_argument("--abort-frac", type=float, default=0.1,
+                    help="fraction of requests cancelled mid-generation "
+                         "(client-disconnect simulation; 0 disables)")
     a = ap.parse_args()
 
     on_gpu = torch.cuda.is_available()
@@ -39,6 +42,8 @@ def main() -> None:
     t_sub: dict[int, float] = {}
     t_done: dict[int, float] = {}
     pending = a.requests
+    to_abort: dict[int, int] = {}
+    aborted: set[int] = set()
     steps = 0
     t0 = time.time()
     while pending > 0 or eng.has_work():
@@ -60,24 +65,37 @@ def main() -> None:
                     top_p=rng.choice([1.0, 0.95]))
                 want[sid] = mnew
                 t_sub[sid] = time.time()
+                if rng.random() < a.abort_frac:
+                    # cancel after a random share of its tokens
+                    to_abort[sid] = max(1, int(mnew * rng.random()))
                 pending -= 1
         for out in eng.step():
             now = time.time()
             t_first.setdefault(out.seq_id, now)
             t_done[out.seq_id] = now
             got[out.seq_id] = got.get(out.seq_id, 0) + 1
+            if (out.seq_id in to_abort and
+                    got[out.seq_id] >= to_abort[out.seq_id]):
+                eng.abort(out.seq_id)
+                aborted.add(out.seq_id)
+                del to_abort[out.seq_id]
         steps += 1
     if on_gpu:
         torch.cuda.synchronize()
     wall = time.time() - t0
 
     bad = [(s, want[s], got.get(s, 0)) for s in want
-           if got.get(s, 0) != want[s]]
+           if s not in aborted and got.get(s, 0) != want[s]]
     assert not bad, f"requests with wrong token counts: {bad[:5]}"
-    total_tokens = sum(want.values())
+    over = [(s, got.get(s, 0)) for s in aborted if got.get(s, 0) > want[s]]
+    assert not over, f"aborted requests kept generating: {over[:5]}"
+    assert eng.kv.allocator.free_count == eng.kv.num_blocks, \
+        "KV blocks leaked across the soak"
+    total_tokens = sum(got.values())
     ttfts = sorted(t_first[s] - t_sub[s] for s in want)
     lats = sorted(t_done[s] - t_sub[s] for s in want)
-    print(f"SOAK OK: {len(want)} requests, {total_tokens} tokens, "
+    print(f"SOAK OK: {len(want)} requests ({len(aborted)} aborted), "
+          f"{total_tokens} tokens, "
           f"{steps} engine steps, {wall:.1f}s wall, "
           f"{total_tokens / wall:.0f} tok/s")
     print(f"  TTFT p50/p95: {ttfts[len(ttfts) // 2] * 1000:.0f}/"
