@@ -49,6 +49,10 @@ def parse_args():
                         "batch*prompt_len, one chunk; cap it for models "
                         "whose prefill activations would not fit beside "
                         "weights+KV, e.g. 32768 for llama2-70b bf16)")
+    p.add_argument("--gpu-mem-fraction", type=float, default=None,
+                   help="KV pool share of free HBM after weights (engine "
+                        "default 0.85; raise toward 0.92 for max-batch "
+                        "experiments — llama2-7b batch>512 needs it)")
     p.add_argument("--kv-dtype", default=None, choices=[None, "fp8"],
                    help="KV-cache dtype override (fp8 e4m3; reduced-precision"
                         " secondary config, NOT the headline)")
@@ -114,6 +118,8 @@ def bench_serve(args) -> None:
                                       args.batch * args.prompt_len),
                        quant=args.quant,
                        kv_dtype=args.kv_dtype,
+                       **({"gpu_mem_fraction": args.gpu_mem_fraction}
+                          if args.gpu_mem_fraction else {}),
                        seed=1234 + st.dp_rank)
     log(f"[bench] building {model_name} engine on {device} ...")
     t_build = time.time()
