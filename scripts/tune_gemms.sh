@@ -10,7 +10,7 @@ set -euo pipefail
 cd "$(dirname "$0")/.."
 
 ARCH=$(python -c "import torch; print(torch.cuda.get_device_properties(0).gcnArchName.split(':')[0])")
-OUT=${1:-tunables/${ARCH}.csv}
+OUT=${1:-substratus_amd/data/tunables/${ARCH}.csv}
 WORK=$(mktemp -d)/tune.csv
 
 export PYTORCH_TUNABLEOP_ENABLED=1

@@ -8,7 +8,7 @@ with the gpurun snapshot).
 
 import os
 
-from setuptools import setup
+from setuptools import find_packages, setup
 
 os.environ.setdefault("PYTORCH_ROCM_ARCH", "gfx950")
 
@@ -26,7 +26,12 @@ sources = [os.path.join(HIP_DIR, f) for f in sorted(os.listdir(HIP_DIR))
 setup(
     name="substratus_amd",
     version="0.1.0",
-    packages=["substratus_amd"],
+    packages=find_packages(include=["substratus_amd", "substratus_amd.*"]),
+    # wheels carry the built _C.so + the GEMM tables; kernel SOURCES ride
+    # the sdist via MANIFEST.in (which can exclude the *_hip.hip siblings
+    # torch's hipify regenerates in-tree on every build)
+    package_data={"substratus_amd": ["data/tunables/*.csv"]},
+
     entry_points={"console_scripts": [
         "sub = substratus_amd.cli.main:main",
         "nbwatch = substratus_amd.containertools.nbwatch:watch",

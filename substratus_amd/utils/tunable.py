@@ -28,9 +28,15 @@ _loaded: bool | None = None
 
 
 def tunables_dir() -> str:
-    root = os.path.dirname(
-        os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
-    return os.path.join(root, "tunables")
+    """Canonical table location is package data (ships in wheels/sdists);
+    a repo-root tunables/ directory, if present, overrides it (operational
+    knob for box-local retuning without touching the package)."""
+    pkg = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    root = os.path.dirname(pkg)
+    override = os.path.join(root, "tunables")
+    if os.path.isdir(override):
+        return override
+    return os.path.join(pkg, "data", "tunables")
 
 
 def load_tunables() -> bool:
