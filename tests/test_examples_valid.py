@@ -1,0 +1,67 @@
+"""Every manifest under examples/ must parse, pass API validation, and
+survive admission (reference: examples/* are the documented entry points;
+a drifted example is a broken front door). Server/Job manifests with GPU
+resources additionally go through the HBM admission arithmetic with the
+model size the registry predicts."""
+
+import glob
+import os
+
+import pytest
+import yaml
+
+from substratus_amd.api import types as apiv1
+
+ROOT = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+MANIFESTS = sorted(glob.glob(os.path.join(ROOT, "examples", "**", "*.yaml"),
+                             recursive=True))
+
+
+@pytest.mark.parametrize("path", MANIFESTS,
+                         ids=[os.path.relpath(p, ROOT) for p in MANIFESTS])
+def test_example_manifest_valid(path):
+    docs = [d for d in yaml.safe_load_all(open(path)) if d]
+    assert docs, f"{path} is empty"
+    for obj in docs:
+        assert obj.get("apiVersion") == "substratus.ai/v1", path
+        assert obj.get("kind") in apiv1.KINDS, path
+        apiv1.validate(obj)
+
+
+def test_examples_cover_baseline_configs():
+    """BASELINE.json's five configs each have a manifest."""
+    names = " ".join(MANIFESTS)
+    for frag in ("facebook-opt-125m", "llama2-7b", "falcon-7b-instruct",
+                 "llama2-13b-chat-gguf", "llama2-70b"):
+        assert frag in names, f"no example for baseline config {frag}"
+
+
+def test_gpu_examples_pass_hbm_admission():
+    """Every example Server with a known model + GPU count admits under
+    the 288 GB x count check with the registry's bf16 weight size."""
+    from substratus_amd.cloud.resources import check_hbm_fit
+    from substratus_amd.models.config import get_config, _REGISTRY
+
+    def bf16_bytes(cfg):
+        # embeddings + per-layer (qkv + o + mlp) + head — close enough for
+        # the admission check (exact accounting tested elsewhere)
+        qkv_out = (cfg.num_heads + 2 * cfg.num_kv_heads) * cfg.head_dim
+        per_layer = (cfg.hidden_size * qkv_out +
+                     cfg.num_heads * cfg.head_dim * cfg.hidden_size +
+                     3 * cfg.hidden_size * cfg.intermediate_size)
+        return 2 * (2 * cfg.vocab_size * cfg.hidden_size +
+                    cfg.num_layers * per_layer)
+
+    checked = 0
+    for path in MANIFESTS:
+        for obj in (d for d in yaml.safe_load_all(open(path)) if d):
+            if obj.get("kind") != "Server":
+                continue
+            gpu = ((obj["spec"].get("resources") or {}).get("gpu") or {})
+            model = (obj["spec"].get("model") or {}).get("name", "")
+            if not gpu.get("count") or model not in _REGISTRY:
+                continue
+            check_hbm_fit(gpu["type"], gpu["count"],
+                          bf16_bytes(get_config(model)))
+            checked += 1
+    assert checked >= 2   # at least the 70b pair
