@@ -342,6 +342,17 @@ def create_app(engine: Optional[Engine] = None,
                 # a dropped client closes the generator (GeneratorExit at
                 # the yield); the finally cancels the request so it stops
                 # occupying batch slots and KV blocks
+                from .tokenizer import StreamDecoder
+                dec = StreamDecoder(tok_obj)
+
+                def chunk_for(text, done):
+                    return {
+                        "id": f"cmpl-{sid}", "object": "text_completion",
+                        "model": model_id,
+                        "choices": [{"index": 0, "text": text,
+                                     "finish_reason": "stop" if done
+                                     else None}],
+                    }
                 try:
                     while True:
                         o: StepOutput = out_q.get(timeout=600)
@@ -351,15 +362,14 @@ def create_app(engine: Optional[Engine] = None,
                             yield f"data: {json.dumps(err)}\n\n"
                             yield "data: [DONE]\n\n"
                             return
-                        chunk = {
-                            "id": f"cmpl-{sid}", "object": "text_completion",
-                            "model": model_id,
-                            "choices": [{"index": 0,
-                                         "text": tok_obj.decode([o.token]),
-                                         "finish_reason": "stop" if o.finished
-                                         else None}],
-                        }
-                        yield f"data: {json.dumps(chunk)}\n\n"
+                        text = dec.push(o.token)
+                        if o.finished:
+                            text += dec.flush()
+                        # one chunk per token (empty delta while a
+                        # multi-byte char is still incomplete)
+                        yield ("data: " +
+                               json.dumps(chunk_for(text, o.finished)) +
+                               "\n\n")
                         if o.finished:
                             yield "data: [DONE]\n\n"
                             return
@@ -426,6 +436,8 @@ def create_app(engine: Optional[Engine] = None,
                                    0, top_p, eos_token=tok_obj.eos)
         if stream:
             def gen():
+                from .tokenizer import StreamDecoder
+                dec = StreamDecoder(tok_obj)
                 try:
                     while True:
                         o: StepOutput = out_q.get(timeout=600)
@@ -435,12 +447,15 @@ def create_app(engine: Optional[Engine] = None,
                                            "type": "server_error"}}) + "\n\n")
                             yield "data: [DONE]\n\n"
                             return
+                        text = dec.push(o.token)
+                        if o.finished:
+                            text += dec.flush()
                         chunk = {
                             "id": f"chatcmpl-{sid}",
                             "object": "chat.completion.chunk",
                             "model": model_id,
                             "choices": [{"index": 0, "delta": {
-                                "content": tok_obj.decode([o.token])},
+                                "content": text},
                                 "finish_reason": "stop" if o.finished
                                 else None}],
                         }

@@ -77,6 +77,38 @@ class HFTokenizer:
                              skip_special_tokens=True)
 
 
+class StreamDecoder:
+    """Incremental detokenizer for SSE streaming: decoding each token id
+    independently emits U+FFFD when a multi-byte UTF-8 character spans
+    token boundaries (byte-level BPE splits them freely). This decodes the
+    CUMULATIVE id list and emits the delta, holding back a trailing
+    replacement char until the next token resolves (or flush() at
+    end-of-stream emits it as-is for genuinely invalid bytes)."""
+
+    def __init__(self, tok):
+        self.tok = tok
+        self.ids: list[int] = []
+        self.emitted = 0            # chars of the cumulative text sent
+
+    def push(self, token_id: int) -> str:
+        self.ids.append(token_id)
+        text = self.tok.decode(self.ids)
+        safe = len(text)
+        if text.endswith("�"):
+            safe -= 1               # maybe a partial sequence: hold back
+        if safe <= self.emitted:
+            return ""
+        out = text[self.emitted:safe]
+        self.emitted = safe
+        return out
+
+    def flush(self) -> str:
+        text = self.tok.decode(self.ids)
+        out = text[self.emitted:]
+        self.emitted = len(text)
+        return out
+
+
 def load(model_dir: Optional[str], vocab_size: int):
     """The server's tokenizer: artifacts tokenizer.json if present, else the
     byte-level fallback."""

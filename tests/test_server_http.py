@@ -188,6 +188,27 @@ def test_worker_survives_engine_failure(tmp_path):
         assert ok.json()["usage"]["completion_tokens"] == 2
 
 
+def test_stream_decoder_multibyte():
+    """SSE streaming must never emit U+FFFD for multi-byte characters that
+    span token boundaries (byte-level BPE splits them freely): the
+    incremental decoder holds back partial sequences and the concatenated
+    deltas reproduce the exact text."""
+    from substratus_amd.serve.tokenizer import ByteTokenizer, StreamDecoder
+    tok = ByteTokenizer(32000)
+    text = "héllo ✓ —丸 ok"
+    ids = tok.encode(text, add_bos=False)
+    dec = StreamDecoder(tok)
+    parts = [dec.push(t) for t in ids]
+    parts.append(dec.flush())
+    assert "".join(parts) == text
+    assert all("\ufffd" not in p for p in parts)
+    # genuinely invalid bytes still flush (as replacement) at end-of-stream
+    dec2 = StreamDecoder(tok)
+    bad = tok.encode("a", add_bos=False) + [2 + 0xC3]  # dangling lead byte
+    out = "".join([dec2.push(t) for t in bad]) + dec2.flush()
+    assert out.startswith("a") and out.endswith("\ufffd")
+
+
 def test_streaming_client_disconnect_aborts(tmp_path):
     """Dropping a streaming client mid-generation cancels the request: the
     engine stops spending steps on it, frees its KV blocks, and the next
