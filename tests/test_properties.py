@@ -126,3 +126,21 @@ def test_kquant_repack_involution(seed, quant):
     packed = QUANT_CODECS[quant][2](w)
     s, qs = repack_k(packed, 768, quant)
     assert torch.equal(unrepack_k(s, qs, 768, quant), packed)
+
+
+@given(st.text(min_size=0, max_size=200))
+@settings(max_examples=60, deadline=None)
+def test_stream_decoder_exact_for_any_text(text):
+    """StreamDecoder pushed one token at a time reconstructs ANY unicode
+    text exactly (byte-level tokenizer splits multi-byte chars across
+    tokens freely), and no intermediate delta ever contains U+FFFD."""
+    from hypothesis import assume
+    from substratus_amd.serve.tokenizer import ByteTokenizer, StreamDecoder
+    assume("�" not in text)   # a literal U+FFFD in the INPUT is emitted as-is
+    tok = ByteTokenizer(32000)
+    ids = tok.encode(text, add_bos=False)
+    dec = StreamDecoder(tok)
+    parts = [dec.push(t) for t in ids]
+    parts.append(dec.flush())
+    assert "".join(parts) == text
+    assert all("�" not in p for p in parts)
