@@ -8,7 +8,7 @@ llama2-70b decode (M=512): o-proj 1.4 -> 1.94 TB/s, down-proj
 
 Tables are produced offline on a target box (see docs/PERF.md; run any
 bench with PYTORCH_TUNABLEOP_ENABLED=1 PYTORCH_TUNABLEOP_TUNING=1 and an
-eager decode path) and shipped under tunables/<arch>.csv. At runtime we
+eager decode path) and shipped under substratus_amd/data/tunables/<arch>.csv (repo-root tunables/ overrides). At runtime we
 load them READ-ONLY: tuning stays disabled, untuned shapes keep the
 heuristic default, and a version/arch validator mismatch simply skips
 the table. Explicitly setting PYTORCH_TUNABLEOP_ENABLED (either value)
