@@ -397,9 +397,15 @@ def resume(name: str, namespace: str = typer.Option("default", "-n")):
 @app.command()
 def infer(server: str = typer.Option(..., "--server", "-s"),
           prompt: str = typer.Argument(...),
-          max_tokens: int = typer.Option(32)):
+          max_tokens: int = typer.Option(32),
+          stream: bool = typer.Option(False, "--stream",
+                                      help="print tokens as they arrive "
+                                           "(SSE; ctrl-c aborts the "
+                                           "request server-side)")):
     """One completion request against a ready Server
     (reference internal/cli/infer.go — the chat TUI stub, made functional)."""
+    import json as _json
+
     import httpx
     c = _client()
     svc = c.get("Service", f"{server}-server")
@@ -408,6 +414,26 @@ def infer(server: str = typer.Option(..., "--server", "-s"),
     if not ep:
         console.print(f"[red]server {server} has no local endpoint[/red]")
         raise typer.Exit(1)
+    if stream:
+        import sys
+        with httpx.stream("POST", f"{ep}/v1/completions",
+                          json={"prompt": prompt, "max_tokens": max_tokens,
+                                "stream": True}, timeout=300) as r:
+            r.raise_for_status()
+            for line in r.iter_lines():
+                if not line.startswith("data:"):
+                    continue
+                payload = line[5:].strip()
+                if payload == "[DONE]":
+                    break
+                obj = _json.loads(payload)
+                if "error" in obj:
+                    console.print(f"[red]{obj['error']['message']}[/red]")
+                    raise typer.Exit(1)
+                sys.stdout.write(obj["choices"][0]["text"])
+                sys.stdout.flush()
+        sys.stdout.write("\n")
+        return
     r = httpx.post(f"{ep}/v1/completions",
                    json={"prompt": prompt, "max_tokens": max_tokens},
                    timeout=300)

@@ -126,6 +126,10 @@ def test_serve_and_infer_end_to_end(daemon, tmp_path):
     r = runner.invoke(cli.app, ["infer", "--server", "cli-srv",
                                 "hello world"])
     assert r.exit_code == 0, r.output
+    # streaming variant rides SSE through the same endpoint
+    r = runner.invoke(cli.app, ["infer", "--server", "cli-srv", "--stream",
+                                "--max-tokens", "3", "hello again"])
+    assert r.exit_code == 0, r.output
 
 
 def test_notebook_conversion(daemon, tmp_path):
