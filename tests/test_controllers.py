@@ -459,3 +459,42 @@ def test_gpu_pods_get_num_gpus_env(harness):
     envs = {e["name"]: e.get("value") for e in c.get("env", [])}
     assert envs.get("NUM_GPUS") == "8"
     assert c["resources"]["limits"]["amd.com/gpu"] == "8"
+
+
+def test_control_plane_churn_leaves_no_residue(harness):
+    """10 create -> ready -> delete cycles across Model+Server: the
+    owner-reference cascade and reconcilers must leave the store empty
+    every cycle (no leaked Jobs/Deployments/Services/ConfigMaps/Pods) and
+    keep working after repeated reuse of the same names."""
+    kube, cloud, sci, mgr = harness
+    child_kinds = ("Job", "Deployment", "Service", "ConfigMap", "Pod")
+    for cycle in range(10):
+        kube.create(apiv1.new_object("Model", "churn-m", {
+            "image": "substratus-amd/model-loader",
+            "params": {"name": "llama-tiny"}}))
+        for _ in range(4):
+            mgr._enqueue(("Model", "churn-m", "default"))
+            mgr.run_once()
+        fake_job_complete(kube, "churn-m-modeller")
+        for _ in range(3):
+            mgr._enqueue(("Model", "churn-m", "default"))
+            mgr.run_once()
+        assert apiv1.get_status_ready(kube.get("Model", "churn-m")), cycle
+        kube.create(apiv1.new_object("Server", "churn-s", {
+            "image": "substratus-amd/model-server",
+            "model": {"name": "churn-m"}}))
+        for _ in range(4):
+            mgr._enqueue(("Server", "churn-s", "default"))
+            mgr.run_once()
+        fake_deployment_ready(kube, "churn-s-server")
+        for _ in range(3):
+            mgr._enqueue(("Server", "churn-s", "default"))
+            mgr.run_once()
+        assert apiv1.get_status_ready(kube.get("Server", "churn-s")), cycle
+        kube.delete("Server", "churn-s")
+        kube.delete("Model", "churn-m")
+        mgr.run_once()
+        for kind in child_kinds + ("Model", "Server"):
+            left = [o["metadata"]["name"] for o in kube.list(kind)
+                    if o["metadata"]["name"].startswith("churn")]
+            assert not left, (cycle, kind, left)
