@@ -65,3 +65,16 @@ def test_gpu_examples_pass_hbm_admission():
                           bf16_bytes(get_config(model)))
             checked += 1
     assert checked >= 2   # at least the 70b pair
+
+
+def test_crd_yaml_in_sync_with_generator():
+    """config/crd/bases/*.yaml must match api/crd.py's render exactly —
+    editing one without the other ships a drifted CRD."""
+    from substratus_amd.api.crd import render_all
+    base = os.path.join(ROOT, "config", "crd", "bases")
+    rendered = render_all()
+    on_disk = set(os.listdir(base))
+    assert on_disk == set(rendered), (on_disk, set(rendered))
+    for name, text in rendered.items():
+        disk = open(os.path.join(base, name)).read()
+        assert disk == text, f"{name} drifted from api/crd.py (re-render)"
