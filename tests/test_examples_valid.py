@@ -102,3 +102,28 @@ def test_install_manifests_parse_and_cover_the_stack():
     up = open(os.path.join(ROOT, "install/kind/up.sh")).read()
     assert "config/crd/bases" in up
     assert os.path.isdir(os.path.join(ROOT, "config/crd/bases"))
+
+
+def test_image_entrypoints_are_importable_modules():
+    """Every Dockerfile ENTRYPOINT that runs `python -m X` must name a
+    module that exists and exposes a main/entry — a renamed module would
+    ship an image that dies at container start."""
+    import importlib
+    import re
+    seen = 0
+    for df in glob.glob(os.path.join(ROOT, "images", "Dockerfile.*")):
+        text = open(df).read()
+        for mod in re.findall(r"-m[\",\s]+([a-zA-Z0-9_.]+)", text):
+            if not mod.startswith("substratus_amd"):
+                continue
+            m = importlib.import_module(mod)
+            runnable = (hasattr(m, "main") or hasattr(m, "run") or
+                        hasattr(m, "watch"))
+            if not runnable and hasattr(m, "__path__"):   # package: -m runs
+                runnable = os.path.exists(os.path.join(   # its __main__.py
+                    list(m.__path__)[0], "__main__.py"))
+            if not runnable:
+                runnable = '__name__ == "__main__"' in open(m.__file__).read()
+            assert runnable, mod
+            seen += 1
+    assert seen >= 6, f"only {seen} entrypoint modules found"
