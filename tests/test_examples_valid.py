@@ -127,3 +127,18 @@ def test_image_entrypoints_are_importable_modules():
             assert runnable, mod
             seen += 1
     assert seen >= 6, f"only {seen} entrypoint modules found"
+
+
+def test_every_source_dir_is_a_real_package():
+    """A subdirectory with .py files but no __init__.py imports fine
+    in-tree (namespace package) but silently VANISHES from wheels
+    (find_packages skips it) — exactly how utils/ went missing from the
+    round-2 release artifact. Keep every source dir a real package."""
+    pkg_root = os.path.join(ROOT, "substratus_amd")
+    for dirpath, dirnames, filenames in os.walk(pkg_root):
+        dirnames[:] = [d for d in dirnames
+                       if d not in ("__pycache__", "hip", "data")]
+        if any(f.endswith(".py") for f in filenames):
+            assert "__init__.py" in filenames or dirpath == pkg_root or \
+                os.path.exists(os.path.join(dirpath, "__init__.py")), \
+                f"{dirpath} has .py files but no __init__.py"
