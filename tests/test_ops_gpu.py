@@ -132,11 +132,31 @@ def test_rope_kv_append(d, theta):
                                       (8, 2, 128), (16, 1, 128),
                                       (16, 1, 64), (8, 2, 64)])
 def test_paged_attn_decode(hq, hkv, d):
+    _run_paged_attn_decode_case(hq, hkv, d, bsz=5,
+                                lens=[1, 16, 17, 130, 77])
+
+
+@pytest.mark.parametrize("hq,hkv,d,bsz", [
+    (32, 32, 128, 64),    # MHA, base=2048 -> nsplit==1 (production path)
+    (64, 8, 128, 128),    # MFMA GQA, base=1024 -> nsplit==1
+    (128, 8, 64, 128),    # MFMA GQA d=64 HPW=4, nsplit==1
+])
+def test_paged_attn_decode_nsplit1(hq, hkv, d, bsz):
+    """The small-batch parametrization above always lands in the
+    flash-decoding nsplit>1 regime (base < 1024 forces 16-way splits);
+    these shapes force nsplit==1 so the kernels' DIRECT-output epilogues
+    — the path every production batch>=64 decode takes — get the same
+    fp32-reference parity check."""
+    torch.manual_seed(11)
+    lens = [1 + (i * 37) % 60 for i in range(bsz)]
+    _run_paged_attn_decode_case(hq, hkv, d, bsz=bsz, lens=lens)
+
+
+def _run_paged_attn_decode_case(hq, hkv, d, bsz, lens):
     from substratus_amd import ops
     from substratus_amd.ops import ref
     torch.manual_seed(7)
-    bsz, bs = 5, 16
-    lens = [1, 16, 17, 130, 77][:bsz]
+    bs = 16
     maxb = (max(lens) + bs - 1) // bs
     nb = sum((n + bs - 1) // bs for n in lens) + 3
     kc = torch.randn(nb, hkv, bs, d, device=DEV).bfloat16()
