@@ -512,18 +512,23 @@ def test_prefill_matches_decode_continuation():
 
 
 # ------------------------------------------------------------- fp8 KV cache
-@pytest.mark.parametrize("hq,hkv", [(8, 4), (8, 1), (16, 1)])
-def test_paged_attn_decode_fp8_kv(hq, hkv):
+@pytest.mark.parametrize("hq,hkv,bsz", [(8, 4, 4), (8, 1, 4), (16, 1, 4),
+                                        # nsplit==1 production regime
+                                        (64, 8, 128)])
+def test_paged_attn_decode_fp8_kv(hq, hkv, bsz):
     """fp8 e4m3 KV cache vs an fp32 reference computed from the SAME
     quantized values (the quantization error itself is checked separately).
-    (8,4)=register path; (8,1)/(16,1)=LDS-staged GQA path HPW 2/4."""
+    (8,4)=register path; (8,1)/(16,1)=MFMA GQA path; (64,8,128) forces
+    the nsplit==1 direct-output epilogue (the fp8@batch-1024 serving
+    config's path)."""
     from substratus_amd.ops import ref
     from substratus_amd import ops
     torch.manual_seed(40)
-    bsz, d, bs = 4, 128, 16
-    lens = [1, 17, 64, 33]
-    maxb = 4
-    nb = 20
+    d, bs = 128, 16
+    lens = ([1, 17, 64, 33] if bsz == 4
+            else [1 + (i * 37) % 60 for i in range(bsz)])
+    maxb = (max(lens) + bs - 1) // bs
+    nb = bsz * maxb + 4
     kc8 = (torch.randn(nb, hkv, bs, d, device=DEV) * 0.5).to(
         torch.float8_e4m3fn)
     vc8 = (torch.randn(nb, hkv, bs, d, device=DEV) * 0.5).to(
