@@ -86,6 +86,20 @@ register(ModelConfig(
     name="llama2-70b", family="llama", hidden_size=8192, intermediate_size=28672,
     num_layers=80, num_heads=64, num_kv_heads=8, vocab_size=32000,
     max_position_embeddings=4096, rms_norm_eps=1e-5))
+# Llama-3 family (beyond reference parity — same llama weight layout, so
+# HF checkpoints map via the existing name translation; GQA groups 4 and 8
+# ride the MFMA decode kernel, rope theta 500k, 128k vocab).
+register(ModelConfig(
+    name="llama3-8b", family="llama", hidden_size=4096,
+    intermediate_size=14336, num_layers=32, num_heads=32, num_kv_heads=8,
+    vocab_size=128256, max_position_embeddings=8192, rope_theta=500000.0,
+    rms_norm_eps=1e-5))
+register(ModelConfig(
+    name="llama3-70b", family="llama", hidden_size=8192,
+    intermediate_size=28672, num_layers=80, num_heads=64, num_kv_heads=8,
+    vocab_size=128256, max_position_embeddings=8192, rope_theta=500000.0,
+    rms_norm_eps=1e-5))
+
 # A tiny llama for CPU tests and CI.
 # Tiny llama for CPU tests/CI and GPU numerics tests (head_dim=64, GQA=2).
 register(ModelConfig(

@@ -142,3 +142,24 @@ def test_every_source_dir_is_a_real_package():
             assert "__init__.py" in filenames or dirpath == pkg_root or \
                 os.path.exists(os.path.join(dirpath, "__init__.py")), \
                 f"{dirpath} has .py files but no __init__.py"
+
+
+def test_llama3_family_configs():
+    """llama3-8b/70b (beyond reference parity): llama layout, GQA 4/8
+    (the MFMA decode path), rope theta 500k reaches the blocks, TP-8
+    divisible, param counts match the published sizes."""
+    import torch
+    from substratus_amd.models import CausalLM
+    from substratus_amd.models.config import get_config
+    for name, params, group in (("llama3-8b", 8.0e9, 4),
+                                ("llama3-70b", 70.6e9, 8)):
+        cfg = get_config(name)
+        assert cfg.num_heads // cfg.num_kv_heads == group
+        assert abs(cfg.param_count() - params) / params < 0.02, \
+            (name, cfg.param_count())
+        for tp in (2, 4, 8):
+            assert cfg.num_heads % tp == 0
+            assert cfg.num_kv_heads % tp == 0 or cfg.num_kv_heads < tp
+            assert cfg.intermediate_size % tp == 0
+        m = CausalLM(cfg, device="meta", dtype=torch.bfloat16)
+        assert m.blocks[0].rope_theta == 500000.0
